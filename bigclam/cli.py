@@ -1,0 +1,153 @@
+"""Command-line entry point: ``python -m bigclam <cmd>``.
+
+The reference has no CLI/config surface at all (hard-coded paths and vars,
+SURVEY.md §2.15); this exposes the full hyperparameter set with the
+reference's defaults.
+
+  fit        fit a K-community model on an edge list, write communities + ckpt
+  select-k   model selection over the geometric K grid (v4 semantics)
+  extract    re-extract communities from a checkpoint
+  bench      one timed sweep loop (see also bench.py at the repo root)
+"""
+from __future__ import annotations
+
+import argparse
+import dataclasses
+import json
+import sys
+
+import numpy as np
+import torch
+
+from .config import BigClamConfig
+from . import comm
+from .engine.extract import extract_communities, write_communities
+from .engine.model_select import select_k
+from .engine.trainer import Trainer
+from .io import load_graph, rmat_graph
+from .utils.metrics import MetricsLogger
+
+
+def _add_common(p: argparse.ArgumentParser):
+    d = BigClamConfig()
+    p.add_argument("--k", type=int, default=d.k)
+    p.add_argument("--alpha", type=float, default=d.alpha)
+    p.add_argument("--beta", type=float, default=d.beta)
+    p.add_argument("--ls-steps", type=int, default=d.ls_steps)
+    p.add_argument("--tol", type=float, default=d.tol)
+    p.add_argument("--max-sweeps", type=int, default=d.max_sweeps)
+    p.add_argument("--dtype", choices=["fp32", "bf16"], default=d.dtype)
+    p.add_argument("--device", default=None, help="cuda|cpu (auto)")
+    p.add_argument("--seed", type=int, default=d.seed)
+    p.add_argument("--init", choices=["seed", "random"], default="seed")
+    p.add_argument("--seed-rank-compat", action="store_true")
+    p.add_argument("--checkpoint-every", type=int, default=0)
+    p.add_argument("--checkpoint-dir", default=None)
+    p.add_argument("--out", default=None)
+    p.add_argument("--metrics", default=None, help="JSONL metrics path")
+    p.add_argument("--quiet", action="store_true")
+
+
+def _cfg_from(args) -> BigClamConfig:
+    device = args.device or ("cuda" if torch.cuda.is_available() else "cpu")
+    return BigClamConfig(
+        k=args.k,
+        alpha=args.alpha,
+        beta=args.beta,
+        ls_steps=args.ls_steps,
+        tol=args.tol,
+        max_sweeps=args.max_sweeps,
+        dtype=args.dtype,
+        device=device,
+        seed=args.seed,
+        seed_rank_compat=args.seed_rank_compat,
+        checkpoint_every=args.checkpoint_every,
+        checkpoint_dir=args.checkpoint_dir,
+        out=args.out,
+        k_min=getattr(args, "k_min", 1000),
+        k_max=getattr(args, "k_max", 9000),
+        k_div=getattr(args, "k_div", 100),
+        k_tol=getattr(args, "k_tol", 1e-3),
+    )
+
+
+def _load(args):
+    if args.edgelist.startswith("rmat:"):
+        # rmat:<scale>:<edge_factor> synthetic graph
+        _, scale, ef = args.edgelist.split(":")
+        return rmat_graph(int(scale), float(ef))
+    return load_graph(args.edgelist)
+
+
+def main(argv=None):
+    ap = argparse.ArgumentParser(prog="bigclam")
+    sub = ap.add_subparsers(dest="cmd", required=True)
+
+    p_fit = sub.add_parser("fit")
+    p_fit.add_argument("edgelist")
+    _add_common(p_fit)
+
+    p_sel = sub.add_parser("select-k")
+    p_sel.add_argument("edgelist")
+    _add_common(p_sel)
+    p_sel.add_argument("--k-min", dest="k_min", type=int, default=1000)
+    p_sel.add_argument("--k-max", dest="k_max", type=int, default=9000)
+    p_sel.add_argument("--k-div", dest="k_div", type=int, default=100)
+    p_sel.add_argument("--k-tol", dest="k_tol", type=float, default=1e-3)
+
+    p_ext = sub.add_parser("extract")
+    p_ext.add_argument("checkpoint_dir")
+    p_ext.add_argument("edgelist")
+    p_ext.add_argument("--out", required=True)
+
+    args = ap.parse_args(argv)
+    rank = comm.init_distributed()
+
+    if args.cmd == "extract":
+        from .ckpt.checkpoint import load_full_F, load_meta
+
+        g = _load(args)
+        F = torch.from_numpy(load_full_F(args.checkpoint_dir))
+        members = extract_communities(F, g.num_edges)
+        write_communities(args.out, members, g.raw_ids)
+        print(json.dumps({"communities": sum(1 for m in members if len(m))}))
+        return 0
+
+    cfg = _cfg_from(args)
+    g = _load(args)
+    metrics = MetricsLogger(args.metrics, rank=rank, quiet=args.quiet)
+
+    if args.cmd == "select-k":
+        out = select_k(g, cfg, metrics=metrics, init=args.init)
+        if rank == 0:
+            print(json.dumps(out))
+        return 0
+
+    tr = Trainer(g, cfg, metrics=metrics)
+    res = tr.fit(init=args.init)
+    F = tr.gather_F()
+    if cfg.checkpoint_dir:  # every rank writes its own shard file
+        from .ckpt.checkpoint import save_shard_checkpoint
+
+        save_shard_checkpoint(cfg.checkpoint_dir, tr, res.sweeps, res.llh)
+    if rank == 0:
+        if cfg.out:
+            members = extract_communities(F, g.num_edges)
+            write_communities(cfg.out, members, g.raw_ids)
+        print(
+            json.dumps(
+                {
+                    "llh": res.llh,
+                    "sweeps": res.sweeps,
+                    "converged": res.converged,
+                    "n": g.num_nodes,
+                    "edges": g.num_edges,
+                    "k": cfg.k,
+                }
+            )
+        )
+    return 0
+
+
+if __name__ == "__main__":
+    sys.exit(main())

@@ -1,0 +1,121 @@
+"""Edge-list ingest: parse, ID-remap, undirected adjacency.
+
+Input contract (same as the reference's ``GraphLoader.edgeListFile``,
+codes/bigclamv3-7.scala:26): one edge per line, ``src<ws>dst``, lines starting
+with ``#`` are comments.  Node IDs are arbitrary non-negative integers.
+
+Unlike the reference — which keeps raw (possibly non-contiguous) GraphX vertex
+IDs and papers over the resulting lookup holes with zero rows
+(``Flookup``, codes/bigclamv3-7.scala:94-104) — we remap IDs to a dense
+``0..N-1`` internal space and keep the ``internal -> raw`` table for output.
+
+Deduplication: SNAP files store undirected graphs either as both directions
+(Email-Enron) or one direction (facebook_combined).  The reference's
+``collectNeighborIds(Either)`` therefore yields neighbor arrays with
+duplicates for both-direction files, silently doubling every edge term.  We
+normalize instead: the canonical internal form is the simple undirected graph
+(self-loops dropped, duplicates merged); every undirected edge appears in both
+endpoints' CSR rows, so each edge is counted exactly twice per LLH sweep —
+the convention of the reference's single-direction datasets.
+"""
+from __future__ import annotations
+
+import io
+import os
+from dataclasses import dataclass
+
+import numpy as np
+
+
+@dataclass
+class Graph:
+    """Undirected graph in CSR form over dense internal IDs.
+
+    ``indptr``/``indices`` cover both directions: ``indices[indptr[u]:indptr[u+1]]``
+    are u's neighbors.  ``raw_ids[i]`` is the original file ID of internal
+    node ``i``.  ``num_edges`` is the undirected edge count
+    (= len(indices) // 2).
+    """
+
+    indptr: np.ndarray  # int64 [N+1]
+    indices: np.ndarray  # int32 [2E]
+    raw_ids: np.ndarray  # int64 [N]
+
+    @property
+    def num_nodes(self) -> int:
+        return len(self.indptr) - 1
+
+    @property
+    def num_edges(self) -> int:
+        return len(self.indices) // 2
+
+    @property
+    def num_directed_edges(self) -> int:
+        return len(self.indices)
+
+    def degrees(self) -> np.ndarray:
+        return np.diff(self.indptr)
+
+    def neighbors(self, u: int) -> np.ndarray:
+        return self.indices[self.indptr[u] : self.indptr[u + 1]]
+
+
+def parse_edge_array(path: str) -> np.ndarray:
+    """Read an edge-list file into an int64 [M, 2] array (fast path: pandas)."""
+    try:
+        import pandas as pd
+
+        df = pd.read_csv(
+            path,
+            sep=r"\s+",
+            comment="#",
+            header=None,
+            usecols=[0, 1],
+            dtype=np.int64,
+            engine="c",
+        )
+        return df.values
+    except Exception:
+        rows = []
+        with io.open(path, "r") as f:
+            for line in f:
+                line = line.strip()
+                if not line or line.startswith("#"):
+                    continue
+                a, b = line.split()[:2]
+                rows.append((int(a), int(b)))
+        return np.asarray(rows, dtype=np.int64).reshape(-1, 2)
+
+
+def build_graph(edges: np.ndarray, drop_self_loops: bool = True) -> Graph:
+    """Build the canonical undirected CSR from a raw [M, 2] edge array."""
+    edges = np.asarray(edges, dtype=np.int64).reshape(-1, 2)
+    if drop_self_loops:
+        edges = edges[edges[:, 0] != edges[:, 1]]
+    # dense ID remap
+    raw_ids, remapped = np.unique(edges.reshape(-1), return_inverse=True)
+    e = remapped.reshape(-1, 2).astype(np.int64)
+    n = len(raw_ids)
+    # canonicalize (min, max) and dedupe
+    lo = np.minimum(e[:, 0], e[:, 1])
+    hi = np.maximum(e[:, 0], e[:, 1])
+    key = lo * n + hi
+    uniq = np.unique(key)
+    lo = (uniq // n).astype(np.int64)
+    hi = (uniq % n).astype(np.int64)
+    # both directions
+    src = np.concatenate([lo, hi])
+    dst = np.concatenate([hi, lo])
+    indptr = np.zeros(n + 1, dtype=np.int64)
+    np.add.at(indptr, src + 1, 1)
+    np.cumsum(indptr, out=indptr)
+    # sort neighbors within each row: lexsort by (src, dst)
+    order = np.lexsort((dst, src))
+    indices = dst[order].astype(np.int32)
+    return Graph(indptr=indptr, indices=indices, raw_ids=raw_ids)
+
+
+def load_graph(path: str) -> Graph:
+    if not os.path.exists(path):
+        raise FileNotFoundError(path)
+    return build_graph(parse_edge_array(path))

@@ -1,0 +1,112 @@
+"""Synthetic graph generators.
+
+There is no network access and the reference's large datasets (com-Youtube,
+com-Amazon) are missing blobs, so benchmarks and tests run on synthetic
+graphs of the same shape (BASELINE.json: "synthetic graphs / random-init F").
+
+- :func:`rmat`: R-MAT power-law graphs (Chakrabarti et al. 2004) — the
+  100M-edge 8-GPU config and the com-Amazon-shaped bench graph.
+- :func:`planted_partition`: graphs with known ground-truth communities for
+  recovery tests (SURVEY.md §4 "planted-partition graphs").
+
+R-MAT has no counterpart in the reference (SURVEY.md §2-B note).
+"""
+from __future__ import annotations
+
+import numpy as np
+
+from .edgelist import Graph, build_graph
+
+
+def rmat_edges(
+    scale: int,
+    edge_factor: float,
+    a: float = 0.57,
+    b: float = 0.19,
+    c: float = 0.19,
+    seed: int = 0,
+) -> np.ndarray:
+    """Generate ~``edge_factor * 2**scale`` R-MAT edge pairs (with dups/loops).
+
+    Vectorized bit-by-bit quadrant sampling; IDs are scrambled so hubs are not
+    clustered at low indices.
+    """
+    n = 1 << scale
+    m = int(edge_factor * n)
+    rng = np.random.default_rng(seed)
+    src = np.zeros(m, dtype=np.int64)
+    dst = np.zeros(m, dtype=np.int64)
+    ab = a + b
+    a_norm = a / ab
+    c_norm = c / (1.0 - ab)
+    for _ in range(scale):
+        r1 = rng.random(m)
+        r2 = rng.random(m)
+        src_bit = r1 > ab
+        dst_bit = np.where(src_bit, r2 > c_norm, r2 > a_norm)
+        src = (src << 1) | src_bit
+        dst = (dst << 1) | dst_bit
+    # scramble ids (fixed permutation derived from seed)
+    perm = rng.permutation(n)
+    return np.stack([perm[src], perm[dst]], axis=1)
+
+
+def rmat_graph(scale: int, edge_factor: float = 16.0, seed: int = 0) -> Graph:
+    return build_graph(rmat_edges(scale, edge_factor, seed=seed))
+
+
+def rmat_graph_with_edges(
+    target_nodes: int, target_edges: int, seed: int = 0
+) -> Graph:
+    """R-MAT graph shaped like (target_nodes, target_edges undirected).
+
+    Used by bench.py to mimic named SNAP datasets (e.g. com-Amazon: 335k
+    nodes / 926k edges) without the actual files.  Oversamples to compensate
+    for dedup losses, then trims.
+    """
+    scale = max(1, int(np.ceil(np.log2(target_nodes))))
+    # oversample: R-MAT dedup + self-loop losses are modest at low density
+    factor = 1.3 * target_edges / (1 << scale)
+    edges = rmat_edges(scale, factor, seed=seed)
+    g = build_graph(edges)
+    if g.num_edges > target_edges:
+        # trim by dropping a random subset of undirected edges
+        rng = np.random.default_rng(seed + 1)
+        lo_mask = g.indices.astype(np.int64) > np.repeat(
+            np.arange(g.num_nodes, dtype=np.int64), g.degrees()
+        )
+        src = np.repeat(np.arange(g.num_nodes, dtype=np.int64), g.degrees())
+        und_src = src[lo_mask]
+        und_dst = g.indices[lo_mask].astype(np.int64)
+        keep = rng.choice(len(und_src), size=target_edges, replace=False)
+        g = build_graph(np.stack([und_src[keep], und_dst[keep]], axis=1))
+    return g
+
+
+def planted_partition(
+    num_communities: int,
+    nodes_per_community: int,
+    p_in: float = 0.3,
+    p_out: float = 0.01,
+    seed: int = 0,
+) -> tuple:
+    """Planted-partition graph; returns (Graph, labels[N])."""
+    rng = np.random.default_rng(seed)
+    n = num_communities * nodes_per_community
+    labels = np.repeat(np.arange(num_communities), nodes_per_community)
+    # sample within-community edges
+    edges = []
+    for c in range(num_communities):
+        base = c * nodes_per_community
+        iu, ju = np.triu_indices(nodes_per_community, k=1)
+        mask = rng.random(len(iu)) < p_in
+        edges.append(np.stack([base + iu[mask], base + ju[mask]], axis=1))
+    # sample cross edges sparsely
+    m_out = int(p_out * n * n / 2)
+    if m_out:
+        s = rng.integers(0, n, m_out)
+        d = rng.integers(0, n, m_out)
+        mask = labels[s] != labels[d]
+        edges.append(np.stack([s[mask], d[mask]], axis=1))
+    all_edges = np.concatenate(edges, axis=0)
+    return build_graph(all_edges), labels

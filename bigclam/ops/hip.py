@@ -1,0 +1,100 @@
+"""GPU op wrappers around the in-tree HIP extension ``bigclam._C``.
+
+The extension is built by ``setup.py build_ext --inplace`` (or
+``__graft_entry__.build()``) with ``hipcc --offload-arch=gfx950`` and ships
+in-tree so the gpurun snapshot carries it.  On a GPU box a missing extension
+is a hard error — there is deliberately no silent eager fallback here
+(see core/state.py).
+"""
+from __future__ import annotations
+
+from typing import Tuple
+
+import torch
+
+from ..config import BigClamConfig
+
+_C = None
+
+
+def ensure_loaded():
+    global _C
+    if _C is None:
+        try:
+            from .. import _C as ext
+        except ImportError as e:  # pragma: no cover
+            raise RuntimeError(
+                "bigclam._C HIP extension not built. Run "
+                "`python setup.py build_ext --inplace` (gfx950) first. "
+                f"Original error: {e}"
+            ) from e
+        _C = ext
+    return _C
+
+
+def edge_grad_llh(
+    F: torch.Tensor,
+    indptr: torch.Tensor,
+    indices: torch.Tensor,
+    sumF: torch.Tensor,
+    order: torch.Tensor,
+    cfg: BigClamConfig,
+) -> Tuple[torch.Tensor, torch.Tensor]:
+    ext = ensure_loaded()
+    n_local = len(indptr) - 1
+    grad = torch.empty(n_local, F.shape[1], device=F.device, dtype=torch.float32)
+    llh = torch.empty(n_local, device=F.device, dtype=torch.float64)
+    ext.edge_grad_llh(
+        F, indptr, indices, sumF, order, grad, llh, cfg.min_p, cfg.max_p
+    )
+    return grad, llh
+
+
+def linesearch(
+    F: torch.Tensor,
+    indptr: torch.Tensor,
+    indices: torch.Tensor,
+    sumF: torch.Tensor,
+    grad: torch.Tensor,
+    llh: torch.Tensor,
+    order: torch.Tensor,
+    cfg: BigClamConfig,
+) -> torch.Tensor:
+    ext = ensure_loaded()
+    n_local = len(indptr) - 1
+    best = torch.empty(n_local, device=F.device, dtype=torch.float32)
+    ladder = torch.tensor(
+        cfg.ladder(), device=F.device, dtype=torch.float32
+    )
+    ext.linesearch(
+        F,
+        indptr,
+        indices,
+        sumF,
+        grad,
+        llh,
+        order,
+        ladder,
+        best,
+        cfg.alpha,
+        cfg.min_p,
+        cfg.max_p,
+        cfg.min_f,
+        cfg.max_f,
+    )
+    return best
+
+
+def full_llh(
+    F: torch.Tensor,
+    indptr: torch.Tensor,
+    indices: torch.Tensor,
+    sumF: torch.Tensor,
+    order: torch.Tensor,
+    cfg: BigClamConfig,
+) -> torch.Tensor:
+    ext = ensure_loaded()
+    n_local = len(indptr) - 1
+    llh = torch.empty(n_local, device=F.device, dtype=torch.float64)
+    ext.llh_only(F, indptr, indices, sumF, order, llh, cfg.min_p, cfg.max_p)
+    return llh.sum()

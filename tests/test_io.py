@@ -1,0 +1,63 @@
+import numpy as np
+import pytest
+
+from bigclam.io import build_graph, load_graph, planted_partition, rmat_graph
+from bigclam.io.edgelist import parse_edge_array
+
+
+def test_build_graph_dedup_and_remap():
+    # duplicate edges (both directions), self loop, gappy raw ids
+    edges = np.array([[10, 20], [20, 10], [10, 30], [30, 30], [20, 30], [10, 20]])
+    g = build_graph(edges)
+    assert g.num_nodes == 3
+    assert g.num_edges == 3
+    assert g.num_directed_edges == 6
+    assert list(g.raw_ids) == [10, 20, 30]
+    # node 0 (raw 10) neighbors: raw 20, 30 -> internal 1, 2
+    assert sorted(g.neighbors(0).tolist()) == [1, 2]
+    # symmetric
+    for u in range(3):
+        for v in g.neighbors(u):
+            assert u in g.neighbors(int(v))
+
+
+def test_rows_sorted():
+    g, _ = planted_partition(2, 10, seed=1)
+    for u in range(g.num_nodes):
+        nb = g.neighbors(u).tolist()
+        assert nb == sorted(nb)
+        assert len(set(nb)) == len(nb)
+        assert u not in nb
+
+
+def test_parse_file(tmp_path):
+    p = tmp_path / "edges.txt"
+    p.write_text("# a comment\n# another\n1 2\n2 3\n3\t1\n")
+    arr = parse_edge_array(str(p))
+    assert arr.shape == (3, 2)
+    g = load_graph(str(p))
+    assert g.num_nodes == 3
+    assert g.num_edges == 3
+
+
+def test_rmat_properties():
+    g = rmat_graph(10, 8.0, seed=3)
+    assert g.num_nodes <= 1024
+    assert g.num_edges > 1000
+    deg = g.degrees()
+    assert deg.min() >= 1
+    # power-law-ish: max degree well above mean
+    assert deg.max() > 5 * deg.mean()
+
+
+def test_rmat_with_target_edges():
+    from bigclam.io import rmat_graph_with_edges
+
+    g = rmat_graph_with_edges(2000, 8000, seed=5)
+    assert g.num_edges == 8000
+
+
+def test_planted_partition_labels():
+    g, labels = planted_partition(3, 15, p_in=0.6, p_out=0.01, seed=2)
+    assert g.num_nodes <= 45
+    assert len(labels) == 45
