@@ -25,7 +25,7 @@ def save_shard_checkpoint(dirpath: str, trainer, sweep: int, llh: float):
     os.makedirs(dirpath, exist_ok=True)
     st = trainer.state
     r = trainer.rank
-    F = st.F_local
+    F = st.F_local_k.contiguous()
     if F.dtype == torch.bfloat16:
         arr = F.view(torch.uint16).cpu().numpy()
         dtype = "bf16"
@@ -35,7 +35,10 @@ def save_shard_checkpoint(dirpath: str, trainer, sweep: int, llh: float):
     np.save(os.path.join(dirpath, f"F_rank{r}.npy"), arr)
     if r == 0:
         np.save(os.path.join(dirpath, "raw_ids.npy"), trainer.graph.raw_ids)
-        np.save(os.path.join(dirpath, "sumF.npy"), st.sumF.cpu().numpy())
+        np.save(
+            os.path.join(dirpath, "sumF.npy"),
+            st.sumF[: trainer.cfg.k].cpu().numpy(),
+        )
         meta = {
             "n": trainer.graph.num_nodes,
             "k": trainer.cfg.k,

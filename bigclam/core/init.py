@@ -131,13 +131,22 @@ def seed_init_local_F(
 
 
 def random_init_local_F(
-    n_total: int, k: int, start: int, stop: int, rng_seed: int = 0
+    n_total: int,
+    k: int,
+    start: int,
+    stop: int,
+    rng_seed: int = 0,
+    scale: Optional[float] = None,
 ) -> np.ndarray:
-    """Uniform(0,1) random init (the BASELINE 'random-init F' configs).
+    """Uniform(0, scale) random init (the BASELINE 'random-init F' configs).
 
-    Rows are generated globally deterministic (seeded by row-block) so a
-    sharded run initializes identically to a single-shard run.
+    ``scale`` defaults to 1/sqrt(K) so initial edge dots x = Fu.Fv are O(1)
+    regardless of K (keeps exp(-x) off its clamps at t=0).  Rows are
+    generated globally deterministic (seeded by row-block) so a sharded run
+    initializes identically to a single-shard run.
     """
+    if scale is None:
+        scale = 1.0 / float(np.sqrt(k))
     out = np.empty((stop - start, k), dtype=np.float32)
     block = 65536
     b0 = start // block
@@ -148,4 +157,5 @@ def random_init_local_F(
         lo = max(start, b * block)
         hi = min(stop, b * block + rows.shape[0])
         out[lo - start : hi - start] = rows[lo - b * block : hi - b * block]
+    out *= np.float32(scale)
     return out

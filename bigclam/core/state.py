@@ -43,6 +43,10 @@ class ShardState:
         self.storage_dtype = (
             torch.bfloat16 if cfg.dtype == "bf16" else torch.float32
         )
+        # pad K to a multiple of 4: rows stay 16B-aligned for float4 kernels;
+        # pad columns are identically zero and stay zero through every op
+        # (grad_pad = -sumF_pad + F_pad = 0, clamp(0 + s*0) = 0).
+        self.kp = (cfg.k + 3) & ~3
 
         dev = self.device
         self.indptr = torch.from_numpy(shard.indptr).to(dev)
@@ -57,9 +61,9 @@ class ShardState:
 
         # F buffer: owned rows [0, n_local) + halo rows [n_local, n_rows)
         self.F = torch.zeros(
-            shard.n_rows, cfg.k, device=dev, dtype=self.storage_dtype
+            shard.n_rows, self.kp, device=dev, dtype=self.storage_dtype
         )
-        self.sumF = torch.zeros(cfg.k, device=dev, dtype=torch.float32)
+        self.sumF = torch.zeros(self.kp, device=dev, dtype=torch.float32)
 
         # halo plan tensors
         plan = shard.plan
@@ -76,6 +80,11 @@ class ShardState:
         )
         if self.use_hip:
             _hip_ops().ensure_loaded()  # fail loudly if the .so is missing
+            if self.storage_dtype == torch.bfloat16:
+                raise NotImplementedError(
+                    "bf16 F storage on GPU: HIP bf16 kernels not built yet "
+                    "(fp32 is the supported GPU dtype in this version)"
+                )
 
     # ------------------------------------------------------------------ util
     @property
@@ -84,7 +93,13 @@ class ShardState:
 
     @property
     def F_local(self) -> torch.Tensor:
+        """Owned rows, padded to kp columns (kernel view)."""
         return self.F[: self.n_local]
+
+    @property
+    def F_local_k(self) -> torch.Tensor:
+        """Owned rows, the true K columns (user/io view)."""
+        return self.F[: self.n_local, : self.cfg.k]
 
     def edge_src(self) -> torch.Tensor:
         if self._edge_src is None:
@@ -93,14 +108,16 @@ class ShardState:
 
     # ------------------------------------------------------------- model init
     def set_local_F(self, F_local: torch.Tensor):
-        """Install owned rows and (re)compute the global sumF."""
+        """Install owned rows (true-K width) and (re)compute the global sumF."""
         self.F = torch.zeros(
             self.shard.n_rows,
-            self.cfg.k,
+            self.kp,
             device=self.device,
             dtype=self.storage_dtype,
         )
-        self.F[: self.n_local] = F_local.to(self.device, self.storage_dtype)
+        self.F[: self.n_local, : self.cfg.k] = F_local.to(
+            self.device, self.storage_dtype
+        )
         self.sumF = self.F_local.float().sum(dim=0)
         comm.all_reduce_(self.sumF)
 

@@ -169,7 +169,7 @@ class Trainer:
     # -------------------------------------------------------------- gather F
     def gather_F(self) -> Optional[torch.Tensor]:
         """Gather the full F to rank 0 (CPU) for extraction/output."""
-        local = self.state.F_local.float()
+        local = self.state.F_local_k.contiguous().float()
         if self.world_size == 1:
             return local.cpu()
         import torch.distributed as dist

@@ -1,0 +1,103 @@
+// Python bindings for the BigCLAM CDNA4 kernels (bigclam._C).
+#include <torch/extension.h>
+
+#include <ATen/hip/HIPContext.h>
+#include <hip/hip_runtime.h>
+
+extern "C" void launch_k1(const float*, const long long*, const int*,
+                          const float*, const int*, float*, double*, int, int,
+                          float, float, hipStream_t);
+extern "C" void launch_k4(const float*, const long long*, const int*,
+                          const float*, const int*, double*, int, int, float,
+                          float, hipStream_t);
+extern "C" void launch_k2(const float*, const long long*, const int*,
+                          const float*, const float*, const double*,
+                          const int*, const float*, float*, int, int, int,
+                          float, float, float, float, float, hipStream_t);
+
+namespace {
+
+#define CHECK_IN(t, type)                                            \
+  TORCH_CHECK(t.is_cuda(), #t " must be on GPU");                    \
+  TORCH_CHECK(t.is_contiguous(), #t " must be contiguous");          \
+  TORCH_CHECK(t.scalar_type() == type, #t " has wrong dtype");
+
+hipStream_t current_stream() {
+  return at::hip::getCurrentHIPStream().stream();
+}
+
+void edge_grad_llh(torch::Tensor F, torch::Tensor indptr,
+                   torch::Tensor indices, torch::Tensor sumF,
+                   torch::Tensor order, torch::Tensor grad, torch::Tensor llh,
+                   double min_p, double max_p) {
+  CHECK_IN(F, torch::kFloat32);
+  CHECK_IN(indptr, torch::kInt64);
+  CHECK_IN(indices, torch::kInt32);
+  CHECK_IN(sumF, torch::kFloat32);
+  CHECK_IN(order, torch::kInt32);
+  CHECK_IN(grad, torch::kFloat32);
+  CHECK_IN(llh, torch::kFloat64);
+  const int n_local = (int)indptr.size(0) - 1;
+  const int K = (int)F.size(1);
+  TORCH_CHECK(K % 4 == 0, "K must be padded to a multiple of 4");
+  TORCH_CHECK(grad.size(0) == n_local && grad.size(1) == K);
+  launch_k1(F.data_ptr<float>(), reinterpret_cast<const long long*>(indptr.data_ptr<int64_t>()),
+            indices.data_ptr<int>(), sumF.data_ptr<float>(),
+            order.data_ptr<int>(), grad.data_ptr<float>(),
+            llh.data_ptr<double>(), n_local, K, (float)min_p, (float)max_p,
+            current_stream());
+}
+
+void llh_only(torch::Tensor F, torch::Tensor indptr, torch::Tensor indices,
+              torch::Tensor sumF, torch::Tensor order, torch::Tensor llh,
+              double min_p, double max_p) {
+  CHECK_IN(F, torch::kFloat32);
+  CHECK_IN(indptr, torch::kInt64);
+  CHECK_IN(indices, torch::kInt32);
+  CHECK_IN(sumF, torch::kFloat32);
+  CHECK_IN(order, torch::kInt32);
+  CHECK_IN(llh, torch::kFloat64);
+  const int n_local = (int)indptr.size(0) - 1;
+  const int K = (int)F.size(1);
+  TORCH_CHECK(K % 4 == 0, "K must be padded to a multiple of 4");
+  launch_k4(F.data_ptr<float>(), reinterpret_cast<const long long*>(indptr.data_ptr<int64_t>()),
+            indices.data_ptr<int>(), sumF.data_ptr<float>(),
+            order.data_ptr<int>(), llh.data_ptr<double>(), n_local, K,
+            (float)min_p, (float)max_p, current_stream());
+}
+
+void linesearch(torch::Tensor F, torch::Tensor indptr, torch::Tensor indices,
+                torch::Tensor sumF, torch::Tensor grad, torch::Tensor llh,
+                torch::Tensor order, torch::Tensor ladder, torch::Tensor best,
+                double alpha, double min_p, double max_p, double min_f,
+                double max_f) {
+  CHECK_IN(F, torch::kFloat32);
+  CHECK_IN(indptr, torch::kInt64);
+  CHECK_IN(indices, torch::kInt32);
+  CHECK_IN(sumF, torch::kFloat32);
+  CHECK_IN(grad, torch::kFloat32);
+  CHECK_IN(llh, torch::kFloat64);
+  CHECK_IN(order, torch::kInt32);
+  CHECK_IN(ladder, torch::kFloat32);
+  CHECK_IN(best, torch::kFloat32);
+  const int n_local = (int)indptr.size(0) - 1;
+  const int K = (int)F.size(1);
+  TORCH_CHECK(K % 4 == 0, "K must be padded to a multiple of 4");
+  launch_k2(F.data_ptr<float>(), reinterpret_cast<const long long*>(indptr.data_ptr<int64_t>()),
+            indices.data_ptr<int>(), sumF.data_ptr<float>(),
+            grad.data_ptr<float>(), llh.data_ptr<double>(),
+            order.data_ptr<int>(), ladder.data_ptr<float>(),
+            best.data_ptr<float>(), n_local, K, (int)ladder.size(0),
+            (float)alpha, (float)min_p, (float)max_p, (float)min_f,
+            (float)max_f, current_stream());
+}
+
+}  // namespace
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("edge_grad_llh", &edge_grad_llh,
+        "K1: fused per-node gradient + local LLH (CDNA4)");
+  m.def("llh_only", &llh_only, "K4: per-node local LLH (CDNA4)");
+  m.def("linesearch", &linesearch,
+        "K2: 16-candidate Armijo line search in one edge pass (CDNA4)");
+}

@@ -22,14 +22,14 @@ def test_checkpoint_roundtrip(tmp_path, small_graph):
     meta = load_meta(str(tmp_path))
     assert meta["n"] == g.num_nodes and meta["k"] == 3
     F = load_full_F(str(tmp_path))
-    np.testing.assert_allclose(F, tr.state.F_local.numpy(), rtol=1e-6)
+    np.testing.assert_allclose(F, tr.state.F_local_k.numpy(), rtol=1e-6)
 
     # resume into a fresh trainer and continue
     tr2 = Trainer(g, cfg, rank=0, world_size=1, device=torch.device("cpu"))
     sweep0 = resume(str(tmp_path), tr2)
     assert sweep0 == res.sweeps
     np.testing.assert_allclose(
-        tr2.state.F_local.numpy(), tr.state.F_local.numpy(), rtol=1e-6
+        tr2.state.F_local_k.numpy(), tr.state.F_local_k.numpy(), rtol=1e-6
     )
     np.testing.assert_allclose(
         tr2.state.sumF.numpy(), tr.state.sumF.numpy(), rtol=1e-5
@@ -46,5 +46,5 @@ def test_checkpoint_bf16(tmp_path, small_graph):
     save_shard_checkpoint(str(tmp_path), tr, sweep=0, llh=0.0)
     F = load_full_F(str(tmp_path))
     np.testing.assert_allclose(
-        F, tr.state.F_local.float().numpy(), rtol=1e-6
+        F, tr.state.F_local_k.float().numpy(), rtol=1e-6
     )

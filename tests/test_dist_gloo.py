@@ -31,7 +31,7 @@ def _single_run(n_sweeps=3):
     tr = Trainer(g, cfg, rank=0, world_size=1, device=torch.device("cpu"))
     tr.init_F("random")
     llh = [tr.sweep()["llh"] for _ in range(n_sweeps)]
-    return llh, tr.state.F_local.numpy().copy()
+    return llh, tr.state.F_local_k.numpy().copy()
 
 
 def _worker(rank, world_size, port, out_dir, n_sweeps):
