@@ -28,7 +28,7 @@ import torch
 from bigclam import comm
 from bigclam.config import BigClamConfig
 from bigclam.engine.trainer import Trainer
-from bigclam.io import rmat_graph_with_edges
+from bigclam.io import shaped_graph
 from bigclam.utils.metrics import MetricsLogger
 
 
@@ -49,7 +49,7 @@ def main():
     use_cuda = torch.cuda.is_available()
     device = torch.device("cuda") if use_cuda else torch.device("cpu")
 
-    graph = rmat_graph_with_edges(args.nodes, args.edges, seed=args.graph_seed)
+    graph = shaped_graph(args.nodes, args.edges, seed=args.graph_seed)
     cfg = BigClamConfig(
         k=args.k,
         dtype=args.dtype,
@@ -105,7 +105,7 @@ def main():
                     "data": "synthetic",
                     "config": {
                         "model": "bigclam",
-                        "graph": "com-Amazon-shaped R-MAT",
+                        "graph": "com-Amazon-shaped synthetic power-law (Chung-Lu)",
                         "nodes": graph.num_nodes,
                         "undirected_edges": graph.num_edges,
                         "directed_edge_visits_per_sweep": graph.num_directed_edges,

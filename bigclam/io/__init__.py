@@ -1,5 +1,11 @@
 from .edgelist import Graph, build_graph, load_graph, parse_edge_array
-from .synthetic import planted_partition, rmat_edges, rmat_graph, rmat_graph_with_edges
+from .synthetic import (
+    planted_partition,
+    rmat_edges,
+    rmat_graph,
+    rmat_graph_with_edges,
+    shaped_graph,
+)
 
 __all__ = [
     "Graph",
@@ -10,4 +16,5 @@ __all__ = [
     "rmat_edges",
     "rmat_graph",
     "rmat_graph_with_edges",
+    "shaped_graph",
 ]

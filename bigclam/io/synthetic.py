@@ -83,6 +83,72 @@ def rmat_graph_with_edges(
     return g
 
 
+def shaped_graph(n: int, e: int, alpha: float = 0.4, seed: int = 0) -> Graph:
+    """Power-law graph with EXACTLY n nodes and e undirected edges.
+
+    Chung-Lu-style: node weights w_i ∝ (i+1)^-alpha (shuffled), a base pass
+    gives every node one incident edge (no isolated nodes, so the node count
+    is exact — plain R-MAT at low edge factor touches far fewer nodes than
+    2**scale), then weighted endpoint sampling fills in the remaining edges;
+    oversampled and trimmed to exactly e after dedup.  alpha=0.4 at
+    com-Amazon's density gives a ~350 max degree, close to the real 549.
+    """
+    rng = np.random.default_rng(seed)
+    w = (np.arange(1, n + 1, dtype=np.float64)) ** (-alpha)
+    rng.shuffle(w)
+    p = w / w.sum()
+    # base: every node gets one weighted partner (guarantees presence)
+    base_src = np.arange(n, dtype=np.int64)
+    base_dst = rng.choice(n, size=n, p=p)
+    fix = base_dst == base_src
+    base_dst[fix] = (base_src[fix] + 1) % n
+    m_extra = int(1.35 * max(0, e - n))
+    src = rng.choice(n, size=m_extra, p=p)
+    dst = rng.choice(n, size=m_extra, p=p)
+    edges = np.concatenate(
+        [
+            np.stack([base_src, base_dst], axis=1),
+            np.stack([src, dst], axis=1),
+        ]
+    )
+    g = build_graph(edges)
+    if g.num_nodes != n:
+        raise RuntimeError("shaped_graph lost nodes unexpectedly")
+    if g.num_edges < e:
+        raise RuntimeError(
+            f"shaped_graph undersampled: {g.num_edges} < {e}; raise oversample"
+        )
+    # trim extra undirected edges at random, never touching the base pass
+    # (trim only edges whose both endpoints keep degree >= 2)
+    srcs = np.repeat(np.arange(n, dtype=np.int64), g.degrees())
+    lo_mask = srcs < g.indices
+    und_src = srcs[lo_mask]
+    und_dst = g.indices[lo_mask].astype(np.int64)
+    n_trim = g.num_edges - e
+    if n_trim > 0:
+        deg = g.degrees().copy()
+        order = rng.permutation(len(und_src))
+        keep = np.ones(len(und_src), dtype=bool)
+        removed = 0
+        for idx in order:
+            if removed == n_trim:
+                break
+            a, b = und_src[idx], und_dst[idx]
+            if deg[a] > 1 and deg[b] > 1:
+                keep[idx] = False
+                deg[a] -= 1
+                deg[b] -= 1
+                removed += 1
+        if removed < n_trim:
+            raise RuntimeError("shaped_graph: could not trim to target")
+        g = build_graph(
+            np.stack([und_src[keep], und_dst[keep]], axis=1)
+        )
+        if g.num_nodes != n or g.num_edges != e:
+            raise RuntimeError("shaped_graph trim broke the target shape")
+    return g
+
+
 def planted_partition(
     num_communities: int,
     nodes_per_community: int,

@@ -1,0 +1,105 @@
+"""CLI end-to-end tests (CPU): fit -> communities + checkpoint -> extract."""
+import json
+import os
+import subprocess
+import sys
+
+import numpy as np
+import pytest
+
+from bigclam.io import planted_partition
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+
+def _write_edgelist(path, g):
+    with open(path, "w") as f:
+        f.write("# test graph\n")
+        for u in range(g.num_nodes):
+            for v in g.neighbors(u):
+                if u < v:
+                    f.write(f"{g.raw_ids[u]} {g.raw_ids[v]}\n")
+
+
+def _run(args):
+    return subprocess.run(
+        [sys.executable, "-m", "bigclam"] + args,
+        cwd=REPO,
+        capture_output=True,
+        text=True,
+        timeout=300,
+    )
+
+
+def test_cli_fit_and_extract(tmp_path):
+    g, _ = planted_partition(3, 12, p_in=0.5, p_out=0.02, seed=21)
+    edges = tmp_path / "g.txt"
+    _write_edgelist(edges, g)
+    out = tmp_path / "coms.txt"
+    ckpt = tmp_path / "ckpt"
+    metrics = tmp_path / "metrics.jsonl"
+    r = _run(
+        [
+            "fit",
+            str(edges),
+            "--k",
+            "3",
+            "--device",
+            "cpu",
+            "--max-sweeps",
+            "10",
+            "--init",
+            "random",
+            "--out",
+            str(out),
+            "--checkpoint-dir",
+            str(ckpt),
+            "--metrics",
+            str(metrics),
+            "--quiet",
+        ]
+    )
+    assert r.returncode == 0, r.stderr
+    res = json.loads(r.stdout.strip().splitlines()[-1])
+    assert res["k"] == 3 and np.isfinite(res["llh"])
+    assert out.exists() and ckpt.exists()
+    # metrics JSONL has per-sweep records
+    lines = [json.loads(l) for l in open(metrics)]
+    assert any("llh" in l for l in lines)
+    # community lines parse: "<cid>: id id id"
+    for line in open(out):
+        cid, rest = line.split(":", 1)
+        int(cid)
+        assert all(tok.isdigit() for tok in rest.split())
+
+    # extract from the checkpoint reproduces a valid output
+    out2 = tmp_path / "coms2.txt"
+    r2 = _run(["extract", str(ckpt), str(edges), "--out", str(out2)])
+    assert r2.returncode == 0, r2.stderr
+    assert out2.read_text() == out.read_text()
+
+
+def test_cli_select_k(tmp_path):
+    g, _ = planted_partition(3, 10, p_in=0.6, p_out=0.02, seed=22)
+    edges = tmp_path / "g.txt"
+    _write_edgelist(edges, g)
+    r = _run(
+        [
+            "select-k",
+            str(edges),
+            "--device",
+            "cpu",
+            "--max-sweeps",
+            "8",
+            "--k-min",
+            "2",
+            "--k-max",
+            "6",
+            "--k-div",
+            "3",
+            "--quiet",
+        ]
+    )
+    assert r.returncode == 0, r.stderr
+    res = json.loads(r.stdout.strip().splitlines()[-1])
+    assert "grid" in res and res["grid"][0] == 2

@@ -61,3 +61,13 @@ def test_planted_partition_labels():
     g, labels = planted_partition(3, 15, p_in=0.6, p_out=0.01, seed=2)
     assert g.num_nodes <= 45
     assert len(labels) == 45
+
+
+def test_shaped_graph_exact_counts():
+    from bigclam.io import shaped_graph
+
+    g = shaped_graph(5000, 14000, seed=9)
+    assert g.num_nodes == 5000
+    assert g.num_edges == 14000
+    assert g.degrees().min() >= 1
+    assert g.degrees().max() > 20  # heavy tail present
