@@ -61,7 +61,17 @@ class Graph:
 
 
 def parse_edge_array(path: str) -> np.ndarray:
-    """Read an edge-list file into an int64 [M, 2] array (fast path: pandas)."""
+    """Read an edge-list file into an int64 [M, 2] array.
+
+    Fast path: the native mmap'd multithreaded parser (bigclam._io_native);
+    falls back to pandas' C engine, then to pure python.
+    """
+    try:
+        from .. import _io_native
+
+        return np.asarray(_io_native.parse_edgelist(path))
+    except ImportError:
+        pass
     try:
         import pandas as pd
 

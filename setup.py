@@ -7,7 +7,8 @@ import os
 
 os.environ.setdefault("PYTORCH_ROCM_ARCH", "gfx950")
 
-from setuptools import setup
+import pybind11
+from setuptools import Extension, setup
 from torch.utils.cpp_extension import BuildExtension, CUDAExtension
 
 setup(
@@ -15,6 +16,13 @@ setup(
     version="0.1.0",
     packages=["bigclam"],
     ext_modules=[
+        Extension(
+            name="bigclam._io_native",
+            sources=["bigclam/kernels/io_native.cpp"],
+            include_dirs=[pybind11.get_include()],
+            extra_compile_args=["-O3", "-std=c++17"],
+            language="c++",
+        ),
         CUDAExtension(
             name="bigclam._C",
             sources=[
