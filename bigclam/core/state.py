@@ -171,13 +171,18 @@ class ShardState:
         )
 
     def apply_step(self, grad: torch.Tensor, steps: torch.Tensor) -> torch.Tensor:
-        """K3 commit: in-place F update; returns the GLOBAL sumF delta
-        (all-reduced, C12) and refreshes sumF."""
-        F_new, delta = ref_ops.apply_step(self.F_local, grad, steps, self.cfg)
-        self.F[: self.n_local] = F_new
-        comm.all_reduce_(delta)
-        self.sumF += delta
-        return delta
+        """K3 commit: in-place projected F update, then refresh sumF by
+        recomputing the exact column sums + all-reduce (C12).  Recomputing
+        (one read pass) keeps the sumF == colsum(F) invariant exact instead
+        of accumulating incremental deltas."""
+        if self.use_hip:
+            _hip_ops().apply_step(self.F_local, grad, steps, self.cfg)
+        else:
+            F_new, _ = ref_ops.apply_step(self.F_local, grad, steps, self.cfg)
+            self.F[: self.n_local] = F_new
+        self.sumF = self.F_local.float().sum(dim=0)
+        comm.all_reduce_(self.sumF)
+        return self.sumF
 
     def full_llh(self) -> torch.Tensor:
         """K4 + C14: global objective (fp64 scalar, all-reduced)."""

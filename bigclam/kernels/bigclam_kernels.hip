@@ -1,27 +1,32 @@
-// BigCLAM CDNA4 (gfx950 / MI355X) HIP kernels.
+// BigCLAM CDNA4 (gfx950 / MI355X) HIP kernels — v2.
 //
-// Hand-written for the CDNA4 execution model: 64-wide wavefronts, LDS-staged
-// per-node gradient accumulators, float4-vectorized HBM access, wave-shuffle
-// reductions, fp64 LLH accumulation.  Kernel worklist per SURVEY.md §2-B:
+// Hand-written for the CDNA4 execution model: 64-wide wavefronts, float4
+// HBM access sized for L1/L2 locality, LDS accumulators without atomics,
+// wave/blocked shuffle reductions, fp64 LLH accumulation.  Kernel worklist
+// per SURVEY.md §2-B:
 //   K1 edge_grad_llh  — fused per-node gradient + local LLH
 //                       (replaces codes/bigclamv3-7.scala:138-150)
 //   K2 linesearch     — all 16 Armijo candidates per node in one edge pass
 //                       (replaces the cartesian fan-out, scala:153-163)
+//   K3 apply_step     — projected commit (scala:89-92, 171)
 //   K4 llh_only       — per-node local LLH (scala:106-120 / 177-200)
 //
-// Conventions:
-//  * F is fp32 row-major [n_rows, K] with K padded to a multiple of 4 (rows
-//    16B-aligned -> clean float4 coalescing).  Pad columns are identically
-//    zero and stay zero through every op (grad_pad = -sumF_pad + F_pad = 0).
-//  * one 256-thread workgroup (4 waves) per LOCAL node; the launch walks
-//    nodes in degree-descending `order` so hub blocks start first.
-//  * wave-per-edge: each wave owns one neighbor at a time -> no
-//    __syncthreads in the edge loop; the K-dim dot is a lane-strided
-//    float4 loop + shfl_xor butterfly (all 64 lanes end with the sum).
-//  * per-node LLH is accumulated in fp64 (the convergence test is a 1e-4
-//    relative change on a ~1e8-magnitude sum).
+// v2 design notes (from profiles/r01_sweep_v1_stats.md):
+//  * K1 is BLOCK-per-edge: the whole 256-thread workgroup walks one
+//    neighbor at a time — fv is read once from HBM for the dot and re-read
+//    L1-hot for the owned-k (no-atomic) accumulate into the LDS gradient
+//    row.  v1's wave-per-edge axpy with LDS atomicAdd was 6x slower than
+//    the same traffic in K4.
+//  * K2 keeps Fu and grad_u row chunks RESIDENT IN REGISTERS (NSLOT
+//    float4 slots per thread, block-stride 1024 elements) so the edge loop
+//    touches only fv from memory; candidate rows are recomputed in-register
+//    (3 VALU per element per candidate) — v1 streamed fu+g+fv (60 KB) per
+//    edge through a 32 KB L1.  The 17 per-node block-reduced node-term
+//    passes are now ONE pass with a 256x16 LDS transpose reduction.
+//  * LLH is accumulated in fp64 (the convergence test is a 1e-4 relative
+//    change on a large-magnitude sum).
 //
-// NOTE: math must match tests/oracle.py bit-for-bit in structure:
+// Math contract (must match tests/oracle.py):
 //   x    = Fu . Fv
 //   p    = clamp(exp(-x), MIN_P, MAX_P)
 //   llh += log1p(-p) + x           ;  grad_acc += Fv / (1-p)
@@ -39,20 +44,12 @@
 // ---------------------------------------------------------------- reductions
 
 __device__ __forceinline__ float wave_allreduce_sum(float v) {
-  // butterfly: every lane ends with the full 64-lane sum
 #pragma unroll
   for (int off = 32; off > 0; off >>= 1) v += __shfl_xor(v, off, WAVE);
   return v;
 }
 
-__device__ __forceinline__ double wave_allreduce_sum(double v) {
-#pragma unroll
-  for (int off = 32; off > 0; off >>= 1) v += __shfl_xor(v, off, WAVE);
-  return v;
-}
-
-// block-wide sum of one float per thread; `red` is 2*NWAVE floats of LDS.
-// Result valid on every thread.  Costs 2 __syncthreads.
+// block-wide sum; `red` is NWAVE floats of LDS; result on every thread.
 __device__ __forceinline__ float block_allreduce_sum(float v, float* red) {
   const int lane = threadIdx.x & (WAVE - 1);
   const int wid = threadIdx.x >> 6;
@@ -80,7 +77,17 @@ __device__ __forceinline__ float dot4(float4 a, float4 b, float acc) {
   return fmaf(a.w, b.w, acc);
 }
 
+__device__ __forceinline__ float4 ld4(const float* p) {
+  return *reinterpret_cast<const float4*>(p);
+}
+
 // ------------------------------------------------------------------- K1
+//
+// One 256-thread block per LOCAL node (launch order = degree-descending so
+// hub blocks start first).  LDS: gradient accumulator gacc[K] (owned-k,
+// no atomics) + reduction scratch.  Per edge the block does a cooperative
+// dot (block_allreduce) and an owned-k accumulate; fv's second read hits L1
+// (the block touches one 4*K-byte row at a time).
 
 extern "C" __global__ void __launch_bounds__(BLOCK) k1_grad_llh(
     const float* __restrict__ F, const long long* __restrict__ indptr,
@@ -91,77 +98,68 @@ extern "C" __global__ void __launch_bounds__(BLOCK) k1_grad_llh(
   const long long e0 = indptr[u];
   const long long e1 = indptr[u + 1];
   const int tid = threadIdx.x;
-  const int lane = tid & (WAVE - 1);
-  const int wid = tid >> 6;
 
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  float* gacc = reinterpret_cast<float*>(smem);               // K floats
-  double* wllh = reinterpret_cast<double*>(smem + (size_t)K * 4);  // NWAVE
-  float* red = reinterpret_cast<float*>(wllh + NWAVE);        // 2*NWAVE
+  float* gacc = reinterpret_cast<float*>(smem);            // K floats
+  float* red = reinterpret_cast<float*>(smem + (size_t)K * 4);  // NWAVE
 
-  for (int k = tid; k < K; k += BLOCK) gacc[k] = 0.f;
+  for (int k = tid * 4; k < K; k += BLOCK * 4) {
+    *reinterpret_cast<float4*>(gacc + k) = float4{0.f, 0.f, 0.f, 0.f};
+  }
   __syncthreads();
 
   const float* __restrict__ fu = F + (size_t)u * K;
-  double llh_w = 0.0;
+  double llh_acc = 0.0;  // accumulated on thread 0 only
 
-  for (long long e = e0 + wid; e < e1; e += NWAVE) {
+  for (long long e = e0; e < e1; ++e) {
     const float* __restrict__ fv = F + (size_t)indices[e] * K;
     float part = 0.f;
-    for (int k = lane * 4; k < K; k += WAVE * 4) {
-      part = dot4(*reinterpret_cast<const float4*>(fu + k),
-                  *reinterpret_cast<const float4*>(fv + k), part);
+    for (int k = tid * 4; k < K; k += BLOCK * 4) {
+      part = dot4(ld4(fu + k), ld4(fv + k), part);
     }
-    const float x = wave_allreduce_sum(part);
+    const float x = block_allreduce_sum(part, red);
     const float p = clamp_p(__expf(-x), min_p, max_p);
     const float w = 1.f / (1.f - p);
-    if (lane == 0) llh_w += (double)log1pf(-p) + (double)x;
-    // axpy into the shared accumulator (other waves write other neighbors
-    // concurrently -> LDS float atomics; fv is L1-hot from the dot pass)
-    for (int k = lane * 4; k < K; k += WAVE * 4) {
-      const float4 b = *reinterpret_cast<const float4*>(fv + k);
-      atomicAdd(&gacc[k + 0], w * b.x);
-      atomicAdd(&gacc[k + 1], w * b.y);
-      atomicAdd(&gacc[k + 2], w * b.z);
-      atomicAdd(&gacc[k + 3], w * b.w);
+    if (tid == 0) llh_acc += (double)log1pf(-p) + (double)x;
+    for (int k = tid * 4; k < K; k += BLOCK * 4) {
+      const float4 b = ld4(fv + k);  // L1-hot: just read by this block
+      float4 g = ld4(gacc + k);
+      g.x = fmaf(w, b.x, g.x);
+      g.y = fmaf(w, b.y, g.y);
+      g.z = fmaf(w, b.z, g.z);
+      g.w = fmaf(w, b.w, g.w);
+      *reinterpret_cast<float4*>(gacc + k) = g;
     }
+    // no barrier needed: each thread owns its gacc elements; the next
+    // edge's block_allreduce syncs before x is consumed.
   }
-  if (lane == 0) wllh[wid] = llh_w;
-  __syncthreads();
 
-  // node terms: -Fu.sumF + Fu.Fu (block-cooperative dots)
+  // node terms: -Fu.sumF + Fu.Fu
   float p_fs = 0.f, p_ff = 0.f;
   for (int k = tid * 4; k < K; k += BLOCK * 4) {
-    const float4 a = *reinterpret_cast<const float4*>(fu + k);
-    const float4 s = *reinterpret_cast<const float4*>(sumF + k);
+    const float4 a = ld4(fu + k);
+    const float4 s = ld4(sumF + k);
     p_fs = dot4(a, s, p_fs);
     p_ff = dot4(a, a, p_ff);
   }
   const float fs = block_allreduce_sum(p_fs, red);
   const float ff = block_allreduce_sum(p_ff, red);
 
-  // write grad = gacc - sumF + fu
   float* __restrict__ gout = grad + (size_t)u * K;
   for (int k = tid * 4; k < K; k += BLOCK * 4) {
-    const float4 g = *reinterpret_cast<const float4*>(&gacc[k]);
-    const float4 s = *reinterpret_cast<const float4*>(sumF + k);
-    const float4 a = *reinterpret_cast<const float4*>(fu + k);
-    float4 o;
-    o.x = g.x - s.x + a.x;
-    o.y = g.y - s.y + a.y;
-    o.z = g.z - s.z + a.z;
-    o.w = g.w - s.w + a.w;
-    *reinterpret_cast<float4*>(gout + k) = o;
+    const float4 g = ld4(gacc + k);
+    const float4 s = ld4(sumF + k);
+    const float4 a = ld4(fu + k);
+    *reinterpret_cast<float4*>(gout + k) =
+        float4{g.x - s.x + a.x, g.y - s.y + a.y, g.z - s.z + a.z,
+               g.w - s.w + a.w};
   }
-  if (tid == 0) {
-    double t = 0.0;
-#pragma unroll
-    for (int wv = 0; wv < NWAVE; ++wv) t += wllh[wv];
-    llh[u] = t + (double)(-fs) + (double)ff;
-  }
+  if (tid == 0) llh[u] = llh_acc + (double)(-fs) + (double)ff;
 }
 
 // ------------------------------------------------------------------- K4
+//
+// Wave-per-edge dots (validated at ~5.8 TB/s effective in the v1 profile).
 
 extern "C" __global__ void __launch_bounds__(BLOCK) k4_llh_only(
     const float* __restrict__ F, const long long* __restrict__ indptr,
@@ -176,7 +174,7 @@ extern "C" __global__ void __launch_bounds__(BLOCK) k4_llh_only(
   const int wid = tid >> 6;
 
   __shared__ __attribute__((aligned(16))) double wllh[NWAVE];
-  __shared__ __attribute__((aligned(16))) float red[2 * NWAVE];
+  __shared__ __attribute__((aligned(16))) float red[NWAVE];
 
   const float* __restrict__ fu = F + (size_t)u * K;
   double llh_w = 0.0;
@@ -184,8 +182,7 @@ extern "C" __global__ void __launch_bounds__(BLOCK) k4_llh_only(
     const float* __restrict__ fv = F + (size_t)indices[e] * K;
     float part = 0.f;
     for (int k = lane * 4; k < K; k += WAVE * 4) {
-      part = dot4(*reinterpret_cast<const float4*>(fu + k),
-                  *reinterpret_cast<const float4*>(fv + k), part);
+      part = dot4(ld4(fu + k), ld4(fv + k), part);
     }
     const float x = wave_allreduce_sum(part);
     const float p = clamp_p(__expf(-x), min_p, max_p);
@@ -196,8 +193,8 @@ extern "C" __global__ void __launch_bounds__(BLOCK) k4_llh_only(
 
   float p_fs = 0.f, p_ff = 0.f;
   for (int k = tid * 4; k < K; k += BLOCK * 4) {
-    const float4 a = *reinterpret_cast<const float4*>(fu + k);
-    const float4 s = *reinterpret_cast<const float4*>(sumF + k);
+    const float4 a = ld4(fu + k);
+    const float4 s = ld4(sumF + k);
     p_fs = dot4(a, s, p_fs);
     p_ff = dot4(a, a, p_ff);
   }
@@ -213,12 +210,19 @@ extern "C" __global__ void __launch_bounds__(BLOCK) k4_llh_only(
 
 // ------------------------------------------------------------------- K2
 //
-// All candidate steps evaluated in ONE pass over u's neighbors: each lane
-// keeps MAX_LS partial dot accumulators; the candidate row
-// cand_j = clamp(Fu + s_j*grad, MIN_F, MAX_F) is recomputed per float4 from
-// the L1-hot Fu/grad rows (16 fma+clamp per loaded Fv float4), so Fv loads
-// amortize over the whole ladder — the reference pays 16 full passes
-// (SURVEY.md §2.8).
+// Block-per-node; K is processed in 512-element chunks.  Per (edge-tile,
+// chunk) the 16 candidate rows cand_j = clamp(Fu + s_j*grad, MIN_F, MAX_F)
+// are built ONCE into LDS (32 KB) and shared by the tile's edges, so the
+// edge loop touches only fv from global memory.  Lane mapping inside a
+// wave: lane = sub*16 + j (sub in 0..3, j = candidate) — each lane owns
+// ONE candidate and a quarter of the k-range, so per-edge dot partials
+// accumulate in a register across all chunks with NO per-chunk cross-lane
+// reduction; one 2-step shfl_xor folds the 4 sub-lanes when the edge
+// completes.  Handles any K (the v2 register-resident variant hit the
+// hipcc candidate-hoisting cliff: 64*NSLOT VGPRs of hoisted cand rows).
+
+#define K2_CHUNK 512
+#define K2_TILE 32  // edges per tile; a wave owns K2_TILE/NWAVE = 8 of them
 
 extern "C" __global__ void __launch_bounds__(BLOCK) k2_linesearch(
     const float* __restrict__ F, const long long* __restrict__ indptr,
@@ -233,115 +237,226 @@ extern "C" __global__ void __launch_bounds__(BLOCK) k2_linesearch(
   const int tid = threadIdx.x;
   const int lane = tid & (WAVE - 1);
   const int wid = tid >> 6;
+  const int sub = lane >> 4;   // 0..3: k-quarter inside the wave
+  const int j = lane & 15;     // candidate index this lane owns
 
-  __shared__ __attribute__((aligned(16))) double wacc[NWAVE][MAX_LS];
-  __shared__ __attribute__((aligned(16))) float red[2 * NWAVE];
-  __shared__ __attribute__((aligned(16))) float s_ladder[MAX_LS];
+  // +4 pad: row stride 2064 B puts each j on its own bank quad
+  __shared__ __attribute__((aligned(16))) float cand[MAX_LS][K2_CHUNK + 4];
+  __shared__ __attribute__((aligned(16))) float s_lad[MAX_LS];
+  __shared__ __attribute__((aligned(16))) double acc_llh[NWAVE][MAX_LS];
+  __shared__ __attribute__((aligned(16))) float acc_nt[NWAVE][MAX_LS];
+  __shared__ __attribute__((aligned(16))) float red[NWAVE];
 
-  if (tid < MAX_LS) s_ladder[tid] = (tid < n_ladder) ? ladder[tid] : 0.f;
-  if (lane == 0) {
-#pragma unroll
-    for (int j = 0; j < MAX_LS; ++j) wacc[wid][j] = 0.0;
-  }
+  if (tid < MAX_LS) s_lad[tid] = (tid < n_ladder) ? ladder[tid] : 0.f;
   __syncthreads();
+  const float sj = s_lad[j];  // this lane's candidate step
 
   const float* __restrict__ fu = F + (size_t)u * K;
   const float* __restrict__ gu = grad + (size_t)u * K;
 
-  float s0, s1_, s2, s3, s4, s5, s6, s7, s8, s9, s10, s11, s12, s13, s14, s15;
-  s0 = s_ladder[0]; s1_ = s_ladder[1]; s2 = s_ladder[2]; s3 = s_ladder[3];
-  s4 = s_ladder[4]; s5 = s_ladder[5]; s6 = s_ladder[6]; s7 = s_ladder[7];
-  s8 = s_ladder[8]; s9 = s_ladder[9]; s10 = s_ladder[10]; s11 = s_ladder[11];
-  s12 = s_ladder[12]; s13 = s_ladder[13]; s14 = s_ladder[14]; s15 = s_ladder[15];
+  double llh_j = 0.0;  // edge-term sum for candidate j (lanes with sub==0)
+  float nt_j = 0.f;    // node term cand_j.(Fu - sumF), partial per lane
+  float p_gg = 0.f;    // grad.grad partial (thread-strided, j-independent)
 
-  double lacc[MAX_LS];
+  for (long long t0 = e0; t0 < e1 || t0 == e0; t0 += K2_TILE) {
+    const int tlen = (int)((e1 - t0) < K2_TILE ? (e1 - t0) : K2_TILE);
+    float acc[K2_TILE / NWAVE];
 #pragma unroll
-  for (int j = 0; j < MAX_LS; ++j) lacc[j] = 0.0;
+    for (int i = 0; i < K2_TILE / NWAVE; ++i) acc[i] = 0.f;
+    const bool first_tile = (t0 == e0);
 
-  for (long long e = e0 + wid; e < e1; e += NWAVE) {
-    const float* __restrict__ fv = F + (size_t)indices[e] * K;
-    float acc[MAX_LS];
+    for (int c0 = 0; c0 < K; c0 += K2_CHUNK) {
+      const int clen = (K - c0) < K2_CHUNK ? (K - c0) : K2_CHUNK;
+      __syncthreads();  // previous chunk fully consumed
+      // build the candidate chunk (fu/g chunk stays L1-hot across the 16 j)
+      for (int jj = 0; jj < MAX_LS; ++jj) {
+        const float sjj = s_lad[jj];
+        for (int kk = tid * 4; kk < clen; kk += BLOCK * 4) {
+          const float4 a = ld4(fu + c0 + kk);
+          const float4 g = ld4(gu + c0 + kk);
+          *reinterpret_cast<float4*>(&cand[jj][kk]) =
+              float4{fminf(fmaxf(fmaf(sjj, g.x, a.x), min_f), max_f),
+                     fminf(fmaxf(fmaf(sjj, g.y, a.y), min_f), max_f),
+                     fminf(fmaxf(fmaf(sjj, g.z, a.z), min_f), max_f),
+                     fminf(fmaxf(fmaf(sjj, g.w, a.w), min_f), max_f)};
+        }
+      }
+      __syncthreads();
+
+      // edges of this tile over this chunk; acc[i] persists across chunks
+      // (full unroll keeps acc[] in registers — runtime indexing would
+      // spill it to scratch)
 #pragma unroll
-    for (int j = 0; j < MAX_LS; ++j) acc[j] = 0.f;
-    for (int k = lane * 4; k < K; k += WAVE * 4) {
-      const float4 a = *reinterpret_cast<const float4*>(fu + k);
-      const float4 g = *reinterpret_cast<const float4*>(gu + k);
-      const float4 b = *reinterpret_cast<const float4*>(fv + k);
-#define LS_STEP(J, S)                                                       \
-  {                                                                         \
-    float4 c;                                                               \
-    c.x = fminf(fmaxf(fmaf(S, g.x, a.x), min_f), max_f);                    \
-    c.y = fminf(fmaxf(fmaf(S, g.y, a.y), min_f), max_f);                    \
-    c.z = fminf(fmaxf(fmaf(S, g.z, a.z), min_f), max_f);                    \
-    c.w = fminf(fmaxf(fmaf(S, g.w, a.w), min_f), max_f);                    \
-    acc[J] = dot4(c, b, acc[J]);                                            \
-  }
-      LS_STEP(0, s0) LS_STEP(1, s1_) LS_STEP(2, s2) LS_STEP(3, s3)
-      LS_STEP(4, s4) LS_STEP(5, s5) LS_STEP(6, s6) LS_STEP(7, s7)
-      LS_STEP(8, s8) LS_STEP(9, s9) LS_STEP(10, s10) LS_STEP(11, s11)
-      LS_STEP(12, s12) LS_STEP(13, s13) LS_STEP(14, s14) LS_STEP(15, s15)
-#undef LS_STEP
-    }
-#pragma unroll
-    for (int j = 0; j < MAX_LS; ++j) {
-      const float x = wave_allreduce_sum(acc[j]);
-      if (lane == 0) {
-        const float p = clamp_p(__expf(-x), min_p, max_p);
-        lacc[j] += (double)log1pf(-p) + (double)x;
+      for (int i = 0; i < K2_TILE / NWAVE; ++i) {
+        const long long e = t0 + (long long)(i * NWAVE + wid);
+        if (e < t0 + tlen) {
+          const float* __restrict__ fv =
+              F + (size_t)indices[e] * K + c0;
+          float a = acc[i];
+#pragma clang loop unroll_count(4)
+          for (int m = sub; m * 4 < clen; m += 4) {
+            a = dot4(ld4(&cand[j][m * 4]), ld4(fv + m * 4), a);
+          }
+          acc[i] = a;
+        }
+      }
+
+      // node term for this chunk: waves split the slot range; once per node
+      if (first_tile) {
+        const int m0 = sub + 4 * wid;  // 16 (sub, wave) ways cover slots mod 16
+        for (int m = m0; m * 4 < clen; m += 16) {
+          const float4 c = ld4(&cand[j][m * 4]);
+          const float4 a = ld4(fu + c0 + m * 4);
+          const float4 sf = ld4(sumF + c0 + m * 4);
+          nt_j = fmaf(c.x, a.x - sf.x, nt_j);
+          nt_j = fmaf(c.y, a.y - sf.y, nt_j);
+          nt_j = fmaf(c.z, a.z - sf.z, nt_j);
+          nt_j = fmaf(c.w, a.w - sf.w, nt_j);
+        }
+      }
+      if (first_tile) {
+        for (int kk = tid * 4; kk < clen; kk += BLOCK * 4) {
+          const float4 g = ld4(gu + c0 + kk);
+          p_gg = dot4(g, g, p_gg);
+        }
       }
     }
-  }
-  if (lane == 0) {
-#pragma unroll
-    for (int j = 0; j < MAX_LS; ++j) wacc[wid][j] = lacc[j];
-  }
-  __syncthreads();
 
-  // per-candidate node terms: cand_j.(Fu - sumF)   (identity: the
-  // -cand.sumF' + cand.cand terms collapse; see header), plus gg = grad.grad.
-  float p_gg = 0.f;
-  for (int k = tid * 4; k < K; k += BLOCK * 4) {
-    const float4 g = *reinterpret_cast<const float4*>(gu + k);
-    p_gg = dot4(g, g, p_gg);
+    // fold the 4 sub-lanes of each edge; lane sub==0 owns the result
+#pragma unroll
+    for (int i = 0; i < K2_TILE / NWAVE; ++i) {
+      float x = acc[i];
+      x += __shfl_xor(x, 16, WAVE);
+      x += __shfl_xor(x, 32, WAVE);
+      const long long e = t0 + (long long)(i * NWAVE + wid);
+      if (sub == 0 && e < t0 + tlen) {
+        const float p = clamp_p(__expf(-x), min_p, max_p);
+        llh_j += (double)log1pf(-p) + (double)x;
+      }
+    }
+    if (e1 == e0) break;  // degree-0 guard (loop ran once for node terms)
+  }
+
+  // fold node-term sub-lanes, stash per-wave results, combine across waves
+  nt_j += __shfl_xor(nt_j, 16, WAVE);
+  nt_j += __shfl_xor(nt_j, 32, WAVE);
+  if (sub == 0) {
+    acc_llh[wid][j] = llh_j;
+    acc_nt[wid][j] = nt_j;
   }
   const float gg = block_allreduce_sum(p_gg, red);
-
-  __shared__ double node_term[MAX_LS];
-  for (int j = 0; j < n_ladder; ++j) {
-    const float sj = s_ladder[j];
-    float part = 0.f;
-    for (int k = tid * 4; k < K; k += BLOCK * 4) {
-      const float4 a = *reinterpret_cast<const float4*>(fu + k);
-      const float4 g = *reinterpret_cast<const float4*>(gu + k);
-      const float4 s = *reinterpret_cast<const float4*>(sumF + k);
-      float4 c;
-      c.x = fminf(fmaxf(fmaf(sj, g.x, a.x), min_f), max_f);
-      c.y = fminf(fmaxf(fmaf(sj, g.y, a.y), min_f), max_f);
-      c.z = fminf(fmaxf(fmaf(sj, g.z, a.z), min_f), max_f);
-      c.w = fminf(fmaxf(fmaf(sj, g.w, a.w), min_f), max_f);
-      part = fmaf(c.x, a.x - s.x, part);
-      part = fmaf(c.y, a.y - s.y, part);
-      part = fmaf(c.z, a.z - s.z, part);
-      part = fmaf(c.w, a.w - s.w, part);
-    }
-    const float nt = block_allreduce_sum(part, red);
-    if (tid == 0) node_term[j] = (double)nt;
-  }
-  __syncthreads();
 
   if (tid == 0) {
     const double llh_u = llh[u];
     float chosen = 0.f;
-    for (int j = 0; j < n_ladder; ++j) {  // descending ladder: first accept
-      double trial = node_term[j];
+    for (int jj = 0; jj < n_ladder; ++jj) {  // descending: first accept wins
+      double trial = 0.0;
 #pragma unroll
-      for (int wv = 0; wv < NWAVE; ++wv) trial += wacc[wv][j];
-      const float sj = s_ladder[j];
-      if (trial >= llh_u + (double)(alpha * sj * gg)) {
-        chosen = sj;
+      for (int wv = 0; wv < NWAVE; ++wv)
+        trial += acc_llh[wv][jj] + (double)acc_nt[wv][jj];
+      const float sjj = s_lad[jj];
+      if (trial >= llh_u + (double)(alpha * sjj * gg)) {
+        chosen = sjj;
         break;
       }
     }
     best[u] = chosen;
   }
+}
+
+// ------------------------------------------------------------------- K3
+//
+// Projected commit: F_u <- clamp(F_u + s_u*grad_u, MIN_F, MAX_F) for nodes
+// with an accepted step.  One block per node; rows with s==0 return without
+// touching memory.  sumF is recomputed afterwards (column sum + allreduce)
+// which keeps the sumF == colsum(F) invariant exact.
+
+extern "C" __global__ void __launch_bounds__(BLOCK) k3_apply_step(
+    float* __restrict__ F, const float* __restrict__ grad,
+    const float* __restrict__ steps, int n_local, int K, float min_f,
+    float max_f) {
+  const int u = blockIdx.x;
+  const float s = steps[u];
+  if (s <= 0.f) return;
+  float* __restrict__ fu = F + (size_t)u * K;
+  const float* __restrict__ gu = grad + (size_t)u * K;
+  for (int k = threadIdx.x * 4; k < K; k += BLOCK * 4) {
+    const float4 a = ld4(fu + k);
+    const float4 g = ld4(gu + k);
+    *reinterpret_cast<float4*>(fu + k) =
+        float4{fminf(fmaxf(fmaf(s, g.x, a.x), min_f), max_f),
+               fminf(fmaxf(fmaf(s, g.y, a.y), min_f), max_f),
+               fminf(fmaxf(fmaf(s, g.z, a.z), min_f), max_f),
+               fminf(fmaxf(fmaf(s, g.w, a.w), min_f), max_f)};
+  }
+}
+
+// ----------------------------------------------------------- host launchers
+
+#include <stdexcept>
+#include <string>
+
+#define HIP_CHECK(expr)                                                    \
+  do {                                                                     \
+    hipError_t _e = (expr);                                                \
+    if (_e != hipSuccess) {                                                \
+      throw std::runtime_error(std::string("HIP error: ") +                \
+                               hipGetErrorString(_e) + " at " __FILE__);   \
+    }                                                                      \
+  } while (0)
+
+static void allow_large_lds(const void* func, size_t bytes) {
+  if (bytes > 65536) {
+    HIP_CHECK(hipFuncSetAttribute(
+        func, hipFuncAttributeMaxDynamicSharedMemorySize, (int)bytes));
+  }
+}
+
+extern "C" void launch_k1(const float* F, const long long* indptr,
+                          const int* indices, const float* sumF,
+                          const int* order, float* grad, double* llh,
+                          int n_local, int K, float min_p, float max_p,
+                          hipStream_t stream) {
+  if (n_local == 0) return;
+  const size_t lds = (size_t)K * 4 + 4 * sizeof(float);
+  allow_large_lds((const void*)k1_grad_llh, lds);
+  hipLaunchKernelGGL(k1_grad_llh, dim3(n_local), dim3(256), lds, stream, F,
+                     indptr, indices, sumF, order, grad, llh, n_local, K,
+                     min_p, max_p);
+  HIP_CHECK(hipGetLastError());
+}
+
+extern "C" void launch_k4(const float* F, const long long* indptr,
+                          const int* indices, const float* sumF,
+                          const int* order, double* llh, int n_local, int K,
+                          float min_p, float max_p, hipStream_t stream) {
+  if (n_local == 0) return;
+  hipLaunchKernelGGL(k4_llh_only, dim3(n_local), dim3(256), 0, stream, F,
+                     indptr, indices, sumF, order, llh, n_local, K, min_p,
+                     max_p);
+  HIP_CHECK(hipGetLastError());
+}
+
+extern "C" void launch_k2(const float* F, const long long* indptr,
+                          const int* indices, const float* sumF,
+                          const float* grad, const double* llh,
+                          const int* order, const float* ladder, float* best,
+                          int n_local, int K, int n_ladder, float alpha,
+                          float min_p, float max_p, float min_f, float max_f,
+                          hipStream_t stream) {
+  if (n_local == 0) return;
+  if (n_ladder > 16) throw std::runtime_error("ladder length > 16 unsupported");
+  hipLaunchKernelGGL(k2_linesearch, dim3(n_local), dim3(256), 0, stream, F,
+                     indptr, indices, sumF, grad, llh, order, ladder, best,
+                     n_local, K, n_ladder, alpha, min_p, max_p, min_f, max_f);
+  HIP_CHECK(hipGetLastError());
+}
+
+extern "C" void launch_k3(float* F, const float* grad, const float* steps,
+                          int n_local, int K, float min_f, float max_f,
+                          hipStream_t stream) {
+  if (n_local == 0) return;
+  hipLaunchKernelGGL(k3_apply_step, dim3(n_local), dim3(256), 0, stream, F,
+                     grad, steps, n_local, K, min_f, max_f);
+  HIP_CHECK(hipGetLastError());
 }

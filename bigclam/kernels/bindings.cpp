@@ -14,6 +14,8 @@ extern "C" void launch_k2(const float*, const long long*, const int*,
                           const float*, const float*, const double*,
                           const int*, const float*, float*, int, int, int,
                           float, float, float, float, float, hipStream_t);
+extern "C" void launch_k3(float*, const float*, const float*, int, int,
+                          float, float, hipStream_t);
 
 namespace {
 
@@ -92,6 +94,21 @@ void linesearch(torch::Tensor F, torch::Tensor indptr, torch::Tensor indices,
             (float)max_f, current_stream());
 }
 
+void apply_step(torch::Tensor F_local, torch::Tensor grad,
+                torch::Tensor steps, double min_f, double max_f) {
+  CHECK_IN(F_local, torch::kFloat32);
+  CHECK_IN(grad, torch::kFloat32);
+  CHECK_IN(steps, torch::kFloat32);
+  const int n_local = (int)F_local.size(0);
+  const int K = (int)F_local.size(1);
+  TORCH_CHECK(K % 4 == 0, "K must be padded to a multiple of 4");
+  TORCH_CHECK(grad.size(0) == n_local && grad.size(1) == K);
+  TORCH_CHECK(steps.size(0) == n_local);
+  launch_k3(F_local.data_ptr<float>(), grad.data_ptr<float>(),
+            steps.data_ptr<float>(), n_local, K, (float)min_f, (float)max_f,
+            current_stream());
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -100,4 +117,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("llh_only", &llh_only, "K4: per-node local LLH (CDNA4)");
   m.def("linesearch", &linesearch,
         "K2: 16-candidate Armijo line search in one edge pass (CDNA4)");
+  m.def("apply_step", &apply_step,
+        "K3: in-place projected commit F += s*grad (CDNA4)");
 }

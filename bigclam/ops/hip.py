@@ -85,6 +85,17 @@ def linesearch(
     return best
 
 
+def apply_step(
+    F_local: torch.Tensor,
+    grad: torch.Tensor,
+    steps: torch.Tensor,
+    cfg: BigClamConfig,
+) -> None:
+    """K3: in-place projected commit on the owned rows."""
+    ext = ensure_loaded()
+    ext.apply_step(F_local, grad, steps, cfg.min_f, cfg.max_f)
+
+
 def full_llh(
     F: torch.Tensor,
     indptr: torch.Tensor,

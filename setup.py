@@ -28,7 +28,6 @@ setup(
             sources=[
                 "bigclam/kernels/bindings.cpp",
                 "bigclam/kernels/bigclam_kernels.hip",
-                "bigclam/kernels/launchers.hip",
             ],
             extra_compile_args={
                 "cxx": ["-O3"],
