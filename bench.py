@@ -65,13 +65,18 @@ def main():
         if use_cuda:
             torch.cuda.synchronize()
 
+    # pipelined loop: each timed step = one full algorithmic sweep
+    # (16-candidate line search -> projected commit -> sumF allreduce ->
+    # halo exchange -> fused grad+LLH pass -> scalar LLH allreduce); the
+    # post-update LLH comes from the next grad pass (see engine/trainer.py).
+    grad, llh_nodes, _ = tr.prologue()
     for _ in range(args.warmup):
-        tr.sweep()
+        grad, llh_nodes, _, _ = tr.pipelined_sweep(grad, llh_nodes)
     comm.barrier()
     sync()
     t0 = time.perf_counter()
     for _ in range(args.steps):
-        tr.sweep()
+        grad, llh_nodes, _, _ = tr.pipelined_sweep(grad, llh_nodes)
     comm.barrier()
     sync()
     elapsed = time.perf_counter() - t0
@@ -112,6 +117,7 @@ def main():
                         "k": args.k,
                         "init": "random",
                         "ladder": 16,
+                        "sweep": "pipelined (grad+llh fused; post-update LLH = next grad pass)",
                         "parallelism": f"dp{world} row-sharded, RCCL halo",
                     },
                 }

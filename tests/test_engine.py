@@ -116,3 +116,22 @@ def test_select_k_small():
     out = select_k(g, cfg)
     assert out["k"] in out["grid"] or out["k"] == 0
     assert len(out["history"]) >= 1
+
+
+def test_pipelined_fit_matches_sweep_trajectory(small_graph):
+    """fit() pipelines K4 away: trajectory must equal the unfused sweeps."""
+    g = small_graph
+    k = 3
+    rng = np.random.default_rng(17)
+    F0 = (rng.random((g.num_nodes, k)) * 0.3).astype(np.float32)
+
+    cfg = BigClamConfig(k=k, device="cpu", max_sweeps=4)
+    tr1 = Trainer(g, cfg, rank=0, world_size=1, device=torch.device("cpu"))
+    tr1.state.set_local_F(torch.from_numpy(F0))
+    unfused = [tr1.sweep()["llh"] for _ in range(4)]
+
+    tr2 = Trainer(g, cfg, rank=0, world_size=1, device=torch.device("cpu"))
+    tr2.state.set_local_F(torch.from_numpy(F0))
+    res = tr2.fit(skip_init=True)
+    for a, b in zip(unfused, res.llh_history):
+        assert abs(a - b) < 1e-9 * max(1.0, abs(a)), (unfused, res.llh_history)
