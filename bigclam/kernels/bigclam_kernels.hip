@@ -89,6 +89,95 @@ __device__ __forceinline__ float4 ld4(const float* p) {
 // dot (block_allreduce) and an owned-k accumulate; fv's second read hits L1
 // (the block touches one 4*K-byte row at a time).
 
+template <int NSLOT>
+__global__ void __launch_bounds__(BLOCK) k1_grad_llh_t(
+    const float* __restrict__ F, const long long* __restrict__ indptr,
+    const int* __restrict__ indices, const float* __restrict__ sumF,
+    const int* __restrict__ order, float* __restrict__ grad,
+    double* __restrict__ llh, int n_local, int K, float min_p, float max_p) {
+  const int u = order[blockIdx.x];
+  const long long e0 = indptr[u];
+  const long long e1 = indptr[u + 1];
+  const int tid = threadIdx.x;
+
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  float* gacc = reinterpret_cast<float*>(smem);            // K floats
+  float* red = reinterpret_cast<float*>(smem + (size_t)K * 4);  // NWAVE
+
+  const float* __restrict__ fu = F + (size_t)u * K;
+  // fu resident in registers: dot pass streams only fv, so the axpy
+  // re-read of fv stays L1-hot (fu+fv would overflow the 32 KB L1)
+  float4 fu4[NSLOT];
+#pragma unroll
+  for (int sl = 0; sl < NSLOT; ++sl) {
+    const int k = tid * 4 + sl * (BLOCK * 4);
+    fu4[sl] = (k < K) ? ld4(fu + k) : float4{0.f, 0.f, 0.f, 0.f};
+    if (k < K) *reinterpret_cast<float4*>(gacc + k) = float4{0.f, 0.f, 0.f, 0.f};
+  }
+  __syncthreads();
+
+  double llh_acc = 0.0;  // accumulated on thread 0 only
+
+#pragma clang loop unroll(disable)
+  for (long long e = e0; e < e1; ++e) {
+    const float* __restrict__ fv = F + (size_t)indices[e] * K;
+    float part = 0.f;
+#pragma unroll
+    for (int sl = 0; sl < NSLOT; ++sl) {
+      const int k = tid * 4 + sl * (BLOCK * 4);
+      if (k < K) part = dot4(fu4[sl], ld4(fv + k), part);
+    }
+    const float x = block_allreduce_sum(part, red);
+    const float p = clamp_p(__expf(-x), min_p, max_p);
+    const float w = 1.f / (1.f - p);
+    if (tid == 0) llh_acc += (double)log1pf(-p) + (double)x;
+#pragma unroll
+    for (int sl = 0; sl < NSLOT; ++sl) {
+      const int k = tid * 4 + sl * (BLOCK * 4);
+      if (k < K) {
+        const float4 b = ld4(fv + k);  // L1-hot: just read by this block
+        float4 g = ld4(gacc + k);
+        g.x = fmaf(w, b.x, g.x);
+        g.y = fmaf(w, b.y, g.y);
+        g.z = fmaf(w, b.z, g.z);
+        g.w = fmaf(w, b.w, g.w);
+        *reinterpret_cast<float4*>(gacc + k) = g;
+      }
+    }
+    // no barrier needed: each thread owns its gacc elements; the next
+    // edge's block_allreduce syncs before x is consumed.
+  }
+
+  // node terms: -Fu.sumF + Fu.Fu
+  float p_fs = 0.f, p_ff = 0.f;
+#pragma unroll
+  for (int sl = 0; sl < NSLOT; ++sl) {
+    const int k = tid * 4 + sl * (BLOCK * 4);
+    if (k < K) {
+      const float4 s = ld4(sumF + k);
+      p_fs = dot4(fu4[sl], s, p_fs);
+      p_ff = dot4(fu4[sl], fu4[sl], p_ff);
+    }
+  }
+  const float fs = block_allreduce_sum(p_fs, red);
+  const float ff = block_allreduce_sum(p_ff, red);
+
+  float* __restrict__ gout = grad + (size_t)u * K;
+#pragma unroll
+  for (int sl = 0; sl < NSLOT; ++sl) {
+    const int k = tid * 4 + sl * (BLOCK * 4);
+    if (k < K) {
+      const float4 g = ld4(gacc + k);
+      const float4 sv = ld4(sumF + k);
+      *reinterpret_cast<float4*>(gout + k) =
+          float4{g.x - sv.x + fu4[sl].x, g.y - sv.y + fu4[sl].y,
+                 g.z - sv.z + fu4[sl].z, g.w - sv.w + fu4[sl].w};
+    }
+  }
+  if (tid == 0) llh[u] = llh_acc + (double)(-fs) + (double)ff;
+}
+
+// generic variant for K > 8192: fu streamed from memory (L2) per edge
 extern "C" __global__ void __launch_bounds__(BLOCK) k1_grad_llh(
     const float* __restrict__ F, const long long* __restrict__ indptr,
     const int* __restrict__ indices, const float* __restrict__ sumF,
@@ -261,8 +350,17 @@ extern "C" __global__ void __launch_bounds__(BLOCK) k2_linesearch(
   for (long long t0 = e0; t0 < e1 || t0 == e0; t0 += K2_TILE) {
     const int tlen = (int)((e1 - t0) < K2_TILE ? (e1 - t0) : K2_TILE);
     float acc[K2_TILE / NWAVE];
+    // hoist the wave's edge row bases out of the chunk loop (the indices[]
+    // load otherwise serializes every chunk behind a vmcnt(0))
+    const float* fvb[K2_TILE / NWAVE];
 #pragma unroll
-    for (int i = 0; i < K2_TILE / NWAVE; ++i) acc[i] = 0.f;
+    for (int i = 0; i < K2_TILE / NWAVE; ++i) {
+      acc[i] = 0.f;
+      const long long e = t0 + (long long)(i * NWAVE + wid);
+      fvb[i] = (e < e1 && e < t0 + K2_TILE)
+                   ? F + (size_t)indices[e] * K
+                   : nullptr;
+    }
     const bool first_tile = (t0 == e0);
 
     for (int c0 = 0; c0 < K; c0 += K2_CHUNK) {
@@ -288,12 +386,10 @@ extern "C" __global__ void __launch_bounds__(BLOCK) k2_linesearch(
       // spill it to scratch)
 #pragma unroll
       for (int i = 0; i < K2_TILE / NWAVE; ++i) {
-        const long long e = t0 + (long long)(i * NWAVE + wid);
-        if (e < t0 + tlen) {
-          const float* __restrict__ fv =
-              F + (size_t)indices[e] * K + c0;
+        if (fvb[i]) {
+          const float* __restrict__ fv = fvb[i] + c0;
           float a = acc[i];
-#pragma clang loop unroll_count(4)
+#pragma clang loop unroll_count(8)
           for (int m = sub; m * 4 < clen; m += 4) {
             a = dot4(ld4(&cand[j][m * 4]), ld4(fv + m * 4), a);
           }
@@ -419,10 +515,28 @@ extern "C" void launch_k1(const float* F, const long long* indptr,
                           hipStream_t stream) {
   if (n_local == 0) return;
   const size_t lds = (size_t)K * 4 + 4 * sizeof(float);
-  allow_large_lds((const void*)k1_grad_llh, lds);
-  hipLaunchKernelGGL(k1_grad_llh, dim3(n_local), dim3(256), lds, stream, F,
-                     indptr, indices, sumF, order, grad, llh, n_local, K,
-                     min_p, max_p);
+#define K1_CASE(NS)                                                         \
+  do {                                                                      \
+    allow_large_lds((const void*)&k1_grad_llh_t<NS>, lds);                  \
+    hipLaunchKernelGGL((k1_grad_llh_t<NS>), dim3(n_local), dim3(256), lds,  \
+                       stream, F, indptr, indices, sumF, order, grad, llh,  \
+                       n_local, K, min_p, max_p);                           \
+  } while (0)
+  if (K <= 1024) {
+    K1_CASE(1);
+  } else if (K <= 2048) {
+    K1_CASE(2);
+  } else if (K <= 4096) {
+    K1_CASE(4);
+  } else if (K <= 8192) {
+    K1_CASE(8);
+  } else {
+    allow_large_lds((const void*)k1_grad_llh, lds);
+    hipLaunchKernelGGL(k1_grad_llh, dim3(n_local), dim3(256), lds, stream, F,
+                       indptr, indices, sumF, order, grad, llh, n_local, K,
+                       min_p, max_p);
+  }
+#undef K1_CASE
   HIP_CHECK(hipGetLastError());
 }
 
