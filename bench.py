@@ -42,6 +42,7 @@ def main():
     ap.add_argument("--edges", type=int, default=925872)  # com-Amazon E
     ap.add_argument("--dtype", choices=["fp32", "bf16"], default="fp32")
     ap.add_argument("--graph-seed", type=int, default=42)
+    ap.add_argument("--locality", type=float, default=0.7)
     args = ap.parse_args()
 
     rank = comm.init_distributed()
@@ -49,7 +50,7 @@ def main():
     use_cuda = torch.cuda.is_available()
     device = torch.device("cuda") if use_cuda else torch.device("cpu")
 
-    graph = shaped_graph(args.nodes, args.edges, seed=args.graph_seed)
+    graph = shaped_graph(args.nodes, args.edges, locality=args.locality, seed=args.graph_seed)
     cfg = BigClamConfig(
         k=args.k,
         dtype=args.dtype,

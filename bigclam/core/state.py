@@ -43,10 +43,13 @@ class ShardState:
         self.storage_dtype = (
             torch.bfloat16 if cfg.dtype == "bf16" else torch.float32
         )
-        # pad K to a multiple of 4: rows stay 16B-aligned for float4 kernels;
-        # pad columns are identically zero and stay zero through every op
-        # (grad_pad = -sumF_pad + F_pad = 0, clamp(0 + s*0) = 0).
-        self.kp = (cfg.k + 3) & ~3
+        # pad K so rows stay 16B-aligned for vector kernels (fp32: float4,
+        # bf16: uint4 of 8 elements); pad columns are identically zero and
+        # stay zero through every op (grad_pad = -sumF_pad + F_pad = 0,
+        # clamp(0 + s*0) = 0).
+        self.kp = (
+            (cfg.k + 7) & ~7 if self.storage_dtype == torch.bfloat16 else (cfg.k + 3) & ~3
+        )
 
         dev = self.device
         self.indptr = torch.from_numpy(shard.indptr).to(dev)
@@ -80,11 +83,6 @@ class ShardState:
         )
         if self.use_hip:
             _hip_ops().ensure_loaded()  # fail loudly if the .so is missing
-            if self.storage_dtype == torch.bfloat16:
-                raise NotImplementedError(
-                    "bf16 F storage on GPU: HIP bf16 kernels not built yet "
-                    "(fp32 is the supported GPU dtype in this version)"
-                )
 
     # ------------------------------------------------------------------ util
     @property

@@ -81,6 +81,47 @@ __device__ __forceinline__ float4 ld4(const float* p) {
   return *reinterpret_cast<const float4*>(p);
 }
 
+
+// ------------------------------------------------------- bf16 storage path
+//
+// F stored as bf16 (K padded to a multiple of 8 -> 16B-aligned uint4 rows);
+// all arithmetic in fp32, grad/sumF stay fp32.  bf16 = the high 16 bits of
+// fp32, so unpacking is one shift/mask per element; packing uses RNE.
+
+typedef unsigned int u32;
+
+__device__ __forceinline__ float bf_lo(u32 v) {
+  return __uint_as_float(v << 16);
+}
+__device__ __forceinline__ float bf_hi(u32 v) {
+  return __uint_as_float(v & 0xffff0000u);
+}
+
+struct f32x8 {
+  float4 a, b;
+};
+
+__device__ __forceinline__ f32x8 ld8bf(const u32* p) {
+  const uint4 u = *reinterpret_cast<const uint4*>(p);
+  f32x8 r;
+  r.a = float4{bf_lo(u.x), bf_hi(u.x), bf_lo(u.y), bf_hi(u.y)};
+  r.b = float4{bf_lo(u.z), bf_hi(u.z), bf_lo(u.w), bf_hi(u.w)};
+  return r;
+}
+
+__device__ __forceinline__ u32 pack_bf16_rne(float lo, float hi) {
+  // round-to-nearest-even bf16 truncation of two fp32 values
+  u32 l = __float_as_uint(lo);
+  u32 h = __float_as_uint(hi);
+  l += 0x7fffu + ((l >> 16) & 1u);
+  h += 0x7fffu + ((h >> 16) & 1u);
+  return (l >> 16) | (h & 0xffff0000u);
+}
+
+__device__ __forceinline__ float dot8(f32x8 a, f32x8 b, float acc) {
+  return dot4(a.b, b.b, dot4(a.a, b.a, acc));
+}
+
 // ------------------------------------------------------------------- K1
 //
 // One 256-thread block per LOCAL node (launch order = degree-descending so
@@ -487,6 +528,339 @@ extern "C" __global__ void __launch_bounds__(BLOCK) k3_apply_step(
   }
 }
 
+
+// --------------------------------------------------------- bf16 kernels
+// Same structures as the fp32 kernels; 8-element (16 B) load granularity.
+// NSLOT covers K <= NSLOT*2048.
+
+template <int NSLOT>
+__global__ void __launch_bounds__(BLOCK) k1_grad_llh_bf16_t(
+    const u32* __restrict__ F, const long long* __restrict__ indptr,
+    const int* __restrict__ indices, const float* __restrict__ sumF,
+    const int* __restrict__ order, float* __restrict__ grad,
+    double* __restrict__ llh, int n_local, int K, float min_p, float max_p) {
+  const int u = order[blockIdx.x];
+  const long long e0 = indptr[u];
+  const long long e1 = indptr[u + 1];
+  const int tid = threadIdx.x;
+
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  float* gacc = reinterpret_cast<float*>(smem);  // K floats
+  float* red = reinterpret_cast<float*>(smem + (size_t)K * 4);
+
+  const u32* __restrict__ fu = F + (size_t)u * (K / 2);
+  f32x8 fu8[NSLOT];
+#pragma unroll
+  for (int sl = 0; sl < NSLOT; ++sl) {
+    const int k = tid * 8 + sl * (BLOCK * 8);
+    if (k < K) {
+      fu8[sl] = ld8bf(fu + k / 2);
+      *reinterpret_cast<float4*>(gacc + k) = float4{0.f, 0.f, 0.f, 0.f};
+      *reinterpret_cast<float4*>(gacc + k + 4) = float4{0.f, 0.f, 0.f, 0.f};
+    } else {
+      fu8[sl].a = fu8[sl].b = float4{0.f, 0.f, 0.f, 0.f};
+    }
+  }
+  __syncthreads();
+
+  double llh_acc = 0.0;
+
+#pragma clang loop unroll(disable)
+  for (long long e = e0; e < e1; ++e) {
+    const u32* __restrict__ fv = F + (size_t)indices[e] * (K / 2);
+    float part = 0.f;
+#pragma unroll
+    for (int sl = 0; sl < NSLOT; ++sl) {
+      const int k = tid * 8 + sl * (BLOCK * 8);
+      if (k < K) part = dot8(fu8[sl], ld8bf(fv + k / 2), part);
+    }
+    const float x = block_allreduce_sum(part, red);
+    const float p = clamp_p(__expf(-x), min_p, max_p);
+    const float w = 1.f / (1.f - p);
+    if (tid == 0) llh_acc += (double)log1pf(-p) + (double)x;
+#pragma unroll
+    for (int sl = 0; sl < NSLOT; ++sl) {
+      const int k = tid * 8 + sl * (BLOCK * 8);
+      if (k < K) {
+        const f32x8 b = ld8bf(fv + k / 2);  // L1-hot
+        float4 g0 = ld4(gacc + k);
+        float4 g1 = ld4(gacc + k + 4);
+        g0.x = fmaf(w, b.a.x, g0.x);
+        g0.y = fmaf(w, b.a.y, g0.y);
+        g0.z = fmaf(w, b.a.z, g0.z);
+        g0.w = fmaf(w, b.a.w, g0.w);
+        g1.x = fmaf(w, b.b.x, g1.x);
+        g1.y = fmaf(w, b.b.y, g1.y);
+        g1.z = fmaf(w, b.b.z, g1.z);
+        g1.w = fmaf(w, b.b.w, g1.w);
+        *reinterpret_cast<float4*>(gacc + k) = g0;
+        *reinterpret_cast<float4*>(gacc + k + 4) = g1;
+      }
+    }
+  }
+
+  float p_fs = 0.f, p_ff = 0.f;
+#pragma unroll
+  for (int sl = 0; sl < NSLOT; ++sl) {
+    const int k = tid * 8 + sl * (BLOCK * 8);
+    if (k < K) {
+      const float4 s0 = ld4(sumF + k);
+      const float4 s1 = ld4(sumF + k + 4);
+      p_fs = dot4(fu8[sl].a, s0, p_fs);
+      p_fs = dot4(fu8[sl].b, s1, p_fs);
+      p_ff = dot8(fu8[sl], fu8[sl], p_ff);
+    }
+  }
+  const float fs = block_allreduce_sum(p_fs, red);
+  const float ff = block_allreduce_sum(p_ff, red);
+
+  float* __restrict__ gout = grad + (size_t)u * K;
+#pragma unroll
+  for (int sl = 0; sl < NSLOT; ++sl) {
+    const int k = tid * 8 + sl * (BLOCK * 8);
+    if (k < K) {
+      const float4 g0 = ld4(gacc + k);
+      const float4 g1 = ld4(gacc + k + 4);
+      const float4 s0 = ld4(sumF + k);
+      const float4 s1 = ld4(sumF + k + 4);
+      *reinterpret_cast<float4*>(gout + k) =
+          float4{g0.x - s0.x + fu8[sl].a.x, g0.y - s0.y + fu8[sl].a.y,
+                 g0.z - s0.z + fu8[sl].a.z, g0.w - s0.w + fu8[sl].a.w};
+      *reinterpret_cast<float4*>(gout + k + 4) =
+          float4{g1.x - s1.x + fu8[sl].b.x, g1.y - s1.y + fu8[sl].b.y,
+                 g1.z - s1.z + fu8[sl].b.z, g1.w - s1.w + fu8[sl].b.w};
+    }
+  }
+  if (tid == 0) llh[u] = llh_acc + (double)(-fs) + (double)ff;
+}
+
+extern "C" __global__ void __launch_bounds__(BLOCK) k4_llh_only_bf16(
+    const u32* __restrict__ F, const long long* __restrict__ indptr,
+    const int* __restrict__ indices, const float* __restrict__ sumF,
+    const int* __restrict__ order, double* __restrict__ llh, int n_local,
+    int K, float min_p, float max_p) {
+  const int u = order[blockIdx.x];
+  const long long e0 = indptr[u];
+  const long long e1 = indptr[u + 1];
+  const int tid = threadIdx.x;
+  const int lane = tid & (WAVE - 1);
+  const int wid = tid >> 6;
+
+  __shared__ __attribute__((aligned(16))) double wllh[NWAVE];
+  __shared__ __attribute__((aligned(16))) float red[NWAVE];
+
+  const u32* __restrict__ fu = F + (size_t)u * (K / 2);
+  double llh_w = 0.0;
+  for (long long e = e0 + wid; e < e1; e += NWAVE) {
+    const u32* __restrict__ fv = F + (size_t)indices[e] * (K / 2);
+    float part = 0.f;
+    for (int k = lane * 8; k < K; k += WAVE * 8) {
+      part = dot8(ld8bf(fu + k / 2), ld8bf(fv + k / 2), part);
+    }
+    const float x = wave_allreduce_sum(part);
+    const float p = clamp_p(__expf(-x), min_p, max_p);
+    if (lane == 0) llh_w += (double)log1pf(-p) + (double)x;
+  }
+  if (lane == 0) wllh[wid] = llh_w;
+  __syncthreads();
+
+  float p_fs = 0.f, p_ff = 0.f;
+  for (int k = tid * 8; k < K; k += BLOCK * 8) {
+    const f32x8 a = ld8bf(fu + k / 2);
+    p_fs = dot4(a.a, ld4(sumF + k), p_fs);
+    p_fs = dot4(a.b, ld4(sumF + k + 4), p_fs);
+    p_ff = dot8(a, a, p_ff);
+  }
+  const float fs = block_allreduce_sum(p_fs, red);
+  const float ff = block_allreduce_sum(p_ff, red);
+  if (tid == 0) {
+    double t = 0.0;
+#pragma unroll
+    for (int wv = 0; wv < NWAVE; ++wv) t += wllh[wv];
+    llh[u] = t + (double)(-fs) + (double)ff;
+  }
+}
+
+extern "C" __global__ void __launch_bounds__(BLOCK) k2_linesearch_bf16(
+    const u32* __restrict__ F, const long long* __restrict__ indptr,
+    const int* __restrict__ indices, const float* __restrict__ sumF,
+    const float* __restrict__ grad, const double* __restrict__ llh,
+    const int* __restrict__ order, const float* __restrict__ ladder,
+    float* __restrict__ best, int n_local, int K, int n_ladder, float alpha,
+    float min_p, float max_p, float min_f, float max_f) {
+  const int u = order[blockIdx.x];
+  const long long e0 = indptr[u];
+  const long long e1 = indptr[u + 1];
+  const int tid = threadIdx.x;
+  const int lane = tid & (WAVE - 1);
+  const int wid = tid >> 6;
+  const int sub = lane >> 4;
+  const int j = lane & 15;
+
+  // +4 pad floats: row stride 2064 B spreads the 16 j rows over bank quads
+  __shared__ __attribute__((aligned(16))) float cand[MAX_LS][K2_CHUNK + 4];
+  __shared__ __attribute__((aligned(16))) float s_lad[MAX_LS];
+  __shared__ __attribute__((aligned(16))) double acc_llh[NWAVE][MAX_LS];
+  __shared__ __attribute__((aligned(16))) float acc_nt[NWAVE][MAX_LS];
+  __shared__ __attribute__((aligned(16))) float red[NWAVE];
+
+  if (tid < MAX_LS) s_lad[tid] = (tid < n_ladder) ? ladder[tid] : 0.f;
+  __syncthreads();
+
+  const u32* __restrict__ fub = F + (size_t)u * (K / 2);
+  const float* __restrict__ gu = grad + (size_t)u * K;
+
+  double llh_j = 0.0;
+  float nt_j = 0.f;
+  float p_gg = 0.f;
+
+  for (long long t0 = e0; t0 < e1 || t0 == e0; t0 += K2_TILE) {
+    const int tlen = (int)((e1 - t0) < K2_TILE ? (e1 - t0) : K2_TILE);
+    float acc[K2_TILE / NWAVE];
+    const u32* fvb[K2_TILE / NWAVE];
+#pragma unroll
+    for (int i = 0; i < K2_TILE / NWAVE; ++i) {
+      acc[i] = 0.f;
+      const long long e = t0 + (long long)(i * NWAVE + wid);
+      fvb[i] = (e < e1 && e < t0 + K2_TILE)
+                   ? F + (size_t)indices[e] * (K / 2)
+                   : nullptr;
+    }
+    const bool first_tile = (t0 == e0);
+
+    for (int c0 = 0; c0 < K; c0 += K2_CHUNK) {
+      const int clen = (K - c0) < K2_CHUNK ? (K - c0) : K2_CHUNK;
+      __syncthreads();
+      for (int jj = 0; jj < MAX_LS; ++jj) {
+        const float sjj = s_lad[jj];
+        for (int kk = tid * 8; kk < clen; kk += BLOCK * 8) {
+          const f32x8 a = ld8bf(fub + (c0 + kk) / 2);
+          const float4 g0 = ld4(gu + c0 + kk);
+          const float4 g1 = ld4(gu + c0 + kk + 4);
+          *reinterpret_cast<float4*>(&cand[jj][kk]) =
+              float4{fminf(fmaxf(fmaf(sjj, g0.x, a.a.x), min_f), max_f),
+                     fminf(fmaxf(fmaf(sjj, g0.y, a.a.y), min_f), max_f),
+                     fminf(fmaxf(fmaf(sjj, g0.z, a.a.z), min_f), max_f),
+                     fminf(fmaxf(fmaf(sjj, g0.w, a.a.w), min_f), max_f)};
+          *reinterpret_cast<float4*>(&cand[jj][kk + 4]) =
+              float4{fminf(fmaxf(fmaf(sjj, g1.x, a.b.x), min_f), max_f),
+                     fminf(fmaxf(fmaf(sjj, g1.y, a.b.y), min_f), max_f),
+                     fminf(fmaxf(fmaf(sjj, g1.z, a.b.z), min_f), max_f),
+                     fminf(fmaxf(fmaf(sjj, g1.w, a.b.w), min_f), max_f)};
+        }
+      }
+      __syncthreads();
+
+#pragma unroll
+      for (int i = 0; i < K2_TILE / NWAVE; ++i) {
+        if (fvb[i]) {
+          const u32* __restrict__ fv = fvb[i] + c0 / 2;
+          float a = acc[i];
+          // m indexes 8-element slots; subs cover slots mod 4
+#pragma clang loop unroll_count(4)
+          for (int m = sub; m * 8 < clen; m += 4) {
+            const f32x8 b = ld8bf(fv + m * 4);
+            a = dot4(ld4(&cand[j][m * 8]), b.a, a);
+            a = dot4(ld4(&cand[j][m * 8 + 4]), b.b, a);
+          }
+          acc[i] = a;
+        }
+      }
+
+      if (first_tile) {
+        const int m0 = sub + 4 * wid;
+        for (int m = m0; m * 8 < clen; m += 16) {
+          const float4 c0v = ld4(&cand[j][m * 8]);
+          const float4 c1v = ld4(&cand[j][m * 8 + 4]);
+          const f32x8 a = ld8bf(fub + (c0 + m * 8) / 2);
+          const float4 s0 = ld4(sumF + c0 + m * 8);
+          const float4 s1 = ld4(sumF + c0 + m * 8 + 4);
+          nt_j = fmaf(c0v.x, a.a.x - s0.x, nt_j);
+          nt_j = fmaf(c0v.y, a.a.y - s0.y, nt_j);
+          nt_j = fmaf(c0v.z, a.a.z - s0.z, nt_j);
+          nt_j = fmaf(c0v.w, a.a.w - s0.w, nt_j);
+          nt_j = fmaf(c1v.x, a.b.x - s1.x, nt_j);
+          nt_j = fmaf(c1v.y, a.b.y - s1.y, nt_j);
+          nt_j = fmaf(c1v.z, a.b.z - s1.z, nt_j);
+          nt_j = fmaf(c1v.w, a.b.w - s1.w, nt_j);
+        }
+        for (int kk = tid * 4; kk < clen; kk += BLOCK * 4) {
+          const float4 g = ld4(gu + c0 + kk);
+          p_gg = dot4(g, g, p_gg);
+        }
+      }
+    }
+
+#pragma unroll
+    for (int i = 0; i < K2_TILE / NWAVE; ++i) {
+      float x = acc[i];
+      x += __shfl_xor(x, 16, WAVE);
+      x += __shfl_xor(x, 32, WAVE);
+      const long long e = t0 + (long long)(i * NWAVE + wid);
+      if (sub == 0 && e < t0 + tlen) {
+        const float p = clamp_p(__expf(-x), min_p, max_p);
+        llh_j += (double)log1pf(-p) + (double)x;
+      }
+    }
+    if (e1 == e0) break;
+  }
+
+  nt_j += __shfl_xor(nt_j, 16, WAVE);
+  nt_j += __shfl_xor(nt_j, 32, WAVE);
+  if (sub == 0) {
+    acc_llh[wid][j] = llh_j;
+    acc_nt[wid][j] = nt_j;
+  }
+  const float gg = block_allreduce_sum(p_gg, red);
+
+  if (tid == 0) {
+    const double llh_u = llh[u];
+    float chosen = 0.f;
+    for (int jj = 0; jj < n_ladder; ++jj) {
+      double trial = 0.0;
+#pragma unroll
+      for (int wv = 0; wv < NWAVE; ++wv)
+        trial += acc_llh[wv][jj] + (double)acc_nt[wv][jj];
+      const float sjj = s_lad[jj];
+      if (trial >= llh_u + (double)(alpha * sjj * gg)) {
+        chosen = sjj;
+        break;
+      }
+    }
+    best[u] = chosen;
+  }
+}
+
+extern "C" __global__ void __launch_bounds__(BLOCK) k3_apply_step_bf16(
+    u32* __restrict__ F, const float* __restrict__ grad,
+    const float* __restrict__ steps, int n_local, int K, float min_f,
+    float max_f) {
+  const int u = blockIdx.x;
+  const float s = steps[u];
+  if (s <= 0.f) return;
+  u32* __restrict__ fu = F + (size_t)u * (K / 2);
+  const float* __restrict__ gu = grad + (size_t)u * K;
+  for (int k = threadIdx.x * 8; k < K; k += BLOCK * 8) {
+    const f32x8 a = ld8bf(fu + k / 2);
+    const float4 g0 = ld4(gu + k);
+    const float4 g1 = ld4(gu + k + 4);
+    float v0 = fminf(fmaxf(fmaf(s, g0.x, a.a.x), min_f), max_f);
+    float v1 = fminf(fmaxf(fmaf(s, g0.y, a.a.y), min_f), max_f);
+    float v2 = fminf(fmaxf(fmaf(s, g0.z, a.a.z), min_f), max_f);
+    float v3 = fminf(fmaxf(fmaf(s, g0.w, a.a.w), min_f), max_f);
+    float v4 = fminf(fmaxf(fmaf(s, g1.x, a.b.x), min_f), max_f);
+    float v5 = fminf(fmaxf(fmaf(s, g1.y, a.b.y), min_f), max_f);
+    float v6 = fminf(fmaxf(fmaf(s, g1.z, a.b.z), min_f), max_f);
+    float v7 = fminf(fmaxf(fmaf(s, g1.w, a.b.w), min_f), max_f);
+    uint4 o;
+    o.x = pack_bf16_rne(v0, v1);
+    o.y = pack_bf16_rne(v2, v3);
+    o.z = pack_bf16_rne(v4, v5);
+    o.w = pack_bf16_rne(v6, v7);
+    *reinterpret_cast<uint4*>(fu + k / 2) = o;
+  }
+}
+
 // ----------------------------------------------------------- host launchers
 
 #include <stdexcept>
@@ -506,6 +880,74 @@ static void allow_large_lds(const void* func, size_t bytes) {
     HIP_CHECK(hipFuncSetAttribute(
         func, hipFuncAttributeMaxDynamicSharedMemorySize, (int)bytes));
   }
+}
+
+extern "C" void launch_k1_bf16(const void* F, const long long* indptr,
+                               const int* indices, const float* sumF,
+                               const int* order, float* grad, double* llh,
+                               int n_local, int K, float min_p, float max_p,
+                               hipStream_t stream) {
+  if (n_local == 0) return;
+  const size_t lds = (size_t)K * 4 + 4 * sizeof(float);
+  const u32* Fb = reinterpret_cast<const u32*>(F);
+#define K1B_CASE(NS)                                                        \
+  do {                                                                      \
+    allow_large_lds((const void*)&k1_grad_llh_bf16_t<NS>, lds);             \
+    hipLaunchKernelGGL((k1_grad_llh_bf16_t<NS>), dim3(n_local), dim3(256),  \
+                       lds, stream, Fb, indptr, indices, sumF, order, grad, \
+                       llh, n_local, K, min_p, max_p);                      \
+  } while (0)
+  if (K <= 2048) {
+    K1B_CASE(1);
+  } else if (K <= 4096) {
+    K1B_CASE(2);
+  } else if (K <= 8192) {
+    K1B_CASE(4);
+  } else if (K <= 16384) {
+    K1B_CASE(8);
+  } else {
+    throw std::runtime_error("bf16 K1: K > 16384 unsupported");
+  }
+#undef K1B_CASE
+  HIP_CHECK(hipGetLastError());
+}
+
+extern "C" void launch_k4_bf16(const void* F, const long long* indptr,
+                               const int* indices, const float* sumF,
+                               const int* order, double* llh, int n_local,
+                               int K, float min_p, float max_p,
+                               hipStream_t stream) {
+  if (n_local == 0) return;
+  hipLaunchKernelGGL(k4_llh_only_bf16, dim3(n_local), dim3(256), 0, stream,
+                     reinterpret_cast<const u32*>(F), indptr, indices, sumF,
+                     order, llh, n_local, K, min_p, max_p);
+  HIP_CHECK(hipGetLastError());
+}
+
+extern "C" void launch_k2_bf16(const void* F, const long long* indptr,
+                               const int* indices, const float* sumF,
+                               const float* grad, const double* llh,
+                               const int* order, const float* ladder,
+                               float* best, int n_local, int K, int n_ladder,
+                               float alpha, float min_p, float max_p,
+                               float min_f, float max_f, hipStream_t stream) {
+  if (n_local == 0) return;
+  if (n_ladder > 16) throw std::runtime_error("ladder length > 16 unsupported");
+  hipLaunchKernelGGL(k2_linesearch_bf16, dim3(n_local), dim3(256), 0, stream,
+                     reinterpret_cast<const u32*>(F), indptr, indices, sumF,
+                     grad, llh, order, ladder, best, n_local, K, n_ladder,
+                     alpha, min_p, max_p, min_f, max_f);
+  HIP_CHECK(hipGetLastError());
+}
+
+extern "C" void launch_k3_bf16(void* F, const float* grad, const float* steps,
+                               int n_local, int K, float min_f, float max_f,
+                               hipStream_t stream) {
+  if (n_local == 0) return;
+  hipLaunchKernelGGL(k3_apply_step_bf16, dim3(n_local), dim3(256), 0, stream,
+                     reinterpret_cast<u32*>(F), grad, steps, n_local, K,
+                     min_f, max_f);
+  HIP_CHECK(hipGetLastError());
 }
 
 extern "C" void launch_k1(const float* F, const long long* indptr,

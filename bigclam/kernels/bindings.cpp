@@ -16,6 +16,19 @@ extern "C" void launch_k2(const float*, const long long*, const int*,
                           float, float, float, float, float, hipStream_t);
 extern "C" void launch_k3(float*, const float*, const float*, int, int,
                           float, float, hipStream_t);
+extern "C" void launch_k1_bf16(const void*, const long long*, const int*,
+                               const float*, const int*, float*, double*,
+                               int, int, float, float, hipStream_t);
+extern "C" void launch_k4_bf16(const void*, const long long*, const int*,
+                               const float*, const int*, double*, int, int,
+                               float, float, hipStream_t);
+extern "C" void launch_k2_bf16(const void*, const long long*, const int*,
+                               const float*, const float*, const double*,
+                               const int*, const float*, float*, int, int,
+                               int, float, float, float, float, float,
+                               hipStream_t);
+extern "C" void launch_k3_bf16(void*, const float*, const float*, int, int,
+                               float, float, hipStream_t);
 
 namespace {
 
@@ -23,6 +36,17 @@ namespace {
   TORCH_CHECK(t.is_cuda(), #t " must be on GPU");                    \
   TORCH_CHECK(t.is_contiguous(), #t " must be contiguous");          \
   TORCH_CHECK(t.scalar_type() == type, #t " has wrong dtype");
+
+#define CHECK_F(t)                                                          \
+  TORCH_CHECK(t.is_cuda(), #t " must be on GPU");                           \
+  TORCH_CHECK(t.is_contiguous(), #t " must be contiguous");                 \
+  TORCH_CHECK(t.scalar_type() == torch::kFloat32 ||                         \
+                  t.scalar_type() == torch::kBFloat16,                      \
+              #t " must be fp32 or bf16");
+
+bool is_bf16(const torch::Tensor& t) {
+  return t.scalar_type() == torch::kBFloat16;
+}
 
 hipStream_t current_stream() {
   return at::hip::getCurrentHIPStream().stream();
@@ -32,7 +56,7 @@ void edge_grad_llh(torch::Tensor F, torch::Tensor indptr,
                    torch::Tensor indices, torch::Tensor sumF,
                    torch::Tensor order, torch::Tensor grad, torch::Tensor llh,
                    double min_p, double max_p) {
-  CHECK_IN(F, torch::kFloat32);
+  CHECK_F(F);
   CHECK_IN(indptr, torch::kInt64);
   CHECK_IN(indices, torch::kInt32);
   CHECK_IN(sumF, torch::kFloat32);
@@ -41,19 +65,27 @@ void edge_grad_llh(torch::Tensor F, torch::Tensor indptr,
   CHECK_IN(llh, torch::kFloat64);
   const int n_local = (int)indptr.size(0) - 1;
   const int K = (int)F.size(1);
-  TORCH_CHECK(K % 4 == 0, "K must be padded to a multiple of 4");
   TORCH_CHECK(grad.size(0) == n_local && grad.size(1) == K);
-  launch_k1(F.data_ptr<float>(), reinterpret_cast<const long long*>(indptr.data_ptr<int64_t>()),
-            indices.data_ptr<int>(), sumF.data_ptr<float>(),
-            order.data_ptr<int>(), grad.data_ptr<float>(),
-            llh.data_ptr<double>(), n_local, K, (float)min_p, (float)max_p,
-            current_stream());
+  const auto ip = reinterpret_cast<const long long*>(indptr.data_ptr<int64_t>());
+  if (is_bf16(F)) {
+    TORCH_CHECK(K % 8 == 0, "bf16 K must be padded to a multiple of 8");
+    launch_k1_bf16(F.data_ptr(), ip, indices.data_ptr<int>(),
+                   sumF.data_ptr<float>(), order.data_ptr<int>(),
+                   grad.data_ptr<float>(), llh.data_ptr<double>(), n_local, K,
+                   (float)min_p, (float)max_p, current_stream());
+  } else {
+    TORCH_CHECK(K % 4 == 0, "K must be padded to a multiple of 4");
+    launch_k1(F.data_ptr<float>(), ip, indices.data_ptr<int>(),
+              sumF.data_ptr<float>(), order.data_ptr<int>(),
+              grad.data_ptr<float>(), llh.data_ptr<double>(), n_local, K,
+              (float)min_p, (float)max_p, current_stream());
+  }
 }
 
 void llh_only(torch::Tensor F, torch::Tensor indptr, torch::Tensor indices,
               torch::Tensor sumF, torch::Tensor order, torch::Tensor llh,
               double min_p, double max_p) {
-  CHECK_IN(F, torch::kFloat32);
+  CHECK_F(F);
   CHECK_IN(indptr, torch::kInt64);
   CHECK_IN(indices, torch::kInt32);
   CHECK_IN(sumF, torch::kFloat32);
@@ -61,11 +93,20 @@ void llh_only(torch::Tensor F, torch::Tensor indptr, torch::Tensor indices,
   CHECK_IN(llh, torch::kFloat64);
   const int n_local = (int)indptr.size(0) - 1;
   const int K = (int)F.size(1);
-  TORCH_CHECK(K % 4 == 0, "K must be padded to a multiple of 4");
-  launch_k4(F.data_ptr<float>(), reinterpret_cast<const long long*>(indptr.data_ptr<int64_t>()),
-            indices.data_ptr<int>(), sumF.data_ptr<float>(),
-            order.data_ptr<int>(), llh.data_ptr<double>(), n_local, K,
-            (float)min_p, (float)max_p, current_stream());
+  const auto ip = reinterpret_cast<const long long*>(indptr.data_ptr<int64_t>());
+  if (is_bf16(F)) {
+    TORCH_CHECK(K % 8 == 0, "bf16 K must be padded to a multiple of 8");
+    launch_k4_bf16(F.data_ptr(), ip, indices.data_ptr<int>(),
+                   sumF.data_ptr<float>(), order.data_ptr<int>(),
+                   llh.data_ptr<double>(), n_local, K, (float)min_p,
+                   (float)max_p, current_stream());
+  } else {
+    TORCH_CHECK(K % 4 == 0, "K must be padded to a multiple of 4");
+    launch_k4(F.data_ptr<float>(), ip, indices.data_ptr<int>(),
+              sumF.data_ptr<float>(), order.data_ptr<int>(),
+              llh.data_ptr<double>(), n_local, K, (float)min_p, (float)max_p,
+              current_stream());
+  }
 }
 
 void linesearch(torch::Tensor F, torch::Tensor indptr, torch::Tensor indices,
@@ -73,7 +114,7 @@ void linesearch(torch::Tensor F, torch::Tensor indptr, torch::Tensor indices,
                 torch::Tensor order, torch::Tensor ladder, torch::Tensor best,
                 double alpha, double min_p, double max_p, double min_f,
                 double max_f) {
-  CHECK_IN(F, torch::kFloat32);
+  CHECK_F(F);
   CHECK_IN(indptr, torch::kInt64);
   CHECK_IN(indices, torch::kInt32);
   CHECK_IN(sumF, torch::kFloat32);
@@ -84,29 +125,47 @@ void linesearch(torch::Tensor F, torch::Tensor indptr, torch::Tensor indices,
   CHECK_IN(best, torch::kFloat32);
   const int n_local = (int)indptr.size(0) - 1;
   const int K = (int)F.size(1);
-  TORCH_CHECK(K % 4 == 0, "K must be padded to a multiple of 4");
-  launch_k2(F.data_ptr<float>(), reinterpret_cast<const long long*>(indptr.data_ptr<int64_t>()),
-            indices.data_ptr<int>(), sumF.data_ptr<float>(),
-            grad.data_ptr<float>(), llh.data_ptr<double>(),
-            order.data_ptr<int>(), ladder.data_ptr<float>(),
-            best.data_ptr<float>(), n_local, K, (int)ladder.size(0),
-            (float)alpha, (float)min_p, (float)max_p, (float)min_f,
-            (float)max_f, current_stream());
+  const auto ip = reinterpret_cast<const long long*>(indptr.data_ptr<int64_t>());
+  if (is_bf16(F)) {
+    TORCH_CHECK(K % 8 == 0, "bf16 K must be padded to a multiple of 8");
+    launch_k2_bf16(F.data_ptr(), ip, indices.data_ptr<int>(),
+                   sumF.data_ptr<float>(), grad.data_ptr<float>(),
+                   llh.data_ptr<double>(), order.data_ptr<int>(),
+                   ladder.data_ptr<float>(), best.data_ptr<float>(), n_local,
+                   K, (int)ladder.size(0), (float)alpha, (float)min_p,
+                   (float)max_p, (float)min_f, (float)max_f,
+                   current_stream());
+  } else {
+    TORCH_CHECK(K % 4 == 0, "K must be padded to a multiple of 4");
+    launch_k2(F.data_ptr<float>(), ip, indices.data_ptr<int>(),
+              sumF.data_ptr<float>(), grad.data_ptr<float>(),
+              llh.data_ptr<double>(), order.data_ptr<int>(),
+              ladder.data_ptr<float>(), best.data_ptr<float>(), n_local, K,
+              (int)ladder.size(0), (float)alpha, (float)min_p, (float)max_p,
+              (float)min_f, (float)max_f, current_stream());
+  }
 }
 
 void apply_step(torch::Tensor F_local, torch::Tensor grad,
                 torch::Tensor steps, double min_f, double max_f) {
-  CHECK_IN(F_local, torch::kFloat32);
+  CHECK_F(F_local);
   CHECK_IN(grad, torch::kFloat32);
   CHECK_IN(steps, torch::kFloat32);
   const int n_local = (int)F_local.size(0);
   const int K = (int)F_local.size(1);
-  TORCH_CHECK(K % 4 == 0, "K must be padded to a multiple of 4");
   TORCH_CHECK(grad.size(0) == n_local && grad.size(1) == K);
   TORCH_CHECK(steps.size(0) == n_local);
-  launch_k3(F_local.data_ptr<float>(), grad.data_ptr<float>(),
-            steps.data_ptr<float>(), n_local, K, (float)min_f, (float)max_f,
-            current_stream());
+  if (is_bf16(F_local)) {
+    TORCH_CHECK(K % 8 == 0, "bf16 K must be padded to a multiple of 8");
+    launch_k3_bf16(F_local.data_ptr(), grad.data_ptr<float>(),
+                   steps.data_ptr<float>(), n_local, K, (float)min_f,
+                   (float)max_f, current_stream());
+  } else {
+    TORCH_CHECK(K % 4 == 0, "K must be padded to a multiple of 4");
+    launch_k3(F_local.data_ptr<float>(), grad.data_ptr<float>(),
+              steps.data_ptr<float>(), n_local, K, (float)min_f,
+              (float)max_f, current_stream());
+  }
 }
 
 }  // namespace
