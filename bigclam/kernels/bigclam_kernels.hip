@@ -398,9 +398,11 @@ extern "C" __global__ void __launch_bounds__(BLOCK) k2_linesearch(
     for (int i = 0; i < K2_TILE / NWAVE; ++i) {
       acc[i] = 0.f;
       const long long e = t0 + (long long)(i * NWAVE + wid);
-      fvb[i] = (e < e1 && e < t0 + K2_TILE)
-                   ? F + (size_t)indices[e] * K
-                   : nullptr;
+      // invalid slots point at row 0: loads stay UNCONDITIONAL (an
+      // if-guard per edge makes hipcc emit exec-masked blocks that
+      // serialize every load behind vmcnt(0)); their acc is never read
+      // (the fold re-checks e < tlen).
+      fvb[i] = (e < e1 && e < t0 + K2_TILE) ? F + (size_t)indices[e] * K : F;
     }
     const bool first_tile = (t0 == e0);
 
@@ -425,17 +427,23 @@ extern "C" __global__ void __launch_bounds__(BLOCK) k2_linesearch(
       // edges of this tile over this chunk; acc[i] persists across chunks
       // (full unroll keeps acc[] in registers — runtime indexing would
       // spill it to scratch)
+      // slot-outer / edge-inner: one LDS cand read serves all 8 edges and
+      // their 8 independent fv loads issue back-to-back (deep MLP) — an
+      // edge-outer nest made hipcc serialize every load behind vmcnt(0)
+      for (int m = sub; m * 4 < clen; m += 4) {
+        // issue all 8 independent loads, then a value barrier: the
+        // compiler emits ONE vmcnt drain for the batch instead of pairing
+        // loads 2-deep behind pk_fma SLP marshalling
+        float4 b[K2_TILE / NWAVE];
 #pragma unroll
-      for (int i = 0; i < K2_TILE / NWAVE; ++i) {
-        if (fvb[i]) {
-          const float* __restrict__ fv = fvb[i] + c0;
-          float a = acc[i];
-#pragma clang loop unroll_count(8)
-          for (int m = sub; m * 4 < clen; m += 4) {
-            a = dot4(ld4(&cand[j][m * 4]), ld4(fv + m * 4), a);
-          }
-          acc[i] = a;
-        }
+        for (int i = 0; i < K2_TILE / NWAVE; ++i)
+          b[i] = ld4(fvb[i] + c0 + m * 4);
+        asm volatile("" ::"v"(b[0].x), "v"(b[1].x), "v"(b[2].x), "v"(b[3].x),
+                     "v"(b[4].x), "v"(b[5].x), "v"(b[6].x), "v"(b[7].x));
+        const float4 c = ld4(&cand[j][m * 4]);
+#pragma unroll
+        for (int i = 0; i < K2_TILE / NWAVE; ++i)
+          acc[i] = dot4(c, b[i], acc[i]);
       }
 
       // node term for this chunk: waves split the slot range; once per node
@@ -722,9 +730,8 @@ extern "C" __global__ void __launch_bounds__(BLOCK) k2_linesearch_bf16(
     for (int i = 0; i < K2_TILE / NWAVE; ++i) {
       acc[i] = 0.f;
       const long long e = t0 + (long long)(i * NWAVE + wid);
-      fvb[i] = (e < e1 && e < t0 + K2_TILE)
-                   ? F + (size_t)indices[e] * (K / 2)
-                   : nullptr;
+      fvb[i] = (e < e1 && e < t0 + K2_TILE) ? F + (size_t)indices[e] * (K / 2)
+                                             : F;  // dummy: see fp32 variant
     }
     const bool first_tile = (t0 == e0);
 
@@ -751,19 +758,25 @@ extern "C" __global__ void __launch_bounds__(BLOCK) k2_linesearch_bf16(
       }
       __syncthreads();
 
+      // slot-outer / edge-inner (see fp32 variant); m indexes 8-elem slots
+      for (int m = sub; m * 8 < clen; m += 4) {
+        uint4 braw[K2_TILE / NWAVE];
 #pragma unroll
-      for (int i = 0; i < K2_TILE / NWAVE; ++i) {
-        if (fvb[i]) {
-          const u32* __restrict__ fv = fvb[i] + c0 / 2;
-          float a = acc[i];
-          // m indexes 8-element slots; subs cover slots mod 4
-#pragma clang loop unroll_count(4)
-          for (int m = sub; m * 8 < clen; m += 4) {
-            const f32x8 b = ld8bf(fv + m * 4);
-            a = dot4(ld4(&cand[j][m * 8]), b.a, a);
-            a = dot4(ld4(&cand[j][m * 8 + 4]), b.b, a);
-          }
-          acc[i] = a;
+        for (int i = 0; i < K2_TILE / NWAVE; ++i)
+          braw[i] = *reinterpret_cast<const uint4*>(fvb[i] + c0 / 2 + m * 4);
+        asm volatile("" ::"v"(braw[0].x), "v"(braw[1].x), "v"(braw[2].x),
+                     "v"(braw[3].x), "v"(braw[4].x), "v"(braw[5].x),
+                     "v"(braw[6].x), "v"(braw[7].x));
+        const float4 cA = ld4(&cand[j][m * 8]);
+        const float4 cB = ld4(&cand[j][m * 8 + 4]);
+#pragma unroll
+        for (int i = 0; i < K2_TILE / NWAVE; ++i) {
+          f32x8 b;
+          b.a = float4{bf_lo(braw[i].x), bf_hi(braw[i].x), bf_lo(braw[i].y),
+                       bf_hi(braw[i].y)};
+          b.b = float4{bf_lo(braw[i].z), bf_hi(braw[i].z), bf_lo(braw[i].w),
+                       bf_hi(braw[i].w)};
+          acc[i] = dot4(cB, b.b, dot4(cA, b.a, acc[i]));
         }
       }
 
