@@ -117,3 +117,49 @@ def test_native_extension_is_loaded():
     assert hasattr(C, "edge_grad_llh")
     assert hasattr(C, "linesearch")
     assert hasattr(C, "llh_only")
+
+
+def _mkstate_dtype(g, k, dtype, seed=0, scale=0.4):
+    cfg = BigClamConfig(k=k, device="cuda", dtype=dtype)
+    shard = make_shard(g, 0, 1)
+    st = ShardState(shard, cfg, device=torch.device("cuda"))
+    rng = np.random.default_rng(seed)
+    F0 = (rng.random((g.num_nodes, k)) * scale).astype(np.float32)
+    st.set_local_F(torch.from_numpy(F0))
+    return cfg, st
+
+
+@pytest.mark.parametrize("k", [16, 500, 2000])
+def test_bf16_k1_matches_fp32(k):
+    g = rmat_graph(10, 8.0, seed=21)
+    cfg32, st32 = _mkstate_dtype(g, k, "fp32", seed=3)
+    cfgb, stb = _mkstate_dtype(g, k, "bf16", seed=3)
+    g32, l32 = st32.grad_llh()
+    gb, lb = stb.grad_llh()
+    kk = min(g32.shape[1], gb.shape[1])
+    # bf16 storage: ~1e-2 relative agreement expected
+    torch.testing.assert_close(
+        gb[:, :kk], g32[:, :kk], rtol=0.05, atol=0.05 * float(g32.abs().mean())
+    )
+    assert abs(lb.sum().item() - l32.sum().item()) < 2e-2 * abs(l32.sum().item())
+
+
+def test_bf16_sweep_converges():
+    g = rmat_graph(10, 6.0, seed=22)
+    cfg = BigClamConfig(k=96, device="cuda", dtype="bf16", max_sweeps=10, seed=4)
+    tr = Trainer(g, cfg, rank=0, world_size=1, device=torch.device("cuda"))
+    res = tr.fit(init="random")
+    assert np.isfinite(res.llh)
+    assert res.llh >= res.llh_history[0] - abs(res.llh_history[0]) * 0.02
+
+
+def test_bf16_k2_steps_mostly_agree_with_fp32():
+    g = rmat_graph(9, 8.0, seed=23)
+    cfg32, st32 = _mkstate_dtype(g, 64, "fp32", seed=5)
+    cfgb, stb = _mkstate_dtype(g, 64, "bf16", seed=5)
+    g32, l32 = st32.grad_llh()
+    gb, lb = stb.grad_llh()
+    b32 = st32.linesearch(g32, l32)
+    bb = stb.linesearch(gb, lb)
+    agree = (b32 == bb).float().mean().item()
+    assert agree > 0.9, agree
