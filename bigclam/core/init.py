@@ -30,33 +30,44 @@ from ..io.edgelist import Graph
 
 
 def conductances(graph: Graph) -> np.ndarray:
-    """Ego-net conductance per node (fp64 [N])."""
-    indptr, indices = graph.indptr, graph.indices
+    """Ego-net conductance per node (fp64 [N]).
+
+    Vectorized closed form (for the simple undirected canonical graph):
+      z_size(u)  = deg(u) + sum_{v in N(u)} deg(v)
+      inside(u)  = 2*deg(u) + sum_{v in N(u)} tri(u,v)    (tri = common nbrs)
+      cut        = z_size - inside;  vol_S = inside
+      vol_T      = total_degree - vol_S - 2*cut
+    with tri row-sums from (A@A) ∘ A (sparse).  Equivalent to the brute-force
+    2-hop pass (tests/test_init.py checks it against tests/oracle.py).
+    """
+    import scipy.sparse as sp
+
     n = graph.num_nodes
-    deg = np.diff(indptr).astype(np.int64)
+    deg = np.diff(graph.indptr).astype(np.int64)
     total_degree = int(deg.sum())
-    cond = np.zeros(n, dtype=np.float64)
-    for u in range(n):
-        nbrs = indices[indptr[u] : indptr[u + 1]]
-        ego = np.append(nbrs.astype(np.int64), u)
-        ego_sorted = np.sort(ego)
-        z_size = int(deg[ego].sum())
-        # count z entries inside ego
-        inside = 0
-        for m in ego:
-            nm = indices[indptr[m] : indptr[m + 1]]
-            pos = np.searchsorted(ego_sorted, nm)
-            pos[pos >= len(ego_sorted)] = len(ego_sorted) - 1
-            inside += int((ego_sorted[pos] == nm).sum())
-        cut = z_size - inside
-        vol_s = z_size - cut
-        vol_t = total_degree - vol_s - 2 * cut
-        if vol_s == 0:
-            cond[u] = 0.0
-        elif vol_t == 0:
-            cond[u] = 1.0
-        else:
-            cond[u] = cut / min(vol_s, vol_t)
+    A = sp.csr_matrix(
+        (
+            np.ones(len(graph.indices), dtype=np.float64),
+            graph.indices.astype(np.int64),
+            graph.indptr,
+        ),
+        shape=(n, n),
+    )
+    z_size = deg + (A @ deg.astype(np.float64)).astype(np.int64)
+    tri_row = np.asarray(((A @ A).multiply(A)).sum(axis=1)).ravel()
+    inside = 2 * deg + tri_row.astype(np.int64)
+    cut = z_size - inside
+    vol_s = inside
+    vol_t = total_degree - vol_s - 2 * cut
+    cond = np.where(
+        vol_s == 0,
+        0.0,
+        np.where(
+            vol_t == 0,
+            1.0,
+            cut / np.maximum(np.minimum(vol_s, vol_t), 1).astype(np.float64),
+        ),
+    )
     return cond
 
 
@@ -68,26 +79,38 @@ def conductance_ranking(
         cond = conductances(graph)
     indptr, indices = graph.indptr, graph.indices
     n = graph.num_nodes
-    picked = {}
-    for x in range(n):
-        nbrs = indices[indptr[x] : indptr[x + 1]].astype(np.int64)
-        if compat:
-            # reference behavior: lowest-id neighbor; isolated -> (x, 10.0)
-            if len(nbrs) == 0:
-                m, c = x, 10.0
-            else:
-                m = int(nbrs.min())
-                c = float(cond[m])
-        else:
-            members = np.append(nbrs, x)
-            cvals = cond[members]
-            # argmin by (conductance, id)
-            best = np.lexsort((members, cvals))[0]
-            m, c = int(members[best]), float(cvals[best])
-        if m not in picked or c < picked[m]:
-            picked[m] = c
-    items = sorted(picked.items(), key=lambda kv: (kv[1], kv[0]))
-    return np.array([m for m, _ in items], dtype=np.int64)
+    deg = np.diff(indptr)
+    dst = indices.astype(np.int64)
+    has_nbrs = deg > 0
+    if compat:
+        # reference behavior (codes/bigclamv3-7.scala:51): Scala tuple min
+        # orders by node id first -> each node picks its lowest-id neighbor.
+        # (The canonical graph has no isolated nodes — build_graph drops
+        # them — so the reference's (x, 10.0) sentinel branch cannot fire.)
+        picked = np.arange(n, dtype=np.int64)
+        starts = indptr[:-1].clip(max=max(len(dst) - 1, 0))
+        mins = np.minimum.reduceat(dst, starts) if len(dst) else picked
+        picked[has_nbrs] = mins[has_nbrs]
+    else:
+        # intended semantics: min-conductance member of the closed ego-net
+        # (ties -> lowest id).  Per row, the neighbor minimizing (cond, id)
+        # falls first after a lexsort keyed (src-major, cond, id).
+        src = np.repeat(np.arange(n, dtype=np.int64), deg)
+        o = np.lexsort((dst, cond[dst], src))
+        best_nbr = np.full(n, -1, dtype=np.int64)
+        best_nbr[has_nbrs] = dst[o][indptr[:-1][has_nbrs]]
+        picked = np.arange(n, dtype=np.int64)
+        take_nbr = has_nbrs & (
+            (cond[best_nbr.clip(min=0)] < cond[picked])
+            | (
+                (cond[best_nbr.clip(min=0)] == cond[picked])
+                & (best_nbr < picked)
+            )
+        )
+        picked[take_nbr] = best_nbr[take_nbr]
+    cands = np.unique(picked)
+    order = np.lexsort((cands, cond[cands]))
+    return cands[order]
 
 
 def seed_init_local_F(
