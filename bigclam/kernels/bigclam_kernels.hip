@@ -162,11 +162,13 @@ __global__ void __launch_bounds__(BLOCK) k1_grad_llh_t(
 #pragma clang loop unroll(disable)
   for (long long e = e0; e < e1; ++e) {
     const float* __restrict__ fv = F + (size_t)indices[e] * K;
+    float4 b[NSLOT];  // fv stays in registers between dot and axpy
     float part = 0.f;
 #pragma unroll
     for (int sl = 0; sl < NSLOT; ++sl) {
       const int k = tid * 4 + sl * (BLOCK * 4);
-      if (k < K) part = dot4(fu4[sl], ld4(fv + k), part);
+      b[sl] = (k < K) ? ld4(fv + k) : float4{0.f, 0.f, 0.f, 0.f};
+      part = dot4(fu4[sl], b[sl], part);
     }
     const float x = block_allreduce_sum(part, red);
     const float p = clamp_p(__expf(-x), min_p, max_p);
@@ -176,12 +178,11 @@ __global__ void __launch_bounds__(BLOCK) k1_grad_llh_t(
     for (int sl = 0; sl < NSLOT; ++sl) {
       const int k = tid * 4 + sl * (BLOCK * 4);
       if (k < K) {
-        const float4 b = ld4(fv + k);  // L1-hot: just read by this block
         float4 g = ld4(gacc + k);
-        g.x = fmaf(w, b.x, g.x);
-        g.y = fmaf(w, b.y, g.y);
-        g.z = fmaf(w, b.z, g.z);
-        g.w = fmaf(w, b.w, g.w);
+        g.x = fmaf(w, b[sl].x, g.x);
+        g.y = fmaf(w, b[sl].y, g.y);
+        g.z = fmaf(w, b[sl].z, g.z);
+        g.w = fmaf(w, b[sl].w, g.w);
         *reinterpret_cast<float4*>(gacc + k) = g;
       }
     }
@@ -340,21 +341,20 @@ extern "C" __global__ void __launch_bounds__(BLOCK) k4_llh_only(
 
 // ------------------------------------------------------------------- K2
 //
-// Block-per-node; K is processed in 512-element chunks.  Per (edge-tile,
-// chunk) the 16 candidate rows cand_j = clamp(Fu + s_j*grad, MIN_F, MAX_F)
-// are built ONCE into LDS (32 KB) and shared by the tile's edges, so the
-// edge loop touches only fv from global memory.  Lane mapping inside a
-// wave: lane = sub*16 + j (sub in 0..3, j = candidate) — each lane owns
-// ONE candidate and a quarter of the k-range, so per-edge dot partials
-// accumulate in a register across all chunks with NO per-chunk cross-lane
-// reduction; one 2-step shfl_xor folds the 4 sub-lanes when the edge
-// completes.  Handles any K (the v2 register-resident variant hit the
-// hipcc candidate-hoisting cliff: 64*NSLOT VGPRs of hoisted cand rows).
+// Block-per-node, wave-per-edge.  v3 design: NO candidate materialization.
+// Per edge element k all 16 trial dots accumulate directly in registers:
+//   acc_j += clamp(fu_k + s_j*g_k, MIN_F, MAX_F) * fv_k
+// (4 VALU per element per candidate, v_pk-packable), so fv streams from
+// HBM exactly once and fu/g are staged once per node in LDS.  The v2
+// kernel rebuilt all 16 candidate rows in LDS per 32-edge tile — for
+// mean-degree-6 graphs that build cost 3-10x the edge work itself
+// (123 ms/sweep measured vs the 48 ms v1; this version targets ~8 ms:
+// one 37 GB fv pass + ~600 G VALU ops at K=5000/com-Amazon).
+// After each edge, 16 wave-allreduces hand candidate j's dot to lane j,
+// which owns its transcendentals (exp/log1p parallel across 16 lanes).
 
-#define K2_CHUNK 512
-#define K2_TILE 32  // edges per tile; a wave owns K2_TILE/NWAVE = 8 of them
-
-extern "C" __global__ void __launch_bounds__(BLOCK) k2_linesearch(
+template <bool STAGED>
+__global__ void __launch_bounds__(BLOCK) k2_ls_v3(
     const float* __restrict__ F, const long long* __restrict__ indptr,
     const int* __restrict__ indices, const float* __restrict__ sumF,
     const float* __restrict__ grad, const double* __restrict__ llh,
@@ -367,128 +367,102 @@ extern "C" __global__ void __launch_bounds__(BLOCK) k2_linesearch(
   const int tid = threadIdx.x;
   const int lane = tid & (WAVE - 1);
   const int wid = tid >> 6;
-  const int sub = lane >> 4;   // 0..3: k-quarter inside the wave
-  const int j = lane & 15;     // candidate index this lane owns
 
-  // +4 pad: row stride 2064 B puts each j on its own bank quad
-  __shared__ __attribute__((aligned(16))) float cand[MAX_LS][K2_CHUNK + 4];
   __shared__ __attribute__((aligned(16))) float s_lad[MAX_LS];
   __shared__ __attribute__((aligned(16))) double acc_llh[NWAVE][MAX_LS];
   __shared__ __attribute__((aligned(16))) float acc_nt[NWAVE][MAX_LS];
   __shared__ __attribute__((aligned(16))) float red[NWAVE];
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  float* fu_s = reinterpret_cast<float*>(smem);  // K floats when STAGED
+  float* g_s = fu_s + K;                         // K floats when STAGED
 
   if (tid < MAX_LS) s_lad[tid] = (tid < n_ladder) ? ladder[tid] : 0.f;
+
+  const float* __restrict__ fu_g = F + (size_t)u * K;
+  const float* __restrict__ gu_g = grad + (size_t)u * K;
+  if (STAGED) {
+    for (int k = tid * 4; k < K; k += BLOCK * 4) {
+      *reinterpret_cast<float4*>(fu_s + k) = ld4(fu_g + k);
+      *reinterpret_cast<float4*>(g_s + k) = ld4(gu_g + k);
+    }
+  }
   __syncthreads();
-  const float sj = s_lad[j];  // this lane's candidate step
+  const float* __restrict__ fu = STAGED ? fu_s : fu_g;
+  const float* __restrict__ gu = STAGED ? g_s : gu_g;
 
-  const float* __restrict__ fu = F + (size_t)u * K;
-  const float* __restrict__ gu = grad + (size_t)u * K;
-
-  double llh_j = 0.0;  // edge-term sum for candidate j (lanes with sub==0)
-  float nt_j = 0.f;    // node term cand_j.(Fu - sumF), partial per lane
-  float p_gg = 0.f;    // grad.grad partial (thread-strided, j-independent)
-
-  for (long long t0 = e0; t0 < e1 || t0 == e0; t0 += K2_TILE) {
-    const int tlen = (int)((e1 - t0) < K2_TILE ? (e1 - t0) : K2_TILE);
-    float acc[K2_TILE / NWAVE];
-    // hoist the wave's edge row bases out of the chunk loop (the indices[]
-    // load otherwise serializes every chunk behind a vmcnt(0))
-    const float* fvb[K2_TILE / NWAVE];
+  float s[MAX_LS];
 #pragma unroll
-    for (int i = 0; i < K2_TILE / NWAVE; ++i) {
-      acc[i] = 0.f;
-      const long long e = t0 + (long long)(i * NWAVE + wid);
-      // invalid slots point at row 0: loads stay UNCONDITIONAL (an
-      // if-guard per edge makes hipcc emit exec-masked blocks that
-      // serialize every load behind vmcnt(0)); their acc is never read
-      // (the fold re-checks e < tlen).
-      fvb[i] = (e < e1 && e < t0 + K2_TILE) ? F + (size_t)indices[e] * K : F;
-    }
-    const bool first_tile = (t0 == e0);
+  for (int j = 0; j < MAX_LS; ++j) s[j] = s_lad[j];
 
-    for (int c0 = 0; c0 < K; c0 += K2_CHUNK) {
-      const int clen = (K - c0) < K2_CHUNK ? (K - c0) : K2_CHUNK;
-      __syncthreads();  // previous chunk fully consumed
-      // build the candidate chunk (fu/g chunk stays L1-hot across the 16 j)
-      for (int jj = 0; jj < MAX_LS; ++jj) {
-        const float sjj = s_lad[jj];
-        for (int kk = tid * 4; kk < clen; kk += BLOCK * 4) {
-          const float4 a = ld4(fu + c0 + kk);
-          const float4 g = ld4(gu + c0 + kk);
-          *reinterpret_cast<float4*>(&cand[jj][kk]) =
-              float4{fminf(fmaxf(fmaf(sjj, g.x, a.x), min_f), max_f),
-                     fminf(fmaxf(fmaf(sjj, g.y, a.y), min_f), max_f),
-                     fminf(fmaxf(fmaf(sjj, g.z, a.z), min_f), max_f),
-                     fminf(fmaxf(fmaf(sjj, g.w, a.w), min_f), max_f)};
-        }
-      }
-      __syncthreads();
+  double llh_lane = 0.0;  // lane j (j<16) accumulates candidate j edge terms
 
-      // edges of this tile over this chunk; acc[i] persists across chunks
-      // (full unroll keeps acc[] in registers — runtime indexing would
-      // spill it to scratch)
-      // slot-outer / edge-inner: one LDS cand read serves all 8 edges and
-      // their 8 independent fv loads issue back-to-back (deep MLP) — an
-      // edge-outer nest made hipcc serialize every load behind vmcnt(0)
-      for (int m = sub; m * 4 < clen; m += 4) {
-        // issue all 8 independent loads, then a value barrier: the
-        // compiler emits ONE vmcnt drain for the batch instead of pairing
-        // loads 2-deep behind pk_fma SLP marshalling
-        float4 b[K2_TILE / NWAVE];
+  for (long long e = e0 + wid; e < e1; e += NWAVE) {
+    const float* __restrict__ fv = F + (size_t)indices[e] * K;
+    float acc[MAX_LS];
 #pragma unroll
-        for (int i = 0; i < K2_TILE / NWAVE; ++i)
-          b[i] = ld4(fvb[i] + c0 + m * 4);
-        asm volatile("" ::"v"(b[0].x), "v"(b[1].x), "v"(b[2].x), "v"(b[3].x),
-                     "v"(b[4].x), "v"(b[5].x), "v"(b[6].x), "v"(b[7].x));
-        const float4 c = ld4(&cand[j][m * 4]);
+    for (int j = 0; j < MAX_LS; ++j) acc[j] = 0.f;
+    for (int k = lane * 4; k < K; k += WAVE * 4) {
+      const float4 b = ld4(fv + k);
+      const float4 a = ld4(fu + k);
+      const float4 g = ld4(gu + k);
 #pragma unroll
-        for (int i = 0; i < K2_TILE / NWAVE; ++i)
-          acc[i] = dot4(c, b[i], acc[i]);
-      }
-
-      // node term for this chunk: waves split the slot range; once per node
-      if (first_tile) {
-        const int m0 = sub + 4 * wid;  // 16 (sub, wave) ways cover slots mod 16
-        for (int m = m0; m * 4 < clen; m += 16) {
-          const float4 c = ld4(&cand[j][m * 4]);
-          const float4 a = ld4(fu + c0 + m * 4);
-          const float4 sf = ld4(sumF + c0 + m * 4);
-          nt_j = fmaf(c.x, a.x - sf.x, nt_j);
-          nt_j = fmaf(c.y, a.y - sf.y, nt_j);
-          nt_j = fmaf(c.z, a.z - sf.z, nt_j);
-          nt_j = fmaf(c.w, a.w - sf.w, nt_j);
-        }
-      }
-      if (first_tile) {
-        for (int kk = tid * 4; kk < clen; kk += BLOCK * 4) {
-          const float4 g = ld4(gu + c0 + kk);
-          p_gg = dot4(g, g, p_gg);
-        }
+      for (int j = 0; j < MAX_LS; ++j) {
+        const float sj = s[j];
+        const float c0 = fminf(fmaxf(fmaf(sj, g.x, a.x), min_f), max_f);
+        const float c1 = fminf(fmaxf(fmaf(sj, g.y, a.y), min_f), max_f);
+        const float c2 = fminf(fmaxf(fmaf(sj, g.z, a.z), min_f), max_f);
+        const float c3 = fminf(fmaxf(fmaf(sj, g.w, a.w), min_f), max_f);
+        float t = acc[j];
+        t = fmaf(c0, b.x, t);
+        t = fmaf(c1, b.y, t);
+        t = fmaf(c2, b.z, t);
+        t = fmaf(c3, b.w, t);
+        acc[j] = t;
       }
     }
-
-    // fold the 4 sub-lanes of each edge; lane sub==0 owns the result
 #pragma unroll
-    for (int i = 0; i < K2_TILE / NWAVE; ++i) {
-      float x = acc[i];
-      x += __shfl_xor(x, 16, WAVE);
-      x += __shfl_xor(x, 32, WAVE);
-      const long long e = t0 + (long long)(i * NWAVE + wid);
-      if (sub == 0 && e < t0 + tlen) {
+    for (int j = 0; j < MAX_LS; ++j) {
+      const float x = wave_allreduce_sum(acc[j]);
+      if (lane == j) {
         const float p = clamp_p(__expf(-x), min_p, max_p);
-        llh_j += (double)log1pf(-p) + (double)x;
+        llh_lane += (double)log1pf(-p) + (double)x;
       }
     }
-    if (e1 == e0) break;  // degree-0 guard (loop ran once for node terms)
   }
 
-  // fold node-term sub-lanes, stash per-wave results, combine across waves
-  nt_j += __shfl_xor(nt_j, 16, WAVE);
-  nt_j += __shfl_xor(nt_j, 32, WAVE);
-  if (sub == 0) {
-    acc_llh[wid][j] = llh_j;
-    acc_nt[wid][j] = nt_j;
+  // node terms cand_j.(Fu - sumF) + grad.grad, block-strided k, once
+  float accn[MAX_LS];
+#pragma unroll
+  for (int j = 0; j < MAX_LS; ++j) accn[j] = 0.f;
+  float p_gg = 0.f;
+  for (int k = tid * 4; k < K; k += BLOCK * 4) {
+    const float4 a = ld4(fu + k);
+    const float4 g = ld4(gu + k);
+    const float4 sf = ld4(sumF + k);
+    const float d0 = a.x - sf.x, d1 = a.y - sf.y;
+    const float d2 = a.z - sf.z, d3 = a.w - sf.w;
+    p_gg = dot4(g, g, p_gg);
+#pragma unroll
+    for (int j = 0; j < MAX_LS; ++j) {
+      const float sj = s[j];
+      const float c0 = fminf(fmaxf(fmaf(sj, g.x, a.x), min_f), max_f);
+      const float c1 = fminf(fmaxf(fmaf(sj, g.y, a.y), min_f), max_f);
+      const float c2 = fminf(fmaxf(fmaf(sj, g.z, a.z), min_f), max_f);
+      const float c3 = fminf(fmaxf(fmaf(sj, g.w, a.w), min_f), max_f);
+      float t = accn[j];
+      t = fmaf(c0, d0, t);
+      t = fmaf(c1, d1, t);
+      t = fmaf(c2, d2, t);
+      t = fmaf(c3, d3, t);
+      accn[j] = t;
+    }
   }
+#pragma unroll
+  for (int j = 0; j < MAX_LS; ++j) {
+    const float x = wave_allreduce_sum(accn[j]);
+    if (lane == j) acc_nt[wid][j] = x;
+  }
+  if (lane < MAX_LS) acc_llh[wid][lane] = llh_lane;
   const float gg = block_allreduce_sum(p_gg, red);
 
   if (tid == 0) {
@@ -576,11 +550,19 @@ __global__ void __launch_bounds__(BLOCK) k1_grad_llh_bf16_t(
 #pragma clang loop unroll(disable)
   for (long long e = e0; e < e1; ++e) {
     const u32* __restrict__ fv = F + (size_t)indices[e] * (K / 2);
+    uint4 braw[NSLOT];  // raw bf16 fv kept in registers between dot and axpy
     float part = 0.f;
 #pragma unroll
     for (int sl = 0; sl < NSLOT; ++sl) {
       const int k = tid * 8 + sl * (BLOCK * 8);
-      if (k < K) part = dot8(fu8[sl], ld8bf(fv + k / 2), part);
+      braw[sl] = (k < K) ? *reinterpret_cast<const uint4*>(fv + k / 2)
+                         : uint4{0u, 0u, 0u, 0u};
+      f32x8 b;
+      b.a = float4{bf_lo(braw[sl].x), bf_hi(braw[sl].x), bf_lo(braw[sl].y),
+                   bf_hi(braw[sl].y)};
+      b.b = float4{bf_lo(braw[sl].z), bf_hi(braw[sl].z), bf_lo(braw[sl].w),
+                   bf_hi(braw[sl].w)};
+      part = dot8(fu8[sl], b, part);
     }
     const float x = block_allreduce_sum(part, red);
     const float p = clamp_p(__expf(-x), min_p, max_p);
@@ -590,7 +572,11 @@ __global__ void __launch_bounds__(BLOCK) k1_grad_llh_bf16_t(
     for (int sl = 0; sl < NSLOT; ++sl) {
       const int k = tid * 8 + sl * (BLOCK * 8);
       if (k < K) {
-        const f32x8 b = ld8bf(fv + k / 2);  // L1-hot
+        f32x8 b;
+        b.a = float4{bf_lo(braw[sl].x), bf_hi(braw[sl].x), bf_lo(braw[sl].y),
+                     bf_hi(braw[sl].y)};
+        b.b = float4{bf_lo(braw[sl].z), bf_hi(braw[sl].z), bf_lo(braw[sl].w),
+                     bf_hi(braw[sl].w)};
         float4 g0 = ld4(gacc + k);
         float4 g1 = ld4(gacc + k + 4);
         g0.x = fmaf(w, b.a.x, g0.x);
@@ -689,7 +675,10 @@ extern "C" __global__ void __launch_bounds__(BLOCK) k4_llh_only_bf16(
   }
 }
 
-extern "C" __global__ void __launch_bounds__(BLOCK) k2_linesearch_bf16(
+// bf16 K2, same v3 structure as k2_ls_v3: fu staged raw bf16 (K*2B LDS),
+// grad staged fp32 (K*4B LDS) — 6 B/element fits K<=25000 in 160 KB LDS.
+template <bool STAGED>
+__global__ void __launch_bounds__(BLOCK) k2_ls_v3_bf16(
     const u32* __restrict__ F, const long long* __restrict__ indptr,
     const int* __restrict__ indices, const float* __restrict__ sumF,
     const float* __restrict__ grad, const double* __restrict__ llh,
@@ -702,128 +691,112 @@ extern "C" __global__ void __launch_bounds__(BLOCK) k2_linesearch_bf16(
   const int tid = threadIdx.x;
   const int lane = tid & (WAVE - 1);
   const int wid = tid >> 6;
-  const int sub = lane >> 4;
-  const int j = lane & 15;
 
-  // +4 pad floats: row stride 2064 B spreads the 16 j rows over bank quads
-  __shared__ __attribute__((aligned(16))) float cand[MAX_LS][K2_CHUNK + 4];
   __shared__ __attribute__((aligned(16))) float s_lad[MAX_LS];
   __shared__ __attribute__((aligned(16))) double acc_llh[NWAVE][MAX_LS];
   __shared__ __attribute__((aligned(16))) float acc_nt[NWAVE][MAX_LS];
   __shared__ __attribute__((aligned(16))) float red[NWAVE];
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  float* g_s = reinterpret_cast<float*>(smem);              // K floats
+  u32* fu_s = reinterpret_cast<u32*>(smem + (size_t)K * 4); // K/2 u32
 
   if (tid < MAX_LS) s_lad[tid] = (tid < n_ladder) ? ladder[tid] : 0.f;
+
+  const u32* __restrict__ fu_g = F + (size_t)u * (K / 2);
+  const float* __restrict__ gu_g = grad + (size_t)u * K;
+  if (STAGED) {
+    for (int k = tid * 8; k < K; k += BLOCK * 8) {
+      *reinterpret_cast<uint4*>(fu_s + k / 2) =
+          *reinterpret_cast<const uint4*>(fu_g + k / 2);
+      *reinterpret_cast<float4*>(g_s + k) = ld4(gu_g + k);
+      *reinterpret_cast<float4*>(g_s + k + 4) = ld4(gu_g + k + 4);
+    }
+  }
   __syncthreads();
+  const u32* __restrict__ fu = STAGED ? fu_s : fu_g;
+  const float* __restrict__ gu = STAGED ? g_s : gu_g;
 
-  const u32* __restrict__ fub = F + (size_t)u * (K / 2);
-  const float* __restrict__ gu = grad + (size_t)u * K;
-
-  double llh_j = 0.0;
-  float nt_j = 0.f;
-  float p_gg = 0.f;
-
-  for (long long t0 = e0; t0 < e1 || t0 == e0; t0 += K2_TILE) {
-    const int tlen = (int)((e1 - t0) < K2_TILE ? (e1 - t0) : K2_TILE);
-    float acc[K2_TILE / NWAVE];
-    const u32* fvb[K2_TILE / NWAVE];
+  float s[MAX_LS];
 #pragma unroll
-    for (int i = 0; i < K2_TILE / NWAVE; ++i) {
-      acc[i] = 0.f;
-      const long long e = t0 + (long long)(i * NWAVE + wid);
-      fvb[i] = (e < e1 && e < t0 + K2_TILE) ? F + (size_t)indices[e] * (K / 2)
-                                             : F;  // dummy: see fp32 variant
-    }
-    const bool first_tile = (t0 == e0);
+  for (int j = 0; j < MAX_LS; ++j) s[j] = s_lad[j];
 
-    for (int c0 = 0; c0 < K; c0 += K2_CHUNK) {
-      const int clen = (K - c0) < K2_CHUNK ? (K - c0) : K2_CHUNK;
-      __syncthreads();
-      for (int jj = 0; jj < MAX_LS; ++jj) {
-        const float sjj = s_lad[jj];
-        for (int kk = tid * 8; kk < clen; kk += BLOCK * 8) {
-          const f32x8 a = ld8bf(fub + (c0 + kk) / 2);
-          const float4 g0 = ld4(gu + c0 + kk);
-          const float4 g1 = ld4(gu + c0 + kk + 4);
-          *reinterpret_cast<float4*>(&cand[jj][kk]) =
-              float4{fminf(fmaxf(fmaf(sjj, g0.x, a.a.x), min_f), max_f),
-                     fminf(fmaxf(fmaf(sjj, g0.y, a.a.y), min_f), max_f),
-                     fminf(fmaxf(fmaf(sjj, g0.z, a.a.z), min_f), max_f),
-                     fminf(fmaxf(fmaf(sjj, g0.w, a.a.w), min_f), max_f)};
-          *reinterpret_cast<float4*>(&cand[jj][kk + 4]) =
-              float4{fminf(fmaxf(fmaf(sjj, g1.x, a.b.x), min_f), max_f),
-                     fminf(fmaxf(fmaf(sjj, g1.y, a.b.y), min_f), max_f),
-                     fminf(fmaxf(fmaf(sjj, g1.z, a.b.z), min_f), max_f),
-                     fminf(fmaxf(fmaf(sjj, g1.w, a.b.w), min_f), max_f)};
-        }
-      }
-      __syncthreads();
+  double llh_lane = 0.0;
 
-      // slot-outer / edge-inner (see fp32 variant); m indexes 8-elem slots
-      for (int m = sub; m * 8 < clen; m += 4) {
-        uint4 braw[K2_TILE / NWAVE];
+  for (long long e = e0 + wid; e < e1; e += NWAVE) {
+    const u32* __restrict__ fv = F + (size_t)indices[e] * (K / 2);
+    float acc[MAX_LS];
 #pragma unroll
-        for (int i = 0; i < K2_TILE / NWAVE; ++i)
-          braw[i] = *reinterpret_cast<const uint4*>(fvb[i] + c0 / 2 + m * 4);
-        asm volatile("" ::"v"(braw[0].x), "v"(braw[1].x), "v"(braw[2].x),
-                     "v"(braw[3].x), "v"(braw[4].x), "v"(braw[5].x),
-                     "v"(braw[6].x), "v"(braw[7].x));
-        const float4 cA = ld4(&cand[j][m * 8]);
-        const float4 cB = ld4(&cand[j][m * 8 + 4]);
+    for (int j = 0; j < MAX_LS; ++j) acc[j] = 0.f;
+    for (int k = lane * 8; k < K; k += WAVE * 8) {
+      const f32x8 b = ld8bf(fv + k / 2);
+      const f32x8 a = ld8bf(fu + k / 2);
+      const float4 g0 = ld4(gu + k);
+      const float4 g1 = ld4(gu + k + 4);
 #pragma unroll
-        for (int i = 0; i < K2_TILE / NWAVE; ++i) {
-          f32x8 b;
-          b.a = float4{bf_lo(braw[i].x), bf_hi(braw[i].x), bf_lo(braw[i].y),
-                       bf_hi(braw[i].y)};
-          b.b = float4{bf_lo(braw[i].z), bf_hi(braw[i].z), bf_lo(braw[i].w),
-                       bf_hi(braw[i].w)};
-          acc[i] = dot4(cB, b.b, dot4(cA, b.a, acc[i]));
-        }
-      }
-
-      if (first_tile) {
-        const int m0 = sub + 4 * wid;
-        for (int m = m0; m * 8 < clen; m += 16) {
-          const float4 c0v = ld4(&cand[j][m * 8]);
-          const float4 c1v = ld4(&cand[j][m * 8 + 4]);
-          const f32x8 a = ld8bf(fub + (c0 + m * 8) / 2);
-          const float4 s0 = ld4(sumF + c0 + m * 8);
-          const float4 s1 = ld4(sumF + c0 + m * 8 + 4);
-          nt_j = fmaf(c0v.x, a.a.x - s0.x, nt_j);
-          nt_j = fmaf(c0v.y, a.a.y - s0.y, nt_j);
-          nt_j = fmaf(c0v.z, a.a.z - s0.z, nt_j);
-          nt_j = fmaf(c0v.w, a.a.w - s0.w, nt_j);
-          nt_j = fmaf(c1v.x, a.b.x - s1.x, nt_j);
-          nt_j = fmaf(c1v.y, a.b.y - s1.y, nt_j);
-          nt_j = fmaf(c1v.z, a.b.z - s1.z, nt_j);
-          nt_j = fmaf(c1v.w, a.b.w - s1.w, nt_j);
-        }
-        for (int kk = tid * 4; kk < clen; kk += BLOCK * 4) {
-          const float4 g = ld4(gu + c0 + kk);
-          p_gg = dot4(g, g, p_gg);
-        }
+      for (int j = 0; j < MAX_LS; ++j) {
+        const float sj = s[j];
+        float t = acc[j];
+        t = fmaf(fminf(fmaxf(fmaf(sj, g0.x, a.a.x), min_f), max_f), b.a.x, t);
+        t = fmaf(fminf(fmaxf(fmaf(sj, g0.y, a.a.y), min_f), max_f), b.a.y, t);
+        t = fmaf(fminf(fmaxf(fmaf(sj, g0.z, a.a.z), min_f), max_f), b.a.z, t);
+        t = fmaf(fminf(fmaxf(fmaf(sj, g0.w, a.a.w), min_f), max_f), b.a.w, t);
+        t = fmaf(fminf(fmaxf(fmaf(sj, g1.x, a.b.x), min_f), max_f), b.b.x, t);
+        t = fmaf(fminf(fmaxf(fmaf(sj, g1.y, a.b.y), min_f), max_f), b.b.y, t);
+        t = fmaf(fminf(fmaxf(fmaf(sj, g1.z, a.b.z), min_f), max_f), b.b.z, t);
+        t = fmaf(fminf(fmaxf(fmaf(sj, g1.w, a.b.w), min_f), max_f), b.b.w, t);
+        acc[j] = t;
       }
     }
-
 #pragma unroll
-    for (int i = 0; i < K2_TILE / NWAVE; ++i) {
-      float x = acc[i];
-      x += __shfl_xor(x, 16, WAVE);
-      x += __shfl_xor(x, 32, WAVE);
-      const long long e = t0 + (long long)(i * NWAVE + wid);
-      if (sub == 0 && e < t0 + tlen) {
+    for (int j = 0; j < MAX_LS; ++j) {
+      const float x = wave_allreduce_sum(acc[j]);
+      if (lane == j) {
         const float p = clamp_p(__expf(-x), min_p, max_p);
-        llh_j += (double)log1pf(-p) + (double)x;
+        llh_lane += (double)log1pf(-p) + (double)x;
       }
     }
-    if (e1 == e0) break;
   }
 
-  nt_j += __shfl_xor(nt_j, 16, WAVE);
-  nt_j += __shfl_xor(nt_j, 32, WAVE);
-  if (sub == 0) {
-    acc_llh[wid][j] = llh_j;
-    acc_nt[wid][j] = nt_j;
+  float accn[MAX_LS];
+#pragma unroll
+  for (int j = 0; j < MAX_LS; ++j) accn[j] = 0.f;
+  float p_gg = 0.f;
+  for (int k = tid * 8; k < K; k += BLOCK * 8) {
+    const f32x8 a = ld8bf(fu + k / 2);
+    const float4 g0 = ld4(gu + k);
+    const float4 g1 = ld4(gu + k + 4);
+    const float4 s0 = ld4(sumF + k);
+    const float4 s1 = ld4(sumF + k + 4);
+    p_gg = dot4(g1, g1, dot4(g0, g0, p_gg));
+#pragma unroll
+    for (int j = 0; j < MAX_LS; ++j) {
+      const float sj = s[j];
+      float t = accn[j];
+      t = fmaf(fminf(fmaxf(fmaf(sj, g0.x, a.a.x), min_f), max_f),
+               a.a.x - s0.x, t);
+      t = fmaf(fminf(fmaxf(fmaf(sj, g0.y, a.a.y), min_f), max_f),
+               a.a.y - s0.y, t);
+      t = fmaf(fminf(fmaxf(fmaf(sj, g0.z, a.a.z), min_f), max_f),
+               a.a.z - s0.z, t);
+      t = fmaf(fminf(fmaxf(fmaf(sj, g0.w, a.a.w), min_f), max_f),
+               a.a.w - s0.w, t);
+      t = fmaf(fminf(fmaxf(fmaf(sj, g1.x, a.b.x), min_f), max_f),
+               a.b.x - s1.x, t);
+      t = fmaf(fminf(fmaxf(fmaf(sj, g1.y, a.b.y), min_f), max_f),
+               a.b.y - s1.y, t);
+      t = fmaf(fminf(fmaxf(fmaf(sj, g1.z, a.b.z), min_f), max_f),
+               a.b.z - s1.z, t);
+      t = fmaf(fminf(fmaxf(fmaf(sj, g1.w, a.b.w), min_f), max_f),
+               a.b.w - s1.w, t);
+      accn[j] = t;
+    }
   }
+#pragma unroll
+  for (int j = 0; j < MAX_LS; ++j) {
+    const float x = wave_allreduce_sum(accn[j]);
+    if (lane == j) acc_nt[wid][j] = x;
+  }
+  if (lane < MAX_LS) acc_llh[wid][lane] = llh_lane;
   const float gg = block_allreduce_sum(p_gg, red);
 
   if (tid == 0) {
@@ -946,10 +919,20 @@ extern "C" void launch_k2_bf16(const void* F, const long long* indptr,
                                float min_f, float max_f, hipStream_t stream) {
   if (n_local == 0) return;
   if (n_ladder > 16) throw std::runtime_error("ladder length > 16 unsupported");
-  hipLaunchKernelGGL(k2_linesearch_bf16, dim3(n_local), dim3(256), 0, stream,
-                     reinterpret_cast<const u32*>(F), indptr, indices, sumF,
-                     grad, llh, order, ladder, best, n_local, K, n_ladder,
-                     alpha, min_p, max_p, min_f, max_f);
+  const u32* Fb = reinterpret_cast<const u32*>(F);
+  const size_t lds = (size_t)K * 6;  // g fp32 + fu raw bf16
+  if (lds + 2048 <= 160 * 1024) {
+    allow_large_lds((const void*)&k2_ls_v3_bf16<true>, lds);
+    hipLaunchKernelGGL((k2_ls_v3_bf16<true>), dim3(n_local), dim3(256), lds,
+                       stream, Fb, indptr, indices, sumF, grad, llh, order,
+                       ladder, best, n_local, K, n_ladder, alpha, min_p,
+                       max_p, min_f, max_f);
+  } else {
+    hipLaunchKernelGGL((k2_ls_v3_bf16<false>), dim3(n_local), dim3(256), 0,
+                       stream, Fb, indptr, indices, sumF, grad, llh, order,
+                       ladder, best, n_local, K, n_ladder, alpha, min_p,
+                       max_p, min_f, max_f);
+  }
   HIP_CHECK(hipGetLastError());
 }
 
@@ -1015,9 +998,19 @@ extern "C" void launch_k2(const float* F, const long long* indptr,
                           hipStream_t stream) {
   if (n_local == 0) return;
   if (n_ladder > 16) throw std::runtime_error("ladder length > 16 unsupported");
-  hipLaunchKernelGGL(k2_linesearch, dim3(n_local), dim3(256), 0, stream, F,
-                     indptr, indices, sumF, grad, llh, order, ladder, best,
-                     n_local, K, n_ladder, alpha, min_p, max_p, min_f, max_f);
+  const size_t lds = (size_t)K * 8;  // fu + g staged fp32
+  if (lds + 2048 <= 160 * 1024) {
+    allow_large_lds((const void*)&k2_ls_v3<true>, lds);
+    hipLaunchKernelGGL((k2_ls_v3<true>), dim3(n_local), dim3(256), lds,
+                       stream, F, indptr, indices, sumF, grad, llh, order,
+                       ladder, best, n_local, K, n_ladder, alpha, min_p,
+                       max_p, min_f, max_f);
+  } else {
+    hipLaunchKernelGGL((k2_ls_v3<false>), dim3(n_local), dim3(256), 0, stream,
+                       F, indptr, indices, sumF, grad, llh, order, ladder,
+                       best, n_local, K, n_ladder, alpha, min_p, max_p, min_f,
+                       max_f);
+  }
   HIP_CHECK(hipGetLastError());
 }
 
