@@ -81,6 +81,20 @@ __device__ __forceinline__ float4 ld4(const float* p) {
   return *reinterpret_cast<const float4*>(p);
 }
 
+// 2-wide fp32 vectors lowered to CDNA packed-math (v_pk_fma_f32 /
+// v_pk_max_f32 / v_pk_min_f32): one instruction per pair, doubling VALU
+// throughput in the K2 inner loop.
+typedef float v2f __attribute__((ext_vector_type(2)));
+
+__device__ __forceinline__ v2f pk_clamp_fma(v2f s, v2f g, v2f a, v2f lo,
+                                            v2f hi) {
+  // v_pk_fma_f32 + two v_med3_f32 (med3 = whole clamp in ONE VALU op;
+  // no packed min/max f32 exists on CDNA4, med3 halves the clamp cost)
+  const v2f t = __builtin_elementwise_fma(s, g, a);
+  return v2f{__builtin_amdgcn_fmed3f(t.x, lo.x, hi.x),
+             __builtin_amdgcn_fmed3f(t.y, lo.y, hi.y)};
+}
+
 
 // ------------------------------------------------------- bf16 storage path
 //
@@ -95,6 +109,10 @@ __device__ __forceinline__ float bf_lo(u32 v) {
 }
 __device__ __forceinline__ float bf_hi(u32 v) {
   return __uint_as_float(v & 0xffff0000u);
+}
+
+__device__ __forceinline__ v2f bf2(u32 v) {  // one packed bf16 pair -> v2f
+  return v2f{bf_lo(v), bf_hi(v)};
 }
 
 struct f32x8 {
@@ -354,7 +372,7 @@ extern "C" __global__ void __launch_bounds__(BLOCK) k4_llh_only(
 // which owns its transcendentals (exp/log1p parallel across 16 lanes).
 
 template <bool STAGED>
-__global__ void __launch_bounds__(BLOCK) k2_ls_v3(
+__global__ void __launch_bounds__(BLOCK, 3) k2_ls_v3(
     const float* __restrict__ F, const long long* __restrict__ indptr,
     const int* __restrict__ indices, const float* __restrict__ sumF,
     const float* __restrict__ grad, const double* __restrict__ llh,
@@ -395,34 +413,51 @@ __global__ void __launch_bounds__(BLOCK) k2_ls_v3(
   for (int j = 0; j < MAX_LS; ++j) s[j] = s_lad[j];
 
   double llh_lane = 0.0;  // lane j (j<16) accumulates candidate j edge terms
+  const v2f lo2 = {min_f, min_f}, hi2 = {max_f, max_f};
 
   for (long long e = e0 + wid; e < e1; e += NWAVE) {
     const float* __restrict__ fv = F + (size_t)indices[e] * K;
-    float acc[MAX_LS];
+    v2f acc[MAX_LS];
 #pragma unroll
-    for (int j = 0; j < MAX_LS; ++j) acc[j] = 0.f;
-    for (int k = lane * 4; k < K; k += WAVE * 4) {
-      const float4 b = ld4(fv + k);
-      const float4 a = ld4(fu + k);
-      const float4 g = ld4(gu + k);
+    for (int j = 0; j < MAX_LS; ++j) acc[j] = v2f{0.f, 0.f};
+    // 4-deep k-unroll: the 4 independent fv loads issue back-to-back
+    // BEFORE the 4 compute bodies, so each wave keeps 4 HBM loads in
+    // flight across the ~165-cycle body (1 load in flight measured
+    // 1.9 TB/s; latency-bound, not bandwidth-bound).
+    constexpr int U = 4;
+    constexpr int KSTR = WAVE * 4;
+    for (int k = lane * 4; k < K; k += U * KSTR) {
+      float4 b[U];  // ONLY the global fv loads are grouped: LDS fu/g reads
+                    // are short-latency and prefetching them cost 32 VGPRs
+                    // (184 total -> 2 waves/SIMD, a 45% regression)
 #pragma unroll
-      for (int j = 0; j < MAX_LS; ++j) {
-        const float sj = s[j];
-        const float c0 = fminf(fmaxf(fmaf(sj, g.x, a.x), min_f), max_f);
-        const float c1 = fminf(fmaxf(fmaf(sj, g.y, a.y), min_f), max_f);
-        const float c2 = fminf(fmaxf(fmaf(sj, g.z, a.z), min_f), max_f);
-        const float c3 = fminf(fmaxf(fmaf(sj, g.w, a.w), min_f), max_f);
-        float t = acc[j];
-        t = fmaf(c0, b.x, t);
-        t = fmaf(c1, b.y, t);
-        t = fmaf(c2, b.z, t);
-        t = fmaf(c3, b.w, t);
-        acc[j] = t;
+      for (int t = 0; t < U; ++t) {
+        const int kk = k + t * KSTR;
+        b[t] = ld4(fv + (kk < K ? kk : (K - 4)));  // clamped: always valid
+      }
+#pragma unroll
+      for (int t = 0; t < U; ++t) {
+        if (k + t * KSTR >= K) break;
+        const float4 a4 = ld4(fu + k + t * KSTR);
+        const float4 g4 = ld4(gu + k + t * KSTR);
+        const v2f b0 = {b[t].x, b[t].y}, b1 = {b[t].z, b[t].w};
+        const v2f a0 = {a4.x, a4.y}, a1 = {a4.z, a4.w};
+        const v2f g0 = {g4.x, g4.y}, g1 = {g4.z, g4.w};
+#pragma unroll
+        for (int j = 0; j < MAX_LS; ++j) {
+          const v2f sj = {s[j], s[j]};
+          v2f t2 = acc[j];
+          t2 = __builtin_elementwise_fma(pk_clamp_fma(sj, g0, a0, lo2, hi2),
+                                         b0, t2);
+          t2 = __builtin_elementwise_fma(pk_clamp_fma(sj, g1, a1, lo2, hi2),
+                                         b1, t2);
+          acc[j] = t2;
+        }
       }
     }
 #pragma unroll
     for (int j = 0; j < MAX_LS; ++j) {
-      const float x = wave_allreduce_sum(acc[j]);
+      const float x = wave_allreduce_sum(acc[j].x + acc[j].y);
       if (lane == j) {
         const float p = clamp_p(__expf(-x), min_p, max_p);
         llh_lane += (double)log1pf(-p) + (double)x;
@@ -431,35 +466,33 @@ __global__ void __launch_bounds__(BLOCK) k2_ls_v3(
   }
 
   // node terms cand_j.(Fu - sumF) + grad.grad, block-strided k, once
-  float accn[MAX_LS];
+  v2f accn[MAX_LS];
 #pragma unroll
-  for (int j = 0; j < MAX_LS; ++j) accn[j] = 0.f;
+  for (int j = 0; j < MAX_LS; ++j) accn[j] = v2f{0.f, 0.f};
   float p_gg = 0.f;
   for (int k = tid * 4; k < K; k += BLOCK * 4) {
     const float4 a = ld4(fu + k);
     const float4 g = ld4(gu + k);
     const float4 sf = ld4(sumF + k);
-    const float d0 = a.x - sf.x, d1 = a.y - sf.y;
-    const float d2 = a.z - sf.z, d3 = a.w - sf.w;
+    const v2f a0 = {a.x, a.y}, a1 = {a.z, a.w};
+    const v2f g0 = {g.x, g.y}, g1 = {g.z, g.w};
+    const v2f d0 = {a.x - sf.x, a.y - sf.y};
+    const v2f d1 = {a.z - sf.z, a.w - sf.w};
     p_gg = dot4(g, g, p_gg);
 #pragma unroll
     for (int j = 0; j < MAX_LS; ++j) {
-      const float sj = s[j];
-      const float c0 = fminf(fmaxf(fmaf(sj, g.x, a.x), min_f), max_f);
-      const float c1 = fminf(fmaxf(fmaf(sj, g.y, a.y), min_f), max_f);
-      const float c2 = fminf(fmaxf(fmaf(sj, g.z, a.z), min_f), max_f);
-      const float c3 = fminf(fmaxf(fmaf(sj, g.w, a.w), min_f), max_f);
-      float t = accn[j];
-      t = fmaf(c0, d0, t);
-      t = fmaf(c1, d1, t);
-      t = fmaf(c2, d2, t);
-      t = fmaf(c3, d3, t);
+      const v2f sj = {s[j], s[j]};
+      v2f t = accn[j];
+      t = __builtin_elementwise_fma(pk_clamp_fma(sj, g0, a0, lo2, hi2), d0,
+                                    t);
+      t = __builtin_elementwise_fma(pk_clamp_fma(sj, g1, a1, lo2, hi2), d1,
+                                    t);
       accn[j] = t;
     }
   }
 #pragma unroll
   for (int j = 0; j < MAX_LS; ++j) {
-    const float x = wave_allreduce_sum(accn[j]);
+    const float x = wave_allreduce_sum(accn[j].x + accn[j].y);
     if (lane == j) acc_nt[wid][j] = x;
   }
   if (lane < MAX_LS) acc_llh[wid][lane] = llh_lane;
@@ -721,35 +754,58 @@ __global__ void __launch_bounds__(BLOCK) k2_ls_v3_bf16(
   for (int j = 0; j < MAX_LS; ++j) s[j] = s_lad[j];
 
   double llh_lane = 0.0;
+  const v2f lo2 = {min_f, min_f}, hi2 = {max_f, max_f};
 
   for (long long e = e0 + wid; e < e1; e += NWAVE) {
     const u32* __restrict__ fv = F + (size_t)indices[e] * (K / 2);
-    float acc[MAX_LS];
+    v2f acc[MAX_LS];
 #pragma unroll
-    for (int j = 0; j < MAX_LS; ++j) acc[j] = 0.f;
-    for (int k = lane * 8; k < K; k += WAVE * 8) {
-      const f32x8 b = ld8bf(fv + k / 2);
-      const f32x8 a = ld8bf(fu + k / 2);
-      const float4 g0 = ld4(gu + k);
-      const float4 g1 = ld4(gu + k + 4);
+    for (int j = 0; j < MAX_LS; ++j) acc[j] = v2f{0.f, 0.f};
+    // U=1 (plain loop): measured best for bf16 — U=2 cost 15% (36.8 vs
+    // 32.1 ms/sweep at K=5000), the unpack temps raise pressure enough
+    // that grouping loses more to scheduling than it gains in MLP.
+    constexpr int U = 1;
+    constexpr int KSTR = WAVE * 8;
+    for (int k = lane * 8; k < K; k += U * KSTR) {
+      uint4 braw[U];  // only the global fv loads grouped (see fp32 variant)
 #pragma unroll
-      for (int j = 0; j < MAX_LS; ++j) {
-        const float sj = s[j];
-        float t = acc[j];
-        t = fmaf(fminf(fmaxf(fmaf(sj, g0.x, a.a.x), min_f), max_f), b.a.x, t);
-        t = fmaf(fminf(fmaxf(fmaf(sj, g0.y, a.a.y), min_f), max_f), b.a.y, t);
-        t = fmaf(fminf(fmaxf(fmaf(sj, g0.z, a.a.z), min_f), max_f), b.a.z, t);
-        t = fmaf(fminf(fmaxf(fmaf(sj, g0.w, a.a.w), min_f), max_f), b.a.w, t);
-        t = fmaf(fminf(fmaxf(fmaf(sj, g1.x, a.b.x), min_f), max_f), b.b.x, t);
-        t = fmaf(fminf(fmaxf(fmaf(sj, g1.y, a.b.y), min_f), max_f), b.b.y, t);
-        t = fmaf(fminf(fmaxf(fmaf(sj, g1.z, a.b.z), min_f), max_f), b.b.z, t);
-        t = fmaf(fminf(fmaxf(fmaf(sj, g1.w, a.b.w), min_f), max_f), b.b.w, t);
-        acc[j] = t;
+      for (int t = 0; t < U; ++t) {
+        const int kk = k + t * KSTR;
+        const int ks = kk < K ? kk : (K - 8);  // clamped: load always valid
+        braw[t] = *reinterpret_cast<const uint4*>(fv + ks / 2);
+      }
+#pragma unroll
+      for (int t = 0; t < U; ++t) {
+        if (k + t * KSTR >= K) break;
+        const int kk = k + t * KSTR;
+        const uint4 araw = *reinterpret_cast<const uint4*>(fu + kk / 2);
+        const float4 gA = ld4(gu + kk);
+        const float4 gB = ld4(gu + kk + 4);
+        const v2f b0 = bf2(braw[t].x), b1 = bf2(braw[t].y),
+                  b2 = bf2(braw[t].z), b3 = bf2(braw[t].w);
+        const v2f a0 = bf2(araw.x), a1 = bf2(araw.y),
+                  a2 = bf2(araw.z), a3 = bf2(araw.w);
+        const v2f g0 = {gA.x, gA.y}, g1 = {gA.z, gA.w};
+        const v2f g2 = {gB.x, gB.y}, g3 = {gB.z, gB.w};
+#pragma unroll
+        for (int j = 0; j < MAX_LS; ++j) {
+          const v2f sj = {s[j], s[j]};
+          v2f t2 = acc[j];
+          t2 = __builtin_elementwise_fma(pk_clamp_fma(sj, g0, a0, lo2, hi2),
+                                         b0, t2);
+          t2 = __builtin_elementwise_fma(pk_clamp_fma(sj, g1, a1, lo2, hi2),
+                                         b1, t2);
+          t2 = __builtin_elementwise_fma(pk_clamp_fma(sj, g2, a2, lo2, hi2),
+                                         b2, t2);
+          t2 = __builtin_elementwise_fma(pk_clamp_fma(sj, g3, a3, lo2, hi2),
+                                         b3, t2);
+          acc[j] = t2;
+        }
       }
     }
 #pragma unroll
     for (int j = 0; j < MAX_LS; ++j) {
-      const float x = wave_allreduce_sum(acc[j]);
+      const float x = wave_allreduce_sum(acc[j].x + acc[j].y);
       if (lane == j) {
         const float p = clamp_p(__expf(-x), min_p, max_p);
         llh_lane += (double)log1pf(-p) + (double)x;
@@ -757,43 +813,37 @@ __global__ void __launch_bounds__(BLOCK) k2_ls_v3_bf16(
     }
   }
 
-  float accn[MAX_LS];
+  v2f accn[MAX_LS];
 #pragma unroll
-  for (int j = 0; j < MAX_LS; ++j) accn[j] = 0.f;
+  for (int j = 0; j < MAX_LS; ++j) accn[j] = v2f{0.f, 0.f};
   float p_gg = 0.f;
   for (int k = tid * 8; k < K; k += BLOCK * 8) {
-    const f32x8 a = ld8bf(fu + k / 2);
-    const float4 g0 = ld4(gu + k);
-    const float4 g1 = ld4(gu + k + 4);
-    const float4 s0 = ld4(sumF + k);
-    const float4 s1 = ld4(sumF + k + 4);
-    p_gg = dot4(g1, g1, dot4(g0, g0, p_gg));
+    const uint4 araw = *reinterpret_cast<const uint4*>(fu + k / 2);
+    const v2f a0 = bf2(araw.x), a1 = bf2(araw.y), a2 = bf2(araw.z),
+              a3 = bf2(araw.w);
+    const float4 gA = ld4(gu + k);
+    const float4 gB = ld4(gu + k + 4);
+    const v2f g0 = {gA.x, gA.y}, g1 = {gA.z, gA.w};
+    const v2f g2 = {gB.x, gB.y}, g3 = {gB.z, gB.w};
+    const float4 sA = ld4(sumF + k);
+    const float4 sB = ld4(sumF + k + 4);
+    const v2f d0 = a0 - v2f{sA.x, sA.y}, d1 = a1 - v2f{sA.z, sA.w};
+    const v2f d2 = a2 - v2f{sB.x, sB.y}, d3 = a3 - v2f{sB.z, sB.w};
+    p_gg = dot4(gA, gA, dot4(gB, gB, p_gg));
 #pragma unroll
     for (int j = 0; j < MAX_LS; ++j) {
-      const float sj = s[j];
-      float t = accn[j];
-      t = fmaf(fminf(fmaxf(fmaf(sj, g0.x, a.a.x), min_f), max_f),
-               a.a.x - s0.x, t);
-      t = fmaf(fminf(fmaxf(fmaf(sj, g0.y, a.a.y), min_f), max_f),
-               a.a.y - s0.y, t);
-      t = fmaf(fminf(fmaxf(fmaf(sj, g0.z, a.a.z), min_f), max_f),
-               a.a.z - s0.z, t);
-      t = fmaf(fminf(fmaxf(fmaf(sj, g0.w, a.a.w), min_f), max_f),
-               a.a.w - s0.w, t);
-      t = fmaf(fminf(fmaxf(fmaf(sj, g1.x, a.b.x), min_f), max_f),
-               a.b.x - s1.x, t);
-      t = fmaf(fminf(fmaxf(fmaf(sj, g1.y, a.b.y), min_f), max_f),
-               a.b.y - s1.y, t);
-      t = fmaf(fminf(fmaxf(fmaf(sj, g1.z, a.b.z), min_f), max_f),
-               a.b.z - s1.z, t);
-      t = fmaf(fminf(fmaxf(fmaf(sj, g1.w, a.b.w), min_f), max_f),
-               a.b.w - s1.w, t);
+      const v2f sj = {s[j], s[j]};
+      v2f t = accn[j];
+      t = __builtin_elementwise_fma(pk_clamp_fma(sj, g0, a0, lo2, hi2), d0, t);
+      t = __builtin_elementwise_fma(pk_clamp_fma(sj, g1, a1, lo2, hi2), d1, t);
+      t = __builtin_elementwise_fma(pk_clamp_fma(sj, g2, a2, lo2, hi2), d2, t);
+      t = __builtin_elementwise_fma(pk_clamp_fma(sj, g3, a3, lo2, hi2), d3, t);
       accn[j] = t;
     }
   }
 #pragma unroll
   for (int j = 0; j < MAX_LS; ++j) {
-    const float x = wave_allreduce_sum(accn[j]);
+    const float x = wave_allreduce_sum(accn[j].x + accn[j].y);
     if (lane == j) acc_nt[wid][j] = x;
   }
   if (lane < MAX_LS) acc_llh[wid][lane] = llh_lane;
@@ -849,6 +899,7 @@ extern "C" __global__ void __launch_bounds__(BLOCK) k3_apply_step_bf16(
 
 // ----------------------------------------------------------- host launchers
 
+#include <cstdlib>
 #include <stdexcept>
 #include <string>
 
@@ -998,8 +1049,9 @@ extern "C" void launch_k2(const float* F, const long long* indptr,
                           hipStream_t stream) {
   if (n_local == 0) return;
   if (n_ladder > 16) throw std::runtime_error("ladder length > 16 unsupported");
+  static const bool nostage = getenv("BIGCLAM_K2_NOSTAGE") != nullptr;
   const size_t lds = (size_t)K * 8;  // fu + g staged fp32
-  if (lds + 2048 <= 160 * 1024) {
+  if (!nostage && lds + 2048 <= 160 * 1024) {
     allow_large_lds((const void*)&k2_ls_v3<true>, lds);
     hipLaunchKernelGGL((k2_ls_v3<true>), dim3(n_local), dim3(256), lds,
                        stream, F, indptr, indices, sumF, grad, llh, order,
