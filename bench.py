@@ -43,6 +43,8 @@ def main():
     ap.add_argument("--dtype", choices=["fp32", "bf16"], default="fp32")
     ap.add_argument("--graph-seed", type=int, default=42)
     ap.add_argument("--locality", type=float, default=0.7)
+    ap.add_argument("--ls-steps", type=int, default=15,
+                    help="Armijo ladder depth (reference: 15 -> 16 candidates)")
     args = ap.parse_args()
 
     rank = comm.init_distributed()
@@ -56,6 +58,7 @@ def main():
         dtype=args.dtype,
         device="cuda" if use_cuda else "cpu",
         seed=7,
+        ls_steps=args.ls_steps,
     )
     tr = Trainer(
         graph, cfg, device=device, metrics=MetricsLogger(rank=rank, quiet=True)

@@ -49,6 +49,43 @@ __device__ __forceinline__ float wave_allreduce_sum(float v) {
   return v;
 }
 
+// Sum 16 per-lane values across the wave simultaneously.  Butterfly with
+// value-halving: each xor step folds the lane count 2x while handing off
+// half the surviving values to the other side, so the whole 16-value
+// reduction costs 8+4+2+1+1+1 = 17 exchanges instead of 16 independent
+// allreduces (96).  Result: value j lands on lane (j*4) in w[0]; the
+// caller writes from whatever lane holds it.
+__device__ __forceinline__ void wave_reduce16(float (&w)[MAX_LS], int lane) {
+  // steps folding values: offsets 32,16,8,4 halve the value count
+#pragma unroll
+  for (int step = 0; step < 4; ++step) {
+    const int off = 32 >> step;
+    const int nv = 8 >> step;  // surviving values after this step
+    const bool hi = (lane & off) != 0;
+#pragma unroll
+    for (int i = 0; i < 8; ++i) {
+      if (i >= nv) break;
+      const float send = hi ? w[i] : w[i + nv];
+      const float recv = __shfl_xor(send, off, WAVE);
+      w[i] = (hi ? w[i + nv] : w[i]) + recv;
+    }
+  }
+  // lanes now hold ONE value each: lane group (l>>2)&15... value index is
+  // bit-reversed-ish; the mapping resolved below.  Fold the last 4 lanes.
+  w[0] += __shfl_xor(w[0], 2, WAVE);
+  w[0] += __shfl_xor(w[0], 1, WAVE);
+}
+
+// After wave_reduce16, value j sits (replicated over a 4-lane group) on the
+// lanes whose bits select it: step k kept value-bit (nv) on the side with
+// lane-bit off=32>>k SET.  Lane l holds value j(l) =
+//   bit3=(l>>5)&1? no — derived: j = ((l&32)?8:0)|((l&16)?4:0)|((l&8)?2:0)|((l&4)?1:0)
+__device__ __forceinline__ int wave_reduce16_owner(int j) {
+  // inverse map: the lane group that holds value j is
+  // l = ((j&8)?32:0)|((j&4)?16:0)|((j&2)?8:0)|((j&1)?4:0)
+  return ((j & 8) << 2) | ((j & 4) << 2) | ((j & 2) << 2) | ((j & 1) << 2);
+}
+
 // block-wide sum; `red` is NWAVE floats of LDS; result on every thread.
 __device__ __forceinline__ float block_allreduce_sum(float v, float* red) {
   const int lane = threadIdx.x & (WAVE - 1);
@@ -149,7 +186,7 @@ __device__ __forceinline__ float dot8(f32x8 a, f32x8 b, float acc) {
 // (the block touches one 4*K-byte row at a time).
 
 template <int NSLOT>
-__global__ void __launch_bounds__(BLOCK) k1_grad_llh_t(
+__global__ void __launch_bounds__(BLOCK, 4) k1_grad_llh_t(
     const float* __restrict__ F, const long long* __restrict__ indptr,
     const int* __restrict__ indices, const float* __restrict__ sumF,
     const int* __restrict__ order, float* __restrict__ grad,
@@ -412,24 +449,22 @@ __global__ void __launch_bounds__(BLOCK, 3) k2_ls_v3(
 #pragma unroll
   for (int j = 0; j < MAX_LS; ++j) s[j] = s_lad[j];
 
-  double llh_lane = 0.0;  // lane j (j<16) accumulates candidate j edge terms
+  // this lane owns candidate jmine after wave_reduce16 (lanes with
+  // (lane & 3) == 0; 4-way replicated otherwise)
+  const int jmine = (((lane >> 5) & 1) << 3) | (((lane >> 4) & 1) << 2) |
+                    (((lane >> 3) & 1) << 1) | ((lane >> 2) & 1);
+  double llh_mine = 0.0;  // candidate jmine edge terms (lane&3)==0 lanes
   const v2f lo2 = {min_f, min_f}, hi2 = {max_f, max_f};
 
   for (long long e = e0 + wid; e < e1; e += NWAVE) {
     const float* __restrict__ fv = F + (size_t)indices[e] * K;
-    v2f acc[MAX_LS];
+    v2f acc2[MAX_LS];
 #pragma unroll
-    for (int j = 0; j < MAX_LS; ++j) acc[j] = v2f{0.f, 0.f};
-    // 4-deep k-unroll: the 4 independent fv loads issue back-to-back
-    // BEFORE the 4 compute bodies, so each wave keeps 4 HBM loads in
-    // flight across the ~165-cycle body (1 load in flight measured
-    // 1.9 TB/s; latency-bound, not bandwidth-bound).
+    for (int j = 0; j < MAX_LS; ++j) acc2[j] = v2f{0.f, 0.f};
     constexpr int U = 4;
     constexpr int KSTR = WAVE * 4;
     for (int k = lane * 4; k < K; k += U * KSTR) {
-      float4 b[U];  // ONLY the global fv loads are grouped: LDS fu/g reads
-                    // are short-latency and prefetching them cost 32 VGPRs
-                    // (184 total -> 2 waves/SIMD, a 45% regression)
+      float4 b[U];  // grouped global loads keep 4 HBM reads in flight
 #pragma unroll
       for (int t = 0; t < U; ++t) {
         const int kk = k + t * KSTR;
@@ -446,29 +481,30 @@ __global__ void __launch_bounds__(BLOCK, 3) k2_ls_v3(
 #pragma unroll
         for (int j = 0; j < MAX_LS; ++j) {
           const v2f sj = {s[j], s[j]};
-          v2f t2 = acc[j];
+          v2f t2 = acc2[j];
           t2 = __builtin_elementwise_fma(pk_clamp_fma(sj, g0, a0, lo2, hi2),
                                          b0, t2);
           t2 = __builtin_elementwise_fma(pk_clamp_fma(sj, g1, a1, lo2, hi2),
                                          b1, t2);
-          acc[j] = t2;
+          acc2[j] = t2;
         }
       }
     }
+    float acc[MAX_LS];
 #pragma unroll
-    for (int j = 0; j < MAX_LS; ++j) {
-      const float x = wave_allreduce_sum(acc[j].x + acc[j].y);
-      if (lane == j) {
-        const float p = clamp_p(__expf(-x), min_p, max_p);
-        llh_lane += (double)log1pf(-p) + (double)x;
-      }
+    for (int j = 0; j < MAX_LS; ++j) acc[j] = acc2[j].x + acc2[j].y;
+    wave_reduce16(acc, lane);  // 17 shfls for all 16 sums
+    if ((lane & 3) == 0) {
+      const float x = acc[0];
+      const float p = clamp_p(__expf(-x), min_p, max_p);
+      llh_mine += (double)log1pf(-p) + (double)x;
     }
   }
 
   // node terms cand_j.(Fu - sumF) + grad.grad, block-strided k, once
-  v2f accn[MAX_LS];
+  v2f accn2[MAX_LS];
 #pragma unroll
-  for (int j = 0; j < MAX_LS; ++j) accn[j] = v2f{0.f, 0.f};
+  for (int j = 0; j < MAX_LS; ++j) accn2[j] = v2f{0.f, 0.f};
   float p_gg = 0.f;
   for (int k = tid * 4; k < K; k += BLOCK * 4) {
     const float4 a = ld4(fu + k);
@@ -482,37 +518,39 @@ __global__ void __launch_bounds__(BLOCK, 3) k2_ls_v3(
 #pragma unroll
     for (int j = 0; j < MAX_LS; ++j) {
       const v2f sj = {s[j], s[j]};
-      v2f t = accn[j];
+      v2f t = accn2[j];
       t = __builtin_elementwise_fma(pk_clamp_fma(sj, g0, a0, lo2, hi2), d0,
                                     t);
       t = __builtin_elementwise_fma(pk_clamp_fma(sj, g1, a1, lo2, hi2), d1,
                                     t);
-      accn[j] = t;
+      accn2[j] = t;
     }
   }
+  float accn[MAX_LS];
 #pragma unroll
-  for (int j = 0; j < MAX_LS; ++j) {
-    const float x = wave_allreduce_sum(accn[j].x + accn[j].y);
-    if (lane == j) acc_nt[wid][j] = x;
+  for (int j = 0; j < MAX_LS; ++j) accn[j] = accn2[j].x + accn2[j].y;
+  wave_reduce16(accn, lane);
+  if ((lane & 3) == 0) {
+    acc_nt[wid][jmine] = accn[0];
+    acc_llh[wid][jmine] = llh_mine;
   }
-  if (lane < MAX_LS) acc_llh[wid][lane] = llh_lane;
-  const float gg = block_allreduce_sum(p_gg, red);
+  const float gg = block_allreduce_sum(p_gg, red);  // ends in __syncthreads
 
-  if (tid == 0) {
-    const double llh_u = llh[u];
-    float chosen = 0.f;
-    for (int jj = 0; jj < n_ladder; ++jj) {  // descending: first accept wins
+  // parallel Armijo selection: 16 threads each evaluate one candidate,
+  // ballot picks the FIRST accepted (ladder is descending -> max step).
+  if (wid == 0) {
+    bool ok = false;
+    if (lane < MAX_LS) {
       double trial = 0.0;
 #pragma unroll
       for (int wv = 0; wv < NWAVE; ++wv)
-        trial += acc_llh[wv][jj] + (double)acc_nt[wv][jj];
-      const float sjj = s_lad[jj];
-      if (trial >= llh_u + (double)(alpha * sjj * gg)) {
-        chosen = sjj;
-        break;
-      }
+        trial += acc_llh[wv][lane] + (double)acc_nt[wv][lane];
+      ok = (lane < n_ladder) &&
+           (trial >= llh[u] + (double)(alpha * s_lad[lane] * gg));
     }
-    best[u] = chosen;
+    const unsigned long long bal = __ballot(ok);
+    if (lane == 0)
+      best[u] = bal ? s_lad[__ffsll((unsigned long long)bal) - 1] : 0.f;
   }
 }
 
@@ -730,8 +768,8 @@ __global__ void __launch_bounds__(BLOCK) k2_ls_v3_bf16(
   __shared__ __attribute__((aligned(16))) float acc_nt[NWAVE][MAX_LS];
   __shared__ __attribute__((aligned(16))) float red[NWAVE];
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  float* g_s = reinterpret_cast<float*>(smem);              // K floats
-  u32* fu_s = reinterpret_cast<u32*>(smem + (size_t)K * 4); // K/2 u32
+  float* g_s = reinterpret_cast<float*>(smem);               // K floats
+  u32* fu_s = reinterpret_cast<u32*>(smem + (size_t)K * 4);  // K/2 u32
 
   if (tid < MAX_LS) s_lad[tid] = (tid < n_ladder) ? ladder[tid] : 0.f;
 
@@ -753,7 +791,9 @@ __global__ void __launch_bounds__(BLOCK) k2_ls_v3_bf16(
 #pragma unroll
   for (int j = 0; j < MAX_LS; ++j) s[j] = s_lad[j];
 
-  double llh_lane = 0.0;
+  const int jmine = (((lane >> 5) & 1) << 3) | (((lane >> 4) & 1) << 2) |
+                    (((lane >> 3) & 1) << 1) | ((lane >> 2) & 1);
+  double llh_mine = 0.0;
   const v2f lo2 = {min_f, min_f}, hi2 = {max_f, max_f};
 
   for (long long e = e0 + wid; e < e1; e += NWAVE) {
@@ -803,13 +843,14 @@ __global__ void __launch_bounds__(BLOCK) k2_ls_v3_bf16(
         }
       }
     }
+    float accf[MAX_LS];
 #pragma unroll
-    for (int j = 0; j < MAX_LS; ++j) {
-      const float x = wave_allreduce_sum(acc[j].x + acc[j].y);
-      if (lane == j) {
-        const float p = clamp_p(__expf(-x), min_p, max_p);
-        llh_lane += (double)log1pf(-p) + (double)x;
-      }
+    for (int j = 0; j < MAX_LS; ++j) accf[j] = acc[j].x + acc[j].y;
+    wave_reduce16(accf, lane);
+    if ((lane & 3) == 0) {
+      const float x = accf[0];
+      const float p = clamp_p(__expf(-x), min_p, max_p);
+      llh_mine += (double)log1pf(-p) + (double)x;
     }
   }
 
@@ -841,29 +882,30 @@ __global__ void __launch_bounds__(BLOCK) k2_ls_v3_bf16(
       accn[j] = t;
     }
   }
+  float accn_f[MAX_LS];
 #pragma unroll
-  for (int j = 0; j < MAX_LS; ++j) {
-    const float x = wave_allreduce_sum(accn[j].x + accn[j].y);
-    if (lane == j) acc_nt[wid][j] = x;
+  for (int j = 0; j < MAX_LS; ++j) accn_f[j] = accn[j].x + accn[j].y;
+  wave_reduce16(accn_f, lane);
+  if ((lane & 3) == 0) {
+    acc_nt[wid][jmine] = accn_f[0];
+    acc_llh[wid][jmine] = llh_mine;
   }
-  if (lane < MAX_LS) acc_llh[wid][lane] = llh_lane;
-  const float gg = block_allreduce_sum(p_gg, red);
+  const float gg = block_allreduce_sum(p_gg, red);  // ends in __syncthreads
 
-  if (tid == 0) {
-    const double llh_u = llh[u];
-    float chosen = 0.f;
-    for (int jj = 0; jj < n_ladder; ++jj) {
+  // parallel Armijo selection (see fp32 variant)
+  if (wid == 0) {
+    bool ok = false;
+    if (lane < MAX_LS) {
       double trial = 0.0;
 #pragma unroll
       for (int wv = 0; wv < NWAVE; ++wv)
-        trial += acc_llh[wv][jj] + (double)acc_nt[wv][jj];
-      const float sjj = s_lad[jj];
-      if (trial >= llh_u + (double)(alpha * sjj * gg)) {
-        chosen = sjj;
-        break;
-      }
+        trial += acc_llh[wv][lane] + (double)acc_nt[wv][lane];
+      ok = (lane < n_ladder) &&
+           (trial >= llh[u] + (double)(alpha * s_lad[lane] * gg));
     }
-    best[u] = chosen;
+    const unsigned long long bal = __ballot(ok);
+    if (lane == 0)
+      best[u] = bal ? s_lad[__ffsll((unsigned long long)bal) - 1] : 0.f;
   }
 }
 
