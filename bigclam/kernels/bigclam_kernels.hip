@@ -11,18 +11,15 @@
 //   K3 apply_step     — projected commit (scala:89-92, 171)
 //   K4 llh_only       — per-node local LLH (scala:106-120 / 177-200)
 //
-// v2 design notes (from profiles/r01_sweep_v1_stats.md):
-//  * K1 is BLOCK-per-edge: the whole 256-thread workgroup walks one
-//    neighbor at a time — fv is read once from HBM for the dot and re-read
-//    L1-hot for the owned-k (no-atomic) accumulate into the LDS gradient
-//    row.  v1's wave-per-edge axpy with LDS atomicAdd was 6x slower than
-//    the same traffic in K4.
-//  * K2 keeps Fu and grad_u row chunks RESIDENT IN REGISTERS (NSLOT
-//    float4 slots per thread, block-stride 1024 elements) so the edge loop
-//    touches only fv from memory; candidate rows are recomputed in-register
-//    (3 VALU per element per candidate) — v1 streamed fu+g+fv (60 KB) per
-//    edge through a 32 KB L1.  The 17 per-node block-reduced node-term
-//    passes are now ONE pass with a 256x16 LDS transpose reduction.
+// Design notes (measured rationale in profiles/r01_kernel_opt_log.md):
+//  * K1 is BLOCK-per-node: per edge the workgroup does one cooperative dot
+//    (block_allreduce) and an owned-k no-atomic accumulate into the LDS
+//    gradient row; fv stays in REGISTERS between the dot and the axpy.
+//  * K2 evaluates all 16 Armijo candidates DIRECTLY per edge element
+//    (clamp-fma + acc-fma in registers, v_med3_f32 clamps) — candidate
+//    rows are never materialized; fu/grad are LDS-staged per node; the 16
+//    per-edge dot sums reduce in ONE 17-shuffle butterfly (wave_reduce16)
+//    and the Armijo pick is a 16-lane ballot.
 //  * LLH is accumulated in fp64 (the convergence test is a 1e-4 relative
 //    change on a large-magnitude sum).
 //
