@@ -86,6 +86,12 @@ def main(argv=None):
     p_fit = sub.add_parser("fit")
     p_fit.add_argument("edgelist")
     _add_common(p_fit)
+    p_fit.add_argument(
+        "--resume",
+        metavar="CKPT_DIR",
+        default=None,
+        help="resume the fit from a checkpoint directory (any world size)",
+    )
 
     p_sel = sub.add_parser("select-k")
     p_sel.add_argument("edgelist")
@@ -124,7 +130,14 @@ def main(argv=None):
         return 0
 
     tr = Trainer(g, cfg, metrics=metrics)
-    res = tr.fit(init=args.init)
+    if getattr(args, "resume", None):
+        from .ckpt.checkpoint import resume as ckpt_resume
+
+        sweep0 = ckpt_resume(args.resume, tr)
+        metrics.log({"note": "resumed", "from_sweep": sweep0})
+        res = tr.fit(skip_init=True)
+    else:
+        res = tr.fit(init=args.init)
     F = tr.gather_F()
     if cfg.checkpoint_dir:  # every rank writes its own shard file
         from .ckpt.checkpoint import save_shard_checkpoint
