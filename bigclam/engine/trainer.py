@@ -68,10 +68,21 @@ class Trainer:
     # ------------------------------------------------------------------ init
     def seeds(self) -> np.ndarray:
         """Conductance seed ranking, computed once and reused across K
-        (codes/bigclam4-7.scala:75)."""
+        (codes/bigclam4-7.scala:75).  On GPU the per-node ego-net
+        conductance runs as the K5 HIP kernel; the ranking itself is a
+        trivial host pass."""
         if self._seeds is None:
+            cond = None
+            if self.state.use_hip:
+                from ..ops import hip as hip_ops
+
+                cond = (
+                    hip_ops.conductance_full_graph(self.graph, self.state.device)
+                    .cpu()
+                    .numpy()
+                )
             self._seeds = conductance_ranking(
-                self.graph, compat=self.cfg.seed_rank_compat
+                self.graph, compat=self.cfg.seed_rank_compat, cond=cond
             )
         return self._seeds
 

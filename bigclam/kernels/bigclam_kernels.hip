@@ -936,6 +936,88 @@ extern "C" __global__ void __launch_bounds__(BLOCK) k3_apply_step_bf16(
   }
 }
 
+// ------------------------------------------------------------------- K5
+//
+// Ego-net conductance per node (replaces codes/bigclamv3-7.scala:39-54;
+// formulas in SURVEY.md §2.4 / core/init.py).  Block per node u; waves
+// split u's edge list; lanes split each neighbor v's adjacency.  The 2-hop
+// endpoint w is "inside" the closed ego-net y = {u} ∪ N(u) iff w == u or w
+// appears in u's SORTED adjacency row (binary search; canonical graphs
+// store sorted rows — the row is L1-hot across the whole block).
+//   cut   = #{2-hop endpoints outside y}
+//   vol_S = |z| - cut,  |z| = deg(u) + Σ_{v∈N(u)} deg(v)
+//   vol_T = Σdeg - vol_S - 2·cut
+//   cond  = cut / max(min(vol_S, vol_T), 1)   (vol_S==0 -> 0, vol_T==0 -> 1)
+
+extern "C" __global__ void __launch_bounds__(BLOCK) k5_conductance(
+    const long long* __restrict__ indptr, const int* __restrict__ indices,
+    double* __restrict__ cond, int n, double total_degree) {
+  const int u = blockIdx.x;
+  const long long e0 = indptr[u];
+  const long long e1 = indptr[u + 1];
+  const int deg = (int)(e1 - e0);
+  const int tid = threadIdx.x;
+  const int lane = tid & (WAVE - 1);
+  const int wid = tid >> 6;
+
+  __shared__ unsigned long long s_cut, s_z;
+  if (tid == 0) {
+    s_cut = 0ull;
+    s_z = 0ull;
+  }
+  __syncthreads();
+  if (deg == 0) {
+    if (tid == 0) cond[u] = 0.0;  // vol_S == 0 guard
+    return;
+  }
+
+  const int* __restrict__ nu = indices + e0;  // sorted row of u
+  unsigned long long cut_l = 0, z_l = 0;
+  for (long long e = e0 + wid; e < e1; e += NWAVE) {
+    const int v = indices[e];
+    const long long f0 = indptr[v];
+    const long long f1 = indptr[v + 1];
+    if (lane == 0) z_l += (unsigned long long)(f1 - f0);
+    for (long long f = f0 + lane; f < f1; f += WAVE) {
+      const int w = indices[f];
+      bool inside = (w == u);
+      if (!inside) {  // binary search w in nu[0..deg)
+        int lo = 0, hi = deg - 1;
+        while (lo <= hi) {
+          const int mid = (lo + hi) >> 1;
+          const int x = nu[mid];
+          if (x == w) {
+            inside = true;
+            break;
+          }
+          if (x < w)
+            lo = mid + 1;
+          else
+            hi = mid - 1;
+        }
+      }
+      if (!inside) ++cut_l;
+    }
+  }
+  if (cut_l) atomicAdd(&s_cut, cut_l);
+  if (z_l) atomicAdd(&s_z, z_l);
+  __syncthreads();
+  if (tid == 0) {
+    const double zsz = (double)deg + (double)s_z;
+    const double cut = (double)s_cut;
+    const double vol_s = zsz - cut;
+    const double vol_t = total_degree - vol_s - 2.0 * cut;
+    double c;
+    if (vol_s == 0.0)
+      c = 0.0;
+    else if (vol_t == 0.0)
+      c = 1.0;
+    else
+      c = cut / fmax(fmin(vol_s, vol_t), 1.0);
+    cond[u] = c;
+  }
+}
+
 // ----------------------------------------------------------- host launchers
 
 #include <cstdlib>
@@ -1111,5 +1193,14 @@ extern "C" void launch_k3(float* F, const float* grad, const float* steps,
   if (n_local == 0) return;
   hipLaunchKernelGGL(k3_apply_step, dim3(n_local), dim3(256), 0, stream, F,
                      grad, steps, n_local, K, min_f, max_f);
+  HIP_CHECK(hipGetLastError());
+}
+
+extern "C" void launch_k5(const long long* indptr, const int* indices,
+                          double* cond, int n, double total_degree,
+                          hipStream_t stream) {
+  if (n == 0) return;
+  hipLaunchKernelGGL(k5_conductance, dim3(n), dim3(256), 0, stream, indptr,
+                     indices, cond, n, total_degree);
   HIP_CHECK(hipGetLastError());
 }

@@ -29,6 +29,8 @@ extern "C" void launch_k2_bf16(const void*, const long long*, const int*,
                                hipStream_t);
 extern "C" void launch_k3_bf16(void*, const float*, const float*, int, int,
                                float, float, hipStream_t);
+extern "C" void launch_k5(const long long*, const int*, double*, int, double,
+                          hipStream_t);
 
 namespace {
 
@@ -168,6 +170,18 @@ void apply_step(torch::Tensor F_local, torch::Tensor grad,
   }
 }
 
+void conductance(torch::Tensor indptr, torch::Tensor indices,
+                 torch::Tensor cond, double total_degree) {
+  CHECK_IN(indptr, torch::kInt64);
+  CHECK_IN(indices, torch::kInt32);
+  CHECK_IN(cond, torch::kFloat64);
+  const int n = (int)indptr.size(0) - 1;
+  TORCH_CHECK(cond.size(0) == n);
+  launch_k5(reinterpret_cast<const long long*>(indptr.data_ptr<int64_t>()),
+            indices.data_ptr<int>(), cond.data_ptr<double>(), n,
+            total_degree, current_stream());
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -178,4 +192,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "K2: 16-candidate Armijo line search in one edge pass (CDNA4)");
   m.def("apply_step", &apply_step,
         "K3: in-place projected commit F += s*grad (CDNA4)");
+  m.def("conductance", &conductance,
+        "K5: ego-net conductance per node (CDNA4)");
 }

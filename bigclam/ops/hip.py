@@ -96,6 +96,23 @@ def apply_step(
     ext.apply_step(F_local, grad, steps, cfg.min_f, cfg.max_f)
 
 
+def conductance_full_graph(graph, device) -> "torch.Tensor":
+    """K5: ego-net conductance of every node of the FULL graph on GPU.
+
+    Used by the seed-init ranking; the ranking itself is a
+    trivial host-side pass over the returned [N] fp64 vector.
+    """
+    import numpy as np
+
+    ext = ensure_loaded()
+    indptr = torch.from_numpy(graph.indptr.astype(np.int64)).to(device)
+    indices = torch.from_numpy(graph.indices.astype(np.int32)).to(device)
+    cond = torch.empty(graph.num_nodes, device=device, dtype=torch.float64)
+    total_degree = float(len(graph.indices))
+    ext.conductance(indptr, indices, cond, total_degree)
+    return cond
+
+
 def full_llh(
     F: torch.Tensor,
     indptr: torch.Tensor,

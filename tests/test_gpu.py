@@ -163,3 +163,19 @@ def test_bf16_k2_steps_mostly_agree_with_fp32():
     bb = stb.linesearch(gb, lb)
     agree = (b32 == bb).float().mean().item()
     assert agree > 0.9, agree
+
+
+def test_k5_conductance_matches_numpy():
+    """K5 HIP ego-net conductance == the vectorized host implementation."""
+    from bigclam.core.init import conductances
+    from bigclam.ops import hip as hip_ops
+
+    for seed, scale, ef in [(31, 9, 6.0), (32, 11, 4.0)]:
+        g = rmat_graph(scale, ef, seed=seed)
+        host = conductances(g)
+        dev = (
+            hip_ops.conductance_full_graph(g, torch.device("cuda"))
+            .cpu()
+            .numpy()
+        )
+        np.testing.assert_allclose(dev, host, rtol=1e-12, atol=1e-12)
