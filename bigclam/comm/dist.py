@@ -64,14 +64,24 @@ def all_to_all(
     inp: torch.Tensor,
     out_splits,
     in_splits,
-) -> torch.Tensor:
-    """Row-wise all_to_all_single on 2-D tensors (no-op at world_size 1)."""
+    async_op: bool = False,
+):
+    """Row-wise all_to_all_single on 2-D tensors (no-op at world_size 1).
+
+    With ``async_op`` returns the Work handle (or None when not
+    distributed) so the halo exchange can overlap with interior-node
+    compute; the caller must ``wait()`` before reading ``out``.
+    """
     if not is_distributed():
-        return out
-    dist.all_to_all_single(
-        out, inp, output_split_sizes=list(out_splits), input_split_sizes=list(in_splits)
+        return None if async_op else out
+    work = dist.all_to_all_single(
+        out,
+        inp,
+        output_split_sizes=list(out_splits),
+        input_split_sizes=list(in_splits),
+        async_op=async_op,
     )
-    return out
+    return work if async_op else out
 
 
 def barrier():

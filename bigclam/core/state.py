@@ -56,9 +56,33 @@ class ShardState:
         self.indices = torch.from_numpy(shard.indices).to(dev)
         # launch order: degree-descending so hub blocks start first
         deg = shard.degrees()
-        self.order = torch.from_numpy(
-            np.argsort(-deg, kind="stable").astype(np.int32)
-        ).to(dev)
+        order_np = np.argsort(-deg, kind="stable").astype(np.int32)
+        self.order = torch.from_numpy(order_np).to(dev)
+        # interior/boundary split for halo-compute overlap: a node is
+        # boundary iff any neighbor lives in the halo section (local row
+        # index >= n_local).  K1 on interior nodes can run while the halo
+        # all_to_all is in flight (halo fractions are large on power-law
+        # graphs: Email-Enron ws=8 halo = 1.95x N).
+        if shard.world_size > 1 and shard.n_halo > 0:
+            row_has_halo = np.zeros(shard.n_local, dtype=bool)
+            remote = shard.indices >= shard.n_local
+            if remote.any():
+                seg = np.add.reduceat(
+                    remote, shard.indptr[:-1].clip(max=max(len(remote) - 1, 0))
+                )
+                row_has_halo = (seg > 0) & (np.diff(shard.indptr) > 0)
+            bnd = order_np[row_has_halo[order_np]]
+            interior = order_np[~row_has_halo[order_np]]
+            self.order_interior = torch.from_numpy(
+                np.ascontiguousarray(interior)
+            ).to(dev)
+            self.order_boundary = torch.from_numpy(
+                np.ascontiguousarray(bnd)
+            ).to(dev)
+        else:
+            self.order_interior = self.order
+            self.order_boundary = self.order[:0]
+        self._halo_send: Optional[torch.Tensor] = None
         # per-edge source row (torch reference path); built lazily on CPU
         self._edge_src: Optional[torch.Tensor] = None
 
@@ -120,13 +144,20 @@ class ShardState:
         comm.all_reduce_(self.sumF)
 
     # ---------------------------------------------------------- communication
-    def halo_exchange(self):
-        """C8: refresh halo rows with peers' current F rows (p2p all-to-all)."""
+    def halo_exchange(self, async_op: bool = False):
+        """C8: refresh halo rows with peers' current F rows (p2p all-to-all).
+
+        With ``async_op`` returns the Work handle (None when nothing to
+        exchange); the send buffer stays referenced until the next call.
+        """
         if self.shard.world_size == 1 or self.shard.n_halo == 0:
-            return
+            return None
         send = self.F_local.index_select(0, self.send_idx)
+        self._halo_send = send  # keep alive while the collective is in flight
         recv = self.F[self.n_local :]
-        comm.all_to_all(recv, send, self.recv_splits, self.send_splits)
+        return comm.all_to_all(
+            recv, send, self.recv_splits, self.send_splits, async_op=async_op
+        )
 
     # ------------------------------------------------------------------- ops
     def grad_llh(self) -> Tuple[torch.Tensor, torch.Tensor]:
@@ -143,6 +174,40 @@ class ShardState:
             n_local=self.n_local,
             edge_src=self.edge_src(),
         )
+
+    def grad_llh_overlap(self, halo_work) -> Tuple[torch.Tensor, torch.Tensor]:
+        """K1 overlapped with the in-flight halo exchange: interior nodes
+        (no halo neighbors) run while the all_to_all completes, boundary
+        nodes after ``halo_work.wait()``.  Falls back to wait-then-full
+        when there is nothing to overlap (world 1, or the CPU reference
+        path, which is vectorized over all nodes at once)."""
+        if (
+            halo_work is not None
+            and self.use_hip
+            and self.order_boundary.numel() > 0
+            and self.order_interior.numel() > 0
+        ):
+            ops = _hip_ops()
+            grad = torch.empty(
+                self.n_local, self.F.shape[1], device=self.device,
+                dtype=torch.float32,
+            )
+            llh = torch.empty(
+                self.n_local, device=self.device, dtype=torch.float64
+            )
+            ops.edge_grad_llh(
+                self.F, self.indptr, self.indices, self.sumF,
+                self.order_interior, self.cfg, out=(grad, llh),
+            )
+            halo_work.wait()
+            ops.edge_grad_llh(
+                self.F, self.indptr, self.indices, self.sumF,
+                self.order_boundary, self.cfg, out=(grad, llh),
+            )
+            return grad, llh
+        if halo_work is not None:
+            halo_work.wait()
+        return self.grad_llh()
 
     def linesearch(self, grad: torch.Tensor, llh: torch.Tensor) -> torch.Tensor:
         if self.use_hip:

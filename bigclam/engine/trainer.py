@@ -143,20 +143,21 @@ class Trainer:
         llh_total).  llh_total is the objective of the CURRENT state
         (the reference v2's initial LLH, codes/Bigclamv2.scala:204)."""
         st = self.state
-        st.halo_exchange()
-        grad, llh_nodes = st.grad_llh()
+        work = st.halo_exchange(async_op=True)
+        grad, llh_nodes = st.grad_llh_overlap(work)
         total = llh_nodes.sum().reshape(1)
         comm.all_reduce_(total)
         return grad, llh_nodes, float(total.item())
 
     def pipelined_sweep(self, grad, llh_nodes):
-        """One iteration: K2 -> K3 commit -> halo -> K1.  Returns
-        (grad', llh_nodes', llh_total_after_commit, steps)."""
+        """One iteration: K2 -> K3 commit -> async halo -> K1 (interior
+        nodes overlap the halo all_to_all; boundary nodes run after it).
+        Returns (grad', llh_nodes', llh_total_after_commit, steps)."""
         st = self.state
         steps = st.linesearch(grad, llh_nodes)
         st.apply_step(grad, steps)
-        st.halo_exchange()
-        grad, llh_nodes = st.grad_llh()
+        work = st.halo_exchange(async_op=True)
+        grad, llh_nodes = st.grad_llh_overlap(work)
         total = llh_nodes.sum().reshape(1)
         comm.all_reduce_(total)
         return grad, llh_nodes, float(total.item()), steps

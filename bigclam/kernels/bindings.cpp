@@ -66,6 +66,8 @@ void edge_grad_llh(torch::Tensor F, torch::Tensor indptr,
   CHECK_IN(grad, torch::kFloat32);
   CHECK_IN(llh, torch::kFloat64);
   const int n_local = (int)indptr.size(0) - 1;
+  // grid = #nodes listed in `order` (may be a subset: halo overlap split)
+  const int n_blocks = (int)order.size(0);
   const int K = (int)F.size(1);
   TORCH_CHECK(grad.size(0) == n_local && grad.size(1) == K);
   const auto ip = reinterpret_cast<const long long*>(indptr.data_ptr<int64_t>());
@@ -73,13 +75,13 @@ void edge_grad_llh(torch::Tensor F, torch::Tensor indptr,
     TORCH_CHECK(K % 8 == 0, "bf16 K must be padded to a multiple of 8");
     launch_k1_bf16(F.data_ptr(), ip, indices.data_ptr<int>(),
                    sumF.data_ptr<float>(), order.data_ptr<int>(),
-                   grad.data_ptr<float>(), llh.data_ptr<double>(), n_local, K,
-                   (float)min_p, (float)max_p, current_stream());
+                   grad.data_ptr<float>(), llh.data_ptr<double>(), n_blocks,
+                   K, (float)min_p, (float)max_p, current_stream());
   } else {
     TORCH_CHECK(K % 4 == 0, "K must be padded to a multiple of 4");
     launch_k1(F.data_ptr<float>(), ip, indices.data_ptr<int>(),
               sumF.data_ptr<float>(), order.data_ptr<int>(),
-              grad.data_ptr<float>(), llh.data_ptr<double>(), n_local, K,
+              grad.data_ptr<float>(), llh.data_ptr<double>(), n_blocks, K,
               (float)min_p, (float)max_p, current_stream());
   }
 }

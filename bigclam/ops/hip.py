@@ -39,11 +39,20 @@ def edge_grad_llh(
     sumF: torch.Tensor,
     order: torch.Tensor,
     cfg: BigClamConfig,
+    out: Tuple[torch.Tensor, torch.Tensor] = None,
 ) -> Tuple[torch.Tensor, torch.Tensor]:
+    """K1 over the nodes listed in ``order`` (grid = len(order)); rows of
+    grad/llh not listed stay untouched, so interior/boundary subsets can be
+    launched separately into the same ``out`` buffers (halo overlap)."""
     ext = ensure_loaded()
     n_local = len(indptr) - 1
-    grad = torch.empty(n_local, F.shape[1], device=F.device, dtype=torch.float32)
-    llh = torch.empty(n_local, device=F.device, dtype=torch.float64)
+    if out is None:
+        grad = torch.empty(
+            n_local, F.shape[1], device=F.device, dtype=torch.float32
+        )
+        llh = torch.empty(n_local, device=F.device, dtype=torch.float64)
+    else:
+        grad, llh = out
     ext.edge_grad_llh(
         F, indptr, indices, sumF, order, grad, llh, cfg.min_p, cfg.max_p
     )

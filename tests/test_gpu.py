@@ -179,3 +179,20 @@ def test_k5_conductance_matches_numpy():
             .numpy()
         )
         np.testing.assert_allclose(dev, host, rtol=1e-12, atol=1e-12)
+
+
+def test_k1_split_order_launches_match_full():
+    """Interior/boundary split launches (halo overlap) == one full launch."""
+    from bigclam.ops import hip as hip_ops
+
+    g = rmat_graph(10, 6.0, seed=41)
+    cfg, st = _mkstate(g, 128, seed=9)
+    grad_full, llh_full = st.grad_llh()
+    perm = torch.randperm(st.n_local, device=st.device).int()
+    o1, o2 = perm[: st.n_local // 3].contiguous(), perm[st.n_local // 3 :].contiguous()
+    grad = torch.empty_like(grad_full)
+    llh = torch.empty_like(llh_full)
+    hip_ops.edge_grad_llh(st.F, st.indptr, st.indices, st.sumF, o1, cfg, out=(grad, llh))
+    hip_ops.edge_grad_llh(st.F, st.indptr, st.indices, st.sumF, o2, cfg, out=(grad, llh))
+    torch.testing.assert_close(grad, grad_full, rtol=0, atol=0)
+    torch.testing.assert_close(llh, llh_full, rtol=0, atol=0)
