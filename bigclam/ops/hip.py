@@ -59,6 +59,9 @@ def edge_grad_llh(
     return grad, llh
 
 
+_ladder_cache = {}
+
+
 def linesearch(
     F: torch.Tensor,
     indptr: torch.Tensor,
@@ -72,9 +75,11 @@ def linesearch(
     ext = ensure_loaded()
     n_local = len(indptr) - 1
     best = torch.empty(n_local, device=F.device, dtype=torch.float32)
-    ladder = torch.tensor(
-        cfg.ladder(), device=F.device, dtype=torch.float32
-    )
+    key = (tuple(cfg.ladder()), str(F.device))
+    ladder = _ladder_cache.get(key)
+    if ladder is None:  # cached: a fresh HtoD copy per sweep costs a sync
+        ladder = torch.tensor(cfg.ladder(), device=F.device, dtype=torch.float32)
+        _ladder_cache[key] = ladder
     ext.linesearch(
         F,
         indptr,
