@@ -70,17 +70,18 @@ def main():
             torch.cuda.synchronize()
 
     # pipelined loop: each timed step = one full algorithmic sweep
-    # (16-candidate line search -> projected commit -> sumF allreduce ->
-    # halo exchange -> fused grad+LLH pass -> scalar LLH allreduce); the
-    # post-update LLH comes from the next grad pass (see engine/trainer.py).
-    grad, llh_nodes, _ = tr.prologue()
+    # (projected commit of the carried Armijo steps -> sumF allreduce ->
+    # halo exchange -> fused grad+LLH+16-candidate line-search pass ->
+    # scalar LLH allreduce); the post-update LLH comes from the next grad
+    # pass (see engine/trainer.py).
+    carry, _ = tr.prologue()
     for _ in range(args.warmup):
-        grad, llh_nodes, _, _ = tr.pipelined_sweep(grad, llh_nodes)
+        carry, _, _ = tr.pipelined_sweep(carry)
     comm.barrier()
     sync()
     t0 = time.perf_counter()
     for _ in range(args.steps):
-        grad, llh_nodes, _, _ = tr.pipelined_sweep(grad, llh_nodes)
+        carry, _, _ = tr.pipelined_sweep(carry)
     comm.barrier()
     sync()
     elapsed = time.perf_counter() - t0

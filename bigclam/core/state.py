@@ -175,6 +175,56 @@ class ShardState:
             edge_src=self.edge_src(),
         )
 
+    @property
+    def fused_ok(self) -> bool:
+        """The fused K1+K2 kernel covers fp32 rows up to K=8192 (LDS holds
+        grad + fu); other shapes use the separate kernels."""
+        return (
+            self.use_hip
+            and self.storage_dtype == torch.float32
+            and self.kp <= 8192
+        )
+
+    def fused_grad_ls_overlap(
+        self, halo_work
+    ) -> Tuple[torch.Tensor, torch.Tensor, torch.Tensor]:
+        """(grad, llh_nodes, best_step) of the current post-halo state —
+        the fused KF kernel where supported (with interior/boundary halo
+        overlap), else K1 (+overlap) followed by K2."""
+        if self.fused_ok:
+            ops = _hip_ops()
+            n, kp = self.n_local, self.F.shape[1]
+            grad = torch.empty(
+                n, kp, device=self.device, dtype=torch.float32
+            )
+            llh = torch.empty(n, device=self.device, dtype=torch.float64)
+            best = torch.empty(n, device=self.device, dtype=torch.float32)
+            out = (grad, llh, best)
+            if (
+                halo_work is not None
+                and self.order_boundary.numel() > 0
+                and self.order_interior.numel() > 0
+            ):
+                ops.fused_grad_ls(
+                    self.F, self.indptr, self.indices, self.sumF,
+                    self.order_interior, self.cfg, out=out,
+                )
+                halo_work.wait()
+                ops.fused_grad_ls(
+                    self.F, self.indptr, self.indices, self.sumF,
+                    self.order_boundary, self.cfg, out=out,
+                )
+            else:
+                if halo_work is not None:
+                    halo_work.wait()
+                ops.fused_grad_ls(
+                    self.F, self.indptr, self.indices, self.sumF,
+                    self.order, self.cfg, out=out,
+                )
+            return grad, llh, best
+        grad, llh = self.grad_llh_overlap(halo_work)
+        return grad, llh, self.linesearch(grad, llh)
+
     def grad_llh_overlap(self, halo_work) -> Tuple[torch.Tensor, torch.Tensor]:
         """K1 overlapped with the in-flight halo exchange: interior nodes
         (no halo neighbors) run while the all_to_all completes, boundary

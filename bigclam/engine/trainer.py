@@ -139,28 +139,30 @@ class Trainer:
     # trajectory, one edge pass fewer).
 
     def prologue(self):
-        """Halo + K1 on the current state; returns (grad, llh_nodes,
-        llh_total).  llh_total is the objective of the CURRENT state
-        (the reference v2's initial LLH, codes/Bigclamv2.scala:204)."""
+        """Halo + fused grad/line-search on the current state; returns
+        (carry, llh_total).  llh_total is the objective of the CURRENT
+        state (the reference v2's initial LLH, codes/Bigclamv2.scala:204);
+        carry = (grad, steps) feeds the first pipelined_sweep."""
         st = self.state
         work = st.halo_exchange(async_op=True)
-        grad, llh_nodes = st.grad_llh_overlap(work)
+        grad, llh_nodes, steps = st.fused_grad_ls_overlap(work)
         total = llh_nodes.sum().reshape(1)
         comm.all_reduce_(total)
-        return grad, llh_nodes, float(total.item())
+        return (grad, steps), float(total.item())
 
-    def pipelined_sweep(self, grad, llh_nodes):
-        """One iteration: K2 -> K3 commit -> async halo -> K1 (interior
-        nodes overlap the halo all_to_all; boundary nodes run after it).
-        Returns (grad', llh_nodes', llh_total_after_commit, steps)."""
+    def pipelined_sweep(self, carry):
+        """One iteration: K3 commit of the carried steps -> async halo ->
+        fused KF (= K1 grad+LLH and K2 line search; interior nodes overlap
+        the halo all_to_all, boundary after).  Returns
+        (carry', llh_total_after_commit, committed_steps)."""
         st = self.state
-        steps = st.linesearch(grad, llh_nodes)
+        grad, steps = carry
         st.apply_step(grad, steps)
         work = st.halo_exchange(async_op=True)
-        grad, llh_nodes = st.grad_llh_overlap(work)
+        grad, llh_nodes, steps_next = st.fused_grad_ls_overlap(work)
         total = llh_nodes.sum().reshape(1)
         comm.all_reduce_(total)
-        return grad, llh_nodes, float(total.item()), steps
+        return (grad, steps_next), float(total.item()), steps
 
     def fit(self, init: str = "seed", skip_init: bool = False) -> FitResult:
         cfg = self.cfg
@@ -168,12 +170,12 @@ class Trainer:
             self.init_F(init)
         res = FitResult()
         llh_old = 0.0
-        grad, llh_nodes, llh0 = self.prologue()
+        carry, llh0 = self.prologue()
         self.metrics.log({"sweep": -1, "llh": llh0, "note": "initial"})
         for i in range(cfg.max_sweeps):
             timer = PhaseTimer(sync=True)
             timer.start("sweep")
-            grad, llh_nodes, llh, steps = self.pipelined_sweep(grad, llh_nodes)
+            carry, llh, steps = self.pipelined_sweep(carry)
             timer.stop()
             res.llh_history.append(llh)
             res.sweeps += 1

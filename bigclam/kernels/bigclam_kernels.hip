@@ -743,6 +743,231 @@ extern "C" __global__ void __launch_bounds__(BLOCK) k4_llh_only_bf16(
   }
 }
 
+// ------------------------------------------------------------------- KF
+//
+// Fused K1+K2: one block per node runs the gradient pass (phase A, K1
+// structure) and the 16-candidate line search (phase B, K2 structure)
+// back-to-back.  grad_u is finalized IN LDS and consumed there by phase B
+// — the separate-kernel flow wrote grad to HBM in K1 and re-staged it in
+// K2 (2 x N x K x 4B), re-staged fu (N x K x 4B) and round-tripped llh.
+// grad still goes to HBM once (K3 needs it); llh[u] likewise.
+// Valid under Jacobi semantics because F is read-only for the whole
+// kernel: node u's trial evaluations only need u's OWN gradient.
+// LDS: gacc/grad K floats + fu_s K floats (+1 KB statics) -> K <= 8192
+// fp32 via the NSLOT template (the K=5000 / K<=500 headline configs).
+
+template <int NSLOT>
+__global__ void __launch_bounds__(BLOCK, 3) kf_fused_t(
+    const float* __restrict__ F, const long long* __restrict__ indptr,
+    const int* __restrict__ indices, const float* __restrict__ sumF,
+    const int* __restrict__ order, float* __restrict__ grad,
+    double* __restrict__ llh, const float* __restrict__ ladder,
+    float* __restrict__ best, int n_local, int K, int n_ladder, float alpha,
+    float min_p, float max_p, float min_f, float max_f) {
+  const int u = order[blockIdx.x];
+  const long long e0 = indptr[u];
+  const long long e1 = indptr[u + 1];
+  const int tid = threadIdx.x;
+  const int lane = tid & (WAVE - 1);
+  const int wid = tid >> 6;
+
+  __shared__ __attribute__((aligned(16))) float s_lad[MAX_LS];
+  __shared__ __attribute__((aligned(16))) double acc_llh[NWAVE][MAX_LS];
+  __shared__ __attribute__((aligned(16))) float acc_nt[NWAVE][MAX_LS];
+  __shared__ __attribute__((aligned(16))) float red[NWAVE];
+  __shared__ double s_llh_u;
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  float* gacc = reinterpret_cast<float*>(smem);  // K floats: grad after A
+  float* fu_s = gacc + K;                        // K floats: fu copy
+
+  if (tid < MAX_LS) s_lad[tid] = (tid < n_ladder) ? ladder[tid] : 0.f;
+
+  // ---------------- phase A: fused gradient + LLH (K1 structure)
+  const float* __restrict__ fu_g = F + (size_t)u * K;
+  float4 fu4[NSLOT];
+#pragma unroll
+  for (int sl = 0; sl < NSLOT; ++sl) {
+    const int k = tid * 4 + sl * (BLOCK * 4);
+    fu4[sl] = (k < K) ? ld4(fu_g + k) : float4{0.f, 0.f, 0.f, 0.f};
+    if (k < K) {
+      *reinterpret_cast<float4*>(gacc + k) = float4{0.f, 0.f, 0.f, 0.f};
+      *reinterpret_cast<float4*>(fu_s + k) = fu4[sl];
+    }
+  }
+  __syncthreads();
+
+  double llh_acc = 0.0;
+#pragma clang loop unroll(disable)
+  for (long long e = e0; e < e1; ++e) {
+    const float* __restrict__ fv = F + (size_t)indices[e] * K;
+    float4 b[NSLOT];
+    float part = 0.f;
+#pragma unroll
+    for (int sl = 0; sl < NSLOT; ++sl) {
+      const int k = tid * 4 + sl * (BLOCK * 4);
+      b[sl] = (k < K) ? ld4(fv + k) : float4{0.f, 0.f, 0.f, 0.f};
+      part = dot4(fu4[sl], b[sl], part);
+    }
+    const float x = block_allreduce_sum(part, red);
+    const float p = clamp_p(__expf(-x), min_p, max_p);
+    const float w = 1.f / (1.f - p);
+    if (tid == 0) llh_acc += (double)log1pf(-p) + (double)x;
+#pragma unroll
+    for (int sl = 0; sl < NSLOT; ++sl) {
+      const int k = tid * 4 + sl * (BLOCK * 4);
+      if (k < K) {
+        float4 g = ld4(gacc + k);
+        g.x = fmaf(w, b[sl].x, g.x);
+        g.y = fmaf(w, b[sl].y, g.y);
+        g.z = fmaf(w, b[sl].z, g.z);
+        g.w = fmaf(w, b[sl].w, g.w);
+        *reinterpret_cast<float4*>(gacc + k) = g;
+      }
+    }
+  }
+
+  float p_fs = 0.f, p_ff = 0.f;
+#pragma unroll
+  for (int sl = 0; sl < NSLOT; ++sl) {
+    const int k = tid * 4 + sl * (BLOCK * 4);
+    if (k < K) {
+      const float4 sv = ld4(sumF + k);
+      p_fs = dot4(fu4[sl], sv, p_fs);
+      p_ff = dot4(fu4[sl], fu4[sl], p_ff);
+    }
+  }
+  const float fs = block_allreduce_sum(p_fs, red);
+  const float ff = block_allreduce_sum(p_ff, red);
+
+  // finalize grad in LDS + write to HBM (K3 needs it); publish llh_u
+  float* __restrict__ gout = grad + (size_t)u * K;
+#pragma unroll
+  for (int sl = 0; sl < NSLOT; ++sl) {
+    const int k = tid * 4 + sl * (BLOCK * 4);
+    if (k < K) {
+      const float4 ga = ld4(gacc + k);
+      const float4 sv = ld4(sumF + k);
+      // same evaluation order as K1 ((gacc - sumF) + fu) for bitwise parity
+      const float4 g = float4{ga.x - sv.x + fu4[sl].x, ga.y - sv.y + fu4[sl].y,
+                              ga.z - sv.z + fu4[sl].z,
+                              ga.w - sv.w + fu4[sl].w};
+      *reinterpret_cast<float4*>(gacc + k) = g;
+      *reinterpret_cast<float4*>(gout + k) = g;
+    }
+  }
+  if (tid == 0) {
+    const double l = llh_acc + (double)(-fs) + (double)ff;
+    llh[u] = l;
+    s_llh_u = l;
+  }
+  __syncthreads();  // gacc = final grad, fu_s, s_llh_u visible
+
+  // ---------------- phase B: 16-candidate line search (K2 structure)
+  const float* __restrict__ fu = fu_s;
+  const float* __restrict__ gu = gacc;
+  float s[MAX_LS];
+#pragma unroll
+  for (int j = 0; j < MAX_LS; ++j) s[j] = s_lad[j];
+  const int jmine = (((lane >> 5) & 1) << 3) | (((lane >> 4) & 1) << 2) |
+                    (((lane >> 3) & 1) << 1) | ((lane >> 2) & 1);
+  double llh_mine = 0.0;
+  const v2f lo2 = {min_f, min_f}, hi2 = {max_f, max_f};
+
+  for (long long e = e0 + wid; e < e1; e += NWAVE) {
+    const float* __restrict__ fv = F + (size_t)indices[e] * K;
+    v2f acc2[MAX_LS];
+#pragma unroll
+    for (int j = 0; j < MAX_LS; ++j) acc2[j] = v2f{0.f, 0.f};
+    constexpr int U = 4;
+    constexpr int KSTR = WAVE * 4;
+    for (int k = lane * 4; k < K; k += U * KSTR) {
+      float4 b[U];
+#pragma unroll
+      for (int t = 0; t < U; ++t) {
+        const int kk = k + t * KSTR;
+        b[t] = ld4(fv + (kk < K ? kk : (K - 4)));
+      }
+#pragma unroll
+      for (int t = 0; t < U; ++t) {
+        if (k + t * KSTR >= K) break;
+        const float4 a4 = ld4(fu + k + t * KSTR);
+        const float4 g4 = ld4(gu + k + t * KSTR);
+        const v2f b0 = {b[t].x, b[t].y}, b1 = {b[t].z, b[t].w};
+        const v2f a0 = {a4.x, a4.y}, a1 = {a4.z, a4.w};
+        const v2f g0 = {g4.x, g4.y}, g1 = {g4.z, g4.w};
+#pragma unroll
+        for (int j = 0; j < MAX_LS; ++j) {
+          const v2f sj = {s[j], s[j]};
+          v2f t2 = acc2[j];
+          t2 = __builtin_elementwise_fma(pk_clamp_fma(sj, g0, a0, lo2, hi2),
+                                         b0, t2);
+          t2 = __builtin_elementwise_fma(pk_clamp_fma(sj, g1, a1, lo2, hi2),
+                                         b1, t2);
+          acc2[j] = t2;
+        }
+      }
+    }
+    float accf[MAX_LS];
+#pragma unroll
+    for (int j = 0; j < MAX_LS; ++j) accf[j] = acc2[j].x + acc2[j].y;
+    wave_reduce16(accf, lane);
+    if ((lane & 3) == 0) {
+      const float x = accf[0];
+      const float p = clamp_p(__expf(-x), min_p, max_p);
+      llh_mine += (double)log1pf(-p) + (double)x;
+    }
+  }
+
+  v2f accn2[MAX_LS];
+#pragma unroll
+  for (int j = 0; j < MAX_LS; ++j) accn2[j] = v2f{0.f, 0.f};
+  float p_gg = 0.f;
+  for (int k = tid * 4; k < K; k += BLOCK * 4) {
+    const float4 a = ld4(fu + k);
+    const float4 g = ld4(gu + k);
+    const float4 sf = ld4(sumF + k);
+    const v2f a0 = {a.x, a.y}, a1 = {a.z, a.w};
+    const v2f g0 = {g.x, g.y}, g1 = {g.z, g.w};
+    const v2f d0 = {a.x - sf.x, a.y - sf.y};
+    const v2f d1 = {a.z - sf.z, a.w - sf.w};
+    p_gg = dot4(g, g, p_gg);
+#pragma unroll
+    for (int j = 0; j < MAX_LS; ++j) {
+      const v2f sj = {s[j], s[j]};
+      v2f t = accn2[j];
+      t = __builtin_elementwise_fma(pk_clamp_fma(sj, g0, a0, lo2, hi2), d0,
+                                    t);
+      t = __builtin_elementwise_fma(pk_clamp_fma(sj, g1, a1, lo2, hi2), d1,
+                                    t);
+      accn2[j] = t;
+    }
+  }
+  float accn[MAX_LS];
+#pragma unroll
+  for (int j = 0; j < MAX_LS; ++j) accn[j] = accn2[j].x + accn2[j].y;
+  wave_reduce16(accn, lane);
+  if ((lane & 3) == 0) {
+    acc_nt[wid][jmine] = accn[0];
+    acc_llh[wid][jmine] = llh_mine;
+  }
+  const float gg = block_allreduce_sum(p_gg, red);  // ends in __syncthreads
+
+  if (wid == 0) {
+    bool ok = false;
+    if (lane < MAX_LS) {
+      double trial = 0.0;
+#pragma unroll
+      for (int wv = 0; wv < NWAVE; ++wv)
+        trial += acc_llh[wv][lane] + (double)acc_nt[wv][lane];
+      ok = (lane < n_ladder) &&
+           (trial >= s_llh_u + (double)(alpha * s_lad[lane] * gg));
+    }
+    const unsigned long long bal = __ballot(ok);
+    if (lane == 0)
+      best[u] = bal ? s_lad[__ffsll((unsigned long long)bal) - 1] : 0.f;
+  }
+}
+
 // K2 tiled variant for K too large to stage whole rows (fp32 K > ~20k,
 // e.g. the com-Amazon K=25000 config): fu/grad are staged per 2048-element
 // k-chunk (16 KB LDS) with the edge loop INSIDE the chunk loop; per-edge
@@ -1379,5 +1604,37 @@ extern "C" void launch_k5(const long long* indptr, const int* indices,
   if (n == 0) return;
   hipLaunchKernelGGL(k5_conductance, dim3(n), dim3(256), 0, stream, indptr,
                      indices, cond, n, total_degree);
+  HIP_CHECK(hipGetLastError());
+}
+
+extern "C" void launch_kf(const float* F, const long long* indptr,
+                          const int* indices, const float* sumF,
+                          const int* order, float* grad, double* llh,
+                          const float* ladder, float* best, int n_local,
+                          int K, int n_ladder, float alpha, float min_p,
+                          float max_p, float min_f, float max_f,
+                          hipStream_t stream) {
+  if (n_local == 0) return;
+  if (n_ladder > 16) throw std::runtime_error("ladder length > 16 unsupported");
+  if (K > 8192) throw std::runtime_error("fused kernel requires K <= 8192");
+  const size_t lds = (size_t)K * 8;  // gacc + fu_s
+#define KF_CASE(NS)                                                          \
+  do {                                                                       \
+    allow_large_lds((const void*)&kf_fused_t<NS>, lds);                      \
+    hipLaunchKernelGGL((kf_fused_t<NS>), dim3(n_local), dim3(256), lds,      \
+                       stream, F, indptr, indices, sumF, order, grad, llh,   \
+                       ladder, best, n_local, K, n_ladder, alpha, min_p,     \
+                       max_p, min_f, max_f);                                 \
+  } while (0)
+  if (K <= 1024) {
+    KF_CASE(1);
+  } else if (K <= 2048) {
+    KF_CASE(2);
+  } else if (K <= 4096) {
+    KF_CASE(4);
+  } else {
+    KF_CASE(8);
+  }
+#undef KF_CASE
   HIP_CHECK(hipGetLastError());
 }

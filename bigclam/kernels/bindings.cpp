@@ -31,6 +31,10 @@ extern "C" void launch_k3_bf16(void*, const float*, const float*, int, int,
                                float, float, hipStream_t);
 extern "C" void launch_k5(const long long*, const int*, double*, int, double,
                           hipStream_t);
+extern "C" void launch_kf(const float*, const long long*, const int*,
+                          const float*, const int*, float*, double*,
+                          const float*, float*, int, int, int, float, float,
+                          float, float, float, hipStream_t);
 
 namespace {
 
@@ -172,6 +176,35 @@ void apply_step(torch::Tensor F_local, torch::Tensor grad,
   }
 }
 
+void fused_grad_ls(torch::Tensor F, torch::Tensor indptr,
+                   torch::Tensor indices, torch::Tensor sumF,
+                   torch::Tensor order, torch::Tensor grad, torch::Tensor llh,
+                   torch::Tensor ladder, torch::Tensor best, double alpha,
+                   double min_p, double max_p, double min_f, double max_f) {
+  CHECK_IN(F, torch::kFloat32);
+  CHECK_IN(indptr, torch::kInt64);
+  CHECK_IN(indices, torch::kInt32);
+  CHECK_IN(sumF, torch::kFloat32);
+  CHECK_IN(order, torch::kInt32);
+  CHECK_IN(grad, torch::kFloat32);
+  CHECK_IN(llh, torch::kFloat64);
+  CHECK_IN(ladder, torch::kFloat32);
+  CHECK_IN(best, torch::kFloat32);
+  const int n_local = (int)indptr.size(0) - 1;
+  const int n_blocks = (int)order.size(0);
+  const int K = (int)F.size(1);
+  TORCH_CHECK(K % 4 == 0 && K <= 8192, "fused kernel: K padded, K <= 8192");
+  TORCH_CHECK(grad.size(0) == n_local && grad.size(1) == K);
+  launch_kf(F.data_ptr<float>(),
+            reinterpret_cast<const long long*>(indptr.data_ptr<int64_t>()),
+            indices.data_ptr<int>(), sumF.data_ptr<float>(),
+            order.data_ptr<int>(), grad.data_ptr<float>(),
+            llh.data_ptr<double>(), ladder.data_ptr<float>(),
+            best.data_ptr<float>(), n_blocks, K, (int)ladder.size(0),
+            (float)alpha, (float)min_p, (float)max_p, (float)min_f,
+            (float)max_f, current_stream());
+}
+
 void conductance(torch::Tensor indptr, torch::Tensor indices,
                  torch::Tensor cond, double total_degree) {
   CHECK_IN(indptr, torch::kInt64);
@@ -196,4 +229,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "K3: in-place projected commit F += s*grad (CDNA4)");
   m.def("conductance", &conductance,
         "K5: ego-net conductance per node (CDNA4)");
+  m.def("fused_grad_ls", &fused_grad_ls,
+        "KF: fused K1 gradient+LLH and K2 line search, one pass per node");
 }

@@ -62,6 +62,46 @@ def edge_grad_llh(
 _ladder_cache = {}
 
 
+def _ladder(cfg: BigClamConfig, device) -> torch.Tensor:
+    key = (tuple(cfg.ladder()), str(device))
+    t = _ladder_cache.get(key)
+    if t is None:  # cached: a fresh HtoD copy per sweep costs a sync
+        t = torch.tensor(cfg.ladder(), device=device, dtype=torch.float32)
+        _ladder_cache[key] = t
+    return t
+
+
+def fused_grad_ls(
+    F: torch.Tensor,
+    indptr: torch.Tensor,
+    indices: torch.Tensor,
+    sumF: torch.Tensor,
+    order: torch.Tensor,
+    cfg: BigClamConfig,
+    out=None,
+):
+    """KF: fused K1 gradient+LLH and K2 line search in one per-node pass.
+
+    Returns (grad [n,K], llh [n] f64, best_step [n]).  ``order`` may be a
+    node subset (halo overlap) writing into shared ``out`` buffers."""
+    ext = ensure_loaded()
+    n_local = len(indptr) - 1
+    if out is None:
+        grad = torch.empty(
+            n_local, F.shape[1], device=F.device, dtype=torch.float32
+        )
+        llh = torch.empty(n_local, device=F.device, dtype=torch.float64)
+        best = torch.empty(n_local, device=F.device, dtype=torch.float32)
+    else:
+        grad, llh, best = out
+    ext.fused_grad_ls(
+        F, indptr, indices, sumF, order, grad, llh,
+        _ladder(cfg, F.device), best,
+        cfg.alpha, cfg.min_p, cfg.max_p, cfg.min_f, cfg.max_f,
+    )
+    return grad, llh, best
+
+
 def linesearch(
     F: torch.Tensor,
     indptr: torch.Tensor,
@@ -75,11 +115,7 @@ def linesearch(
     ext = ensure_loaded()
     n_local = len(indptr) - 1
     best = torch.empty(n_local, device=F.device, dtype=torch.float32)
-    key = (tuple(cfg.ladder()), str(F.device))
-    ladder = _ladder_cache.get(key)
-    if ladder is None:  # cached: a fresh HtoD copy per sweep costs a sync
-        ladder = torch.tensor(cfg.ladder(), device=F.device, dtype=torch.float32)
-        _ladder_cache[key] = ladder
+    ladder = _ladder(cfg, F.device)
     ext.linesearch(
         F,
         indptr,

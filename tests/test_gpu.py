@@ -212,3 +212,20 @@ def test_k2_tiled_matches_reference(monkeypatch):
     )
     agree = (best == rbest).float().mean().item()
     assert agree > 0.98, f"only {agree:.3f} of best-steps agree"
+
+
+def test_fused_matches_separate_kernels():
+    """KF (fused K1+K2) == the separate K1 then K2 launches, bitwise."""
+    from bigclam.ops import hip as hip_ops
+
+    g = rmat_graph(10, 7.0, seed=51)
+    cfg, st = _mkstate(g, 192, seed=15)
+    assert st.fused_ok
+    grad_s, llh_s = st.grad_llh()
+    best_s = st.linesearch(grad_s, llh_s)
+    grad_f, llh_f, best_f = hip_ops.fused_grad_ls(
+        st.F, st.indptr, st.indices, st.sumF, st.order, cfg
+    )
+    torch.testing.assert_close(grad_f, grad_s, rtol=0, atol=0)
+    torch.testing.assert_close(llh_f, llh_s, rtol=0, atol=0)
+    torch.testing.assert_close(best_f, best_s, rtol=0, atol=0)
