@@ -177,13 +177,14 @@ class ShardState:
 
     @property
     def fused_ok(self) -> bool:
-        """The fused K1+K2 kernel covers fp32 rows up to K=8192 (LDS holds
-        grad + fu); other shapes use the separate kernels."""
-        return (
-            self.use_hip
-            and self.storage_dtype == torch.float32
-            and self.kp <= 8192
-        )
+        """The fused K1+K2 kernel covers fp32 rows up to K=8192 and bf16
+        rows up to K=16384 (LDS holds grad + fu); other shapes use the
+        separate kernels."""
+        if not self.use_hip:
+            return False
+        if self.storage_dtype == torch.float32:
+            return self.kp <= 8192
+        return self.kp <= 16384
 
     def fused_grad_ls_overlap(
         self, halo_work

@@ -968,6 +968,253 @@ __global__ void __launch_bounds__(BLOCK, 3) kf_fused_t(
   }
 }
 
+// bf16 fused K1+K2 (same structure as kf_fused_t): gacc fp32 K*4B +
+// fu raw bf16 K*2B LDS -> covers K <= 16384 via the NSLOT template.
+template <int NSLOT>
+__global__ void __launch_bounds__(BLOCK, 3) kf_fused_bf16_t(
+    const u32* __restrict__ F, const long long* __restrict__ indptr,
+    const int* __restrict__ indices, const float* __restrict__ sumF,
+    const int* __restrict__ order, float* __restrict__ grad,
+    double* __restrict__ llh, const float* __restrict__ ladder,
+    float* __restrict__ best, int n_local, int K, int n_ladder, float alpha,
+    float min_p, float max_p, float min_f, float max_f) {
+  const int u = order[blockIdx.x];
+  const long long e0 = indptr[u];
+  const long long e1 = indptr[u + 1];
+  const int tid = threadIdx.x;
+  const int lane = tid & (WAVE - 1);
+  const int wid = tid >> 6;
+
+  __shared__ __attribute__((aligned(16))) float s_lad[MAX_LS];
+  __shared__ __attribute__((aligned(16))) double acc_llh[NWAVE][MAX_LS];
+  __shared__ __attribute__((aligned(16))) float acc_nt[NWAVE][MAX_LS];
+  __shared__ __attribute__((aligned(16))) float red[NWAVE];
+  __shared__ double s_llh_u;
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  float* gacc = reinterpret_cast<float*>(smem);               // K floats
+  u32* fu_s = reinterpret_cast<u32*>(smem + (size_t)K * 4);   // K/2 u32
+
+  if (tid < MAX_LS) s_lad[tid] = (tid < n_ladder) ? ladder[tid] : 0.f;
+
+  // ---------------- phase A (k1_grad_llh_bf16_t structure)
+  const u32* __restrict__ fu_g = F + (size_t)u * (K / 2);
+  f32x8 fu8[NSLOT];
+#pragma unroll
+  for (int sl = 0; sl < NSLOT; ++sl) {
+    const int k = tid * 8 + sl * (BLOCK * 8);
+    if (k < K) {
+      const uint4 raw = *reinterpret_cast<const uint4*>(fu_g + k / 2);
+      *reinterpret_cast<uint4*>(fu_s + k / 2) = raw;
+      fu8[sl].a = float4{bf_lo(raw.x), bf_hi(raw.x), bf_lo(raw.y),
+                         bf_hi(raw.y)};
+      fu8[sl].b = float4{bf_lo(raw.z), bf_hi(raw.z), bf_lo(raw.w),
+                         bf_hi(raw.w)};
+      *reinterpret_cast<float4*>(gacc + k) = float4{0.f, 0.f, 0.f, 0.f};
+      *reinterpret_cast<float4*>(gacc + k + 4) = float4{0.f, 0.f, 0.f, 0.f};
+    } else {
+      fu8[sl].a = fu8[sl].b = float4{0.f, 0.f, 0.f, 0.f};
+    }
+  }
+  __syncthreads();
+
+  double llh_acc = 0.0;
+#pragma clang loop unroll(disable)
+  for (long long e = e0; e < e1; ++e) {
+    const u32* __restrict__ fv = F + (size_t)indices[e] * (K / 2);
+    uint4 braw[NSLOT];
+    float part = 0.f;
+#pragma unroll
+    for (int sl = 0; sl < NSLOT; ++sl) {
+      const int k = tid * 8 + sl * (BLOCK * 8);
+      braw[sl] = (k < K) ? *reinterpret_cast<const uint4*>(fv + k / 2)
+                         : uint4{0u, 0u, 0u, 0u};
+      f32x8 b;
+      b.a = float4{bf_lo(braw[sl].x), bf_hi(braw[sl].x), bf_lo(braw[sl].y),
+                   bf_hi(braw[sl].y)};
+      b.b = float4{bf_lo(braw[sl].z), bf_hi(braw[sl].z), bf_lo(braw[sl].w),
+                   bf_hi(braw[sl].w)};
+      part = dot8(fu8[sl], b, part);
+    }
+    const float x = block_allreduce_sum(part, red);
+    const float p = clamp_p(__expf(-x), min_p, max_p);
+    const float w = 1.f / (1.f - p);
+    if (tid == 0) llh_acc += (double)log1pf(-p) + (double)x;
+#pragma unroll
+    for (int sl = 0; sl < NSLOT; ++sl) {
+      const int k = tid * 8 + sl * (BLOCK * 8);
+      if (k < K) {
+        f32x8 b;
+        b.a = float4{bf_lo(braw[sl].x), bf_hi(braw[sl].x), bf_lo(braw[sl].y),
+                     bf_hi(braw[sl].y)};
+        b.b = float4{bf_lo(braw[sl].z), bf_hi(braw[sl].z), bf_lo(braw[sl].w),
+                     bf_hi(braw[sl].w)};
+        float4 g0 = ld4(gacc + k);
+        float4 g1 = ld4(gacc + k + 4);
+        g0.x = fmaf(w, b.a.x, g0.x);
+        g0.y = fmaf(w, b.a.y, g0.y);
+        g0.z = fmaf(w, b.a.z, g0.z);
+        g0.w = fmaf(w, b.a.w, g0.w);
+        g1.x = fmaf(w, b.b.x, g1.x);
+        g1.y = fmaf(w, b.b.y, g1.y);
+        g1.z = fmaf(w, b.b.z, g1.z);
+        g1.w = fmaf(w, b.b.w, g1.w);
+        *reinterpret_cast<float4*>(gacc + k) = g0;
+        *reinterpret_cast<float4*>(gacc + k + 4) = g1;
+      }
+    }
+  }
+
+  float p_fs = 0.f, p_ff = 0.f;
+#pragma unroll
+  for (int sl = 0; sl < NSLOT; ++sl) {
+    const int k = tid * 8 + sl * (BLOCK * 8);
+    if (k < K) {
+      const float4 s0 = ld4(sumF + k);
+      const float4 s1 = ld4(sumF + k + 4);
+      p_fs = dot4(fu8[sl].a, s0, p_fs);
+      p_fs = dot4(fu8[sl].b, s1, p_fs);
+      p_ff = dot8(fu8[sl], fu8[sl], p_ff);
+    }
+  }
+  const float fs = block_allreduce_sum(p_fs, red);
+  const float ff = block_allreduce_sum(p_ff, red);
+
+  float* __restrict__ gout = grad + (size_t)u * K;
+#pragma unroll
+  for (int sl = 0; sl < NSLOT; ++sl) {
+    const int k = tid * 8 + sl * (BLOCK * 8);
+    if (k < K) {
+      const float4 g0a = ld4(gacc + k);
+      const float4 g1a = ld4(gacc + k + 4);
+      const float4 s0 = ld4(sumF + k);
+      const float4 s1 = ld4(sumF + k + 4);
+      const float4 g0 =
+          float4{g0a.x - s0.x + fu8[sl].a.x, g0a.y - s0.y + fu8[sl].a.y,
+                 g0a.z - s0.z + fu8[sl].a.z, g0a.w - s0.w + fu8[sl].a.w};
+      const float4 g1 =
+          float4{g1a.x - s1.x + fu8[sl].b.x, g1a.y - s1.y + fu8[sl].b.y,
+                 g1a.z - s1.z + fu8[sl].b.z, g1a.w - s1.w + fu8[sl].b.w};
+      *reinterpret_cast<float4*>(gacc + k) = g0;
+      *reinterpret_cast<float4*>(gacc + k + 4) = g1;
+      *reinterpret_cast<float4*>(gout + k) = g0;
+      *reinterpret_cast<float4*>(gout + k + 4) = g1;
+    }
+  }
+  if (tid == 0) {
+    const double l = llh_acc + (double)(-fs) + (double)ff;
+    llh[u] = l;
+    s_llh_u = l;
+  }
+  __syncthreads();
+
+  // ---------------- phase B (k2_ls_v3_bf16 structure; fu/g from LDS)
+  const u32* __restrict__ fu = fu_s;
+  const float* __restrict__ gu = gacc;
+  float s[MAX_LS];
+#pragma unroll
+  for (int j = 0; j < MAX_LS; ++j) s[j] = s_lad[j];
+  const int jmine = (((lane >> 5) & 1) << 3) | (((lane >> 4) & 1) << 2) |
+                    (((lane >> 3) & 1) << 1) | ((lane >> 2) & 1);
+  double llh_mine = 0.0;
+  const v2f lo2 = {min_f, min_f}, hi2 = {max_f, max_f};
+
+  for (long long e = e0 + wid; e < e1; e += NWAVE) {
+    const u32* __restrict__ fv = F + (size_t)indices[e] * (K / 2);
+    v2f acc[MAX_LS];
+#pragma unroll
+    for (int j = 0; j < MAX_LS; ++j) acc[j] = v2f{0.f, 0.f};
+    constexpr int KSTR = WAVE * 8;
+    for (int k = lane * 8; k < K; k += KSTR) {
+      const uint4 braw = *reinterpret_cast<const uint4*>(fv + k / 2);
+      const uint4 araw = *reinterpret_cast<const uint4*>(fu + k / 2);
+      const float4 gA = ld4(gu + k);
+      const float4 gB = ld4(gu + k + 4);
+      const v2f b0 = bf2(braw.x), b1 = bf2(braw.y), b2 = bf2(braw.z),
+                b3 = bf2(braw.w);
+      const v2f a0 = bf2(araw.x), a1 = bf2(araw.y), a2 = bf2(araw.z),
+                a3 = bf2(araw.w);
+      const v2f g0 = {gA.x, gA.y}, g1 = {gA.z, gA.w};
+      const v2f g2 = {gB.x, gB.y}, g3 = {gB.z, gB.w};
+#pragma unroll
+      for (int j = 0; j < MAX_LS; ++j) {
+        const v2f sj = {s[j], s[j]};
+        v2f t2 = acc[j];
+        t2 = __builtin_elementwise_fma(pk_clamp_fma(sj, g0, a0, lo2, hi2),
+                                       b0, t2);
+        t2 = __builtin_elementwise_fma(pk_clamp_fma(sj, g1, a1, lo2, hi2),
+                                       b1, t2);
+        t2 = __builtin_elementwise_fma(pk_clamp_fma(sj, g2, a2, lo2, hi2),
+                                       b2, t2);
+        t2 = __builtin_elementwise_fma(pk_clamp_fma(sj, g3, a3, lo2, hi2),
+                                       b3, t2);
+        acc[j] = t2;
+      }
+    }
+    float accf[MAX_LS];
+#pragma unroll
+    for (int j = 0; j < MAX_LS; ++j) accf[j] = acc[j].x + acc[j].y;
+    wave_reduce16(accf, lane);
+    if ((lane & 3) == 0) {
+      const float x = accf[0];
+      const float p = clamp_p(__expf(-x), min_p, max_p);
+      llh_mine += (double)log1pf(-p) + (double)x;
+    }
+  }
+
+  v2f accn2[MAX_LS];
+#pragma unroll
+  for (int j = 0; j < MAX_LS; ++j) accn2[j] = v2f{0.f, 0.f};
+  float p_gg = 0.f;
+  for (int k = tid * 8; k < K; k += BLOCK * 8) {
+    const uint4 araw = *reinterpret_cast<const uint4*>(fu + k / 2);
+    const v2f a0 = bf2(araw.x), a1 = bf2(araw.y), a2 = bf2(araw.z),
+              a3 = bf2(araw.w);
+    const float4 gA = ld4(gu + k);
+    const float4 gB = ld4(gu + k + 4);
+    const v2f g0 = {gA.x, gA.y}, g1 = {gA.z, gA.w};
+    const v2f g2 = {gB.x, gB.y}, g3 = {gB.z, gB.w};
+    const float4 sA = ld4(sumF + k);
+    const float4 sB = ld4(sumF + k + 4);
+    const v2f d0 = a0 - v2f{sA.x, sA.y}, d1 = a1 - v2f{sA.z, sA.w};
+    const v2f d2 = a2 - v2f{sB.x, sB.y}, d3 = a3 - v2f{sB.z, sB.w};
+    p_gg = dot4(gA, gA, dot4(gB, gB, p_gg));
+#pragma unroll
+    for (int j = 0; j < MAX_LS; ++j) {
+      const v2f sj = {s[j], s[j]};
+      v2f t = accn2[j];
+      t = __builtin_elementwise_fma(pk_clamp_fma(sj, g0, a0, lo2, hi2), d0, t);
+      t = __builtin_elementwise_fma(pk_clamp_fma(sj, g1, a1, lo2, hi2), d1, t);
+      t = __builtin_elementwise_fma(pk_clamp_fma(sj, g2, a2, lo2, hi2), d2, t);
+      t = __builtin_elementwise_fma(pk_clamp_fma(sj, g3, a3, lo2, hi2), d3, t);
+      accn2[j] = t;
+    }
+  }
+  float accn[MAX_LS];
+#pragma unroll
+  for (int j = 0; j < MAX_LS; ++j) accn[j] = accn2[j].x + accn2[j].y;
+  wave_reduce16(accn, lane);
+  if ((lane & 3) == 0) {
+    acc_nt[wid][jmine] = accn[0];
+    acc_llh[wid][jmine] = llh_mine;
+  }
+  const float gg = block_allreduce_sum(p_gg, red);  // ends in __syncthreads
+
+  if (wid == 0) {
+    bool ok = false;
+    if (lane < MAX_LS) {
+      double trial = 0.0;
+#pragma unroll
+      for (int wv = 0; wv < NWAVE; ++wv)
+        trial += acc_llh[wv][lane] + (double)acc_nt[wv][lane];
+      ok = (lane < n_ladder) &&
+           (trial >= s_llh_u + (double)(alpha * s_lad[lane] * gg));
+    }
+    const unsigned long long bal = __ballot(ok);
+    if (lane == 0)
+      best[u] = bal ? s_lad[__ffsll((unsigned long long)bal) - 1] : 0.f;
+  }
+}
+
 // K2 tiled variant for K too large to stage whole rows (fp32 K > ~20k,
 // e.g. the com-Amazon K=25000 config): fu/grad are staged per 2048-element
 // k-chunk (16 KB LDS) with the edge loop INSIDE the chunk loop; per-edge
@@ -1636,5 +1883,38 @@ extern "C" void launch_kf(const float* F, const long long* indptr,
     KF_CASE(8);
   }
 #undef KF_CASE
+  HIP_CHECK(hipGetLastError());
+}
+
+extern "C" void launch_kf_bf16(const void* F, const long long* indptr,
+                               const int* indices, const float* sumF,
+                               const int* order, float* grad, double* llh,
+                               const float* ladder, float* best, int n_local,
+                               int K, int n_ladder, float alpha, float min_p,
+                               float max_p, float min_f, float max_f,
+                               hipStream_t stream) {
+  if (n_local == 0) return;
+  if (n_ladder > 16) throw std::runtime_error("ladder length > 16 unsupported");
+  if (K > 16384) throw std::runtime_error("bf16 fused kernel requires K <= 16384");
+  const size_t lds = (size_t)K * 6;  // gacc fp32 + fu raw bf16
+  const u32* Fb = reinterpret_cast<const u32*>(F);
+#define KFB_CASE(NS)                                                          \
+  do {                                                                        \
+    allow_large_lds((const void*)&kf_fused_bf16_t<NS>, lds);                  \
+    hipLaunchKernelGGL((kf_fused_bf16_t<NS>), dim3(n_local), dim3(256), lds,  \
+                       stream, Fb, indptr, indices, sumF, order, grad, llh,   \
+                       ladder, best, n_local, K, n_ladder, alpha, min_p,      \
+                       max_p, min_f, max_f);                                  \
+  } while (0)
+  if (K <= 2048) {
+    KFB_CASE(1);
+  } else if (K <= 4096) {
+    KFB_CASE(2);
+  } else if (K <= 8192) {
+    KFB_CASE(4);
+  } else {
+    KFB_CASE(8);
+  }
+#undef KFB_CASE
   HIP_CHECK(hipGetLastError());
 }

@@ -35,6 +35,10 @@ extern "C" void launch_kf(const float*, const long long*, const int*,
                           const float*, const int*, float*, double*,
                           const float*, float*, int, int, int, float, float,
                           float, float, float, hipStream_t);
+extern "C" void launch_kf_bf16(const void*, const long long*, const int*,
+                               const float*, const int*, float*, double*,
+                               const float*, float*, int, int, int, float,
+                               float, float, float, float, hipStream_t);
 
 namespace {
 
@@ -181,7 +185,7 @@ void fused_grad_ls(torch::Tensor F, torch::Tensor indptr,
                    torch::Tensor order, torch::Tensor grad, torch::Tensor llh,
                    torch::Tensor ladder, torch::Tensor best, double alpha,
                    double min_p, double max_p, double min_f, double max_f) {
-  CHECK_IN(F, torch::kFloat32);
+  CHECK_F(F);
   CHECK_IN(indptr, torch::kInt64);
   CHECK_IN(indices, torch::kInt32);
   CHECK_IN(sumF, torch::kFloat32);
@@ -193,16 +197,28 @@ void fused_grad_ls(torch::Tensor F, torch::Tensor indptr,
   const int n_local = (int)indptr.size(0) - 1;
   const int n_blocks = (int)order.size(0);
   const int K = (int)F.size(1);
-  TORCH_CHECK(K % 4 == 0 && K <= 8192, "fused kernel: K padded, K <= 8192");
   TORCH_CHECK(grad.size(0) == n_local && grad.size(1) == K);
-  launch_kf(F.data_ptr<float>(),
-            reinterpret_cast<const long long*>(indptr.data_ptr<int64_t>()),
-            indices.data_ptr<int>(), sumF.data_ptr<float>(),
-            order.data_ptr<int>(), grad.data_ptr<float>(),
-            llh.data_ptr<double>(), ladder.data_ptr<float>(),
-            best.data_ptr<float>(), n_blocks, K, (int)ladder.size(0),
-            (float)alpha, (float)min_p, (float)max_p, (float)min_f,
-            (float)max_f, current_stream());
+  const auto ip =
+      reinterpret_cast<const long long*>(indptr.data_ptr<int64_t>());
+  if (is_bf16(F)) {
+    TORCH_CHECK(K % 8 == 0 && K <= 16384,
+                "bf16 fused kernel: K padded to 8, K <= 16384");
+    launch_kf_bf16(F.data_ptr(), ip, indices.data_ptr<int>(),
+                   sumF.data_ptr<float>(), order.data_ptr<int>(),
+                   grad.data_ptr<float>(), llh.data_ptr<double>(),
+                   ladder.data_ptr<float>(), best.data_ptr<float>(), n_blocks,
+                   K, (int)ladder.size(0), (float)alpha, (float)min_p,
+                   (float)max_p, (float)min_f, (float)max_f,
+                   current_stream());
+  } else {
+    TORCH_CHECK(K % 4 == 0 && K <= 8192, "fused kernel: K padded, K <= 8192");
+    launch_kf(F.data_ptr<float>(), ip, indices.data_ptr<int>(),
+              sumF.data_ptr<float>(), order.data_ptr<int>(),
+              grad.data_ptr<float>(), llh.data_ptr<double>(),
+              ladder.data_ptr<float>(), best.data_ptr<float>(), n_blocks, K,
+              (int)ladder.size(0), (float)alpha, (float)min_p, (float)max_p,
+              (float)min_f, (float)max_f, current_stream());
+  }
 }
 
 void conductance(torch::Tensor indptr, torch::Tensor indices,

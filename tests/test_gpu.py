@@ -229,3 +229,19 @@ def test_fused_matches_separate_kernels():
     torch.testing.assert_close(grad_f, grad_s, rtol=0, atol=0)
     torch.testing.assert_close(llh_f, llh_s, rtol=0, atol=0)
     torch.testing.assert_close(best_f, best_s, rtol=0, atol=0)
+
+
+def test_fused_bf16_matches_separate_kernels():
+    from bigclam.ops import hip as hip_ops
+
+    g = rmat_graph(10, 7.0, seed=52)
+    cfg, st = _mkstate_dtype(g, 192, "bf16", seed=16)
+    assert st.fused_ok
+    grad_s, llh_s = st.grad_llh()
+    best_s = st.linesearch(grad_s, llh_s)
+    grad_f, llh_f, best_f = hip_ops.fused_grad_ls(
+        st.F, st.indptr, st.indices, st.sumF, st.order, cfg
+    )
+    torch.testing.assert_close(grad_f, grad_s, rtol=0, atol=0)
+    torch.testing.assert_close(llh_f, llh_s, rtol=0, atol=0)
+    torch.testing.assert_close(best_f, best_s, rtol=0, atol=0)
