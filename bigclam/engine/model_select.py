@@ -44,8 +44,9 @@ def select_k(
         llh = res.llh
         history.append({"k": k, "llh": llh, "sweeps": res.sweeps})
         metrics.log({"select_k": k, "llh": llh, "sweeps": res.sweeps})
+        k_for_c = k  # if the grid exhausts without flattening, keep the
+        # largest K tried (the reference loop simply runs out of grid)
         if llh_old is not None and (1.0 - llh / llh_old) < cfg.k_tol:
-            k_for_c = k
             break
         llh_old = llh
     return {"k": k_for_c, "history": history, "grid": ks}

@@ -135,3 +135,18 @@ def test_pipelined_fit_matches_sweep_trajectory(small_graph):
     res = tr2.fit(skip_init=True)
     for a, b in zip(unfused, res.llh_history):
         assert abs(a - b) < 1e-9 * max(1.0, abs(a)), (unfused, res.llh_history)
+
+
+def test_select_k_exhausted_grid_returns_last_k():
+    """If no K flattens the LLH gain, the largest K tried is selected
+    (never 0 — regression guard)."""
+    from bigclam.engine.model_select import select_k
+    from bigclam.io import planted_partition
+
+    g, _ = planted_partition(4, 12, p_in=0.6, p_out=0.02, seed=13)
+    cfg = BigClamConfig(
+        k=4, device="cpu", max_sweeps=4, k_min=2, k_max=8, k_div=2,
+        k_tol=1e-30, seed=3,  # tol so tight the gain never flattens
+    )
+    out = select_k(g, cfg, init="random")
+    assert out["k"] == out["grid"][-1] != 0
