@@ -73,15 +73,10 @@ __device__ __forceinline__ void wave_reduce16(float (&w)[MAX_LS], int lane) {
   w[0] += __shfl_xor(w[0], 1, WAVE);
 }
 
-// After wave_reduce16, value j sits (replicated over a 4-lane group) on the
-// lanes whose bits select it: step k kept value-bit (nv) on the side with
-// lane-bit off=32>>k SET.  Lane l holds value j(l) =
-//   bit3=(l>>5)&1? no — derived: j = ((l&32)?8:0)|((l&16)?4:0)|((l&8)?2:0)|((l&4)?1:0)
-__device__ __forceinline__ int wave_reduce16_owner(int j) {
-  // inverse map: the lane group that holds value j is
-  // l = ((j&8)?32:0)|((j&4)?16:0)|((j&2)?8:0)|((j&1)?4:0)
-  return ((j & 8) << 2) | ((j & 4) << 2) | ((j & 2) << 2) | ((j & 1) << 2);
-}
+// After wave_reduce16, value j sits (replicated over a 4-lane group) on
+// the lanes whose bits select it: lane l holds value
+//   j(l) = ((l>>5)&1)<<3 | ((l>>4)&1)<<2 | ((l>>3)&1)<<1 | ((l>>2)&1)
+// ("jmine" in the kernels; lanes with (l & 3) == 0 are the owners).
 
 // block-wide sum; `red` is NWAVE floats of LDS; result on every thread.
 __device__ __forceinline__ float block_allreduce_sum(float v, float* red) {
