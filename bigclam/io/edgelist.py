@@ -113,15 +113,16 @@ def build_graph(edges: np.ndarray, drop_self_loops: bool = True) -> Graph:
     uniq = np.unique(key)
     lo = (uniq // n).astype(np.int64)
     hi = (uniq % n).astype(np.int64)
-    # both directions
-    src = np.concatenate([lo, hi])
-    dst = np.concatenate([hi, lo])
+    # both directions; one int64 key sort gives row-major + sorted rows
+    # (np.sort on the key is ~2x np.lexsort((dst, src)) at 100M+ edges,
+    # and np.bincount beats np.add.at by ~10x)
+    key2 = np.concatenate([lo * n + hi, hi * n + lo])
+    key2.sort(kind="stable")
+    src = key2 // n
+    indices = (key2 % n).astype(np.int32)
+    counts = np.bincount(src, minlength=n)
     indptr = np.zeros(n + 1, dtype=np.int64)
-    np.add.at(indptr, src + 1, 1)
-    np.cumsum(indptr, out=indptr)
-    # sort neighbors within each row: lexsort by (src, dst)
-    order = np.lexsort((dst, src))
-    indices = dst[order].astype(np.int32)
+    np.cumsum(counts, out=indptr[1:])
     return Graph(indptr=indptr, indices=indices, raw_ids=raw_ids)
 
 

@@ -40,10 +40,10 @@ def rmat_edges(
     a_norm = a / ab
     c_norm = c / (1.0 - ab)
     for _ in range(scale):
-        r1 = rng.random(m)
-        r2 = rng.random(m)
-        src_bit = r1 > ab
-        dst_bit = np.where(src_bit, r2 > c_norm, r2 > a_norm)
+        r1 = rng.random(m, dtype=np.float32)
+        r2 = rng.random(m, dtype=np.float32)
+        src_bit = r1 > np.float32(ab)
+        dst_bit = np.where(src_bit, r2 > np.float32(c_norm), r2 > np.float32(a_norm))
         src = (src << 1) | src_bit
         dst = (dst << 1) | dst_bit
     # scramble ids (fixed permutation derived from seed)
