@@ -82,6 +82,23 @@ class ShardState:
         else:
             self.order_interior = self.order
             self.order_boundary = self.order[:0]
+            interior = order_np
+            bnd = order_np[:0]
+        # MFMA split: each launch list is degree-descending, so the nodes
+        # taking the MFMA phase-B kernel (deg >= threshold) are a prefix;
+        # precompute the split points.  BIGCLAM_MFMA_DEG=0 disables (all
+        # nodes on the direct kernel).  Measured on the com-Amazon-shaped
+        # CSR (profiles/r01_kernel_opt_log.md): bf16 MFMA wins for EVERY
+        # degree (16.4 vs 22.6 ms even with tile padding at mean degree
+        # 5.5), so the bf16 default routes all nodes to it.
+        default_thr = "1" if self.storage_dtype == torch.bfloat16 else "0"
+        thr = int(os.environ.get("BIGCLAM_MFMA_DEG", default_thr))
+        if thr > 0:
+            self.n_mfma = int((deg[order_np] >= thr).sum())
+            self.n_mfma_interior = int((deg[interior] >= thr).sum())
+            self.n_mfma_boundary = int((deg[bnd] >= thr).sum())
+        else:
+            self.n_mfma = self.n_mfma_interior = self.n_mfma_boundary = 0
         self._halo_send: Optional[torch.Tensor] = None
         # per-edge source row (torch reference path); built lazily on CPU
         self._edge_src: Optional[torch.Tensor] = None
@@ -209,11 +226,13 @@ class ShardState:
                 ops.fused_grad_ls(
                     self.F, self.indptr, self.indices, self.sumF,
                     self.order_interior, self.cfg, out=out,
+                    n_mfma=self.n_mfma_interior,
                 )
                 halo_work.wait()
                 ops.fused_grad_ls(
                     self.F, self.indptr, self.indices, self.sumF,
                     self.order_boundary, self.cfg, out=out,
+                    n_mfma=self.n_mfma_boundary,
                 )
             else:
                 if halo_work is not None:
@@ -221,6 +240,7 @@ class ShardState:
                 ops.fused_grad_ls(
                     self.F, self.indptr, self.indices, self.sumF,
                     self.order, self.cfg, out=out,
+                    n_mfma=self.n_mfma,
                 )
             return grad, llh, best
         grad, llh = self.grad_llh_overlap(halo_work)

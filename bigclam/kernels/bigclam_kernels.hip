@@ -169,6 +169,32 @@ __device__ __forceinline__ float dot8(f32x8 a, f32x8 b, float acc) {
   return dot4(a.b, b.b, dot4(a.a, b.a, acc));
 }
 
+// ------------------------------------------------------------- MFMA types
+//
+// Fragment types for the CDNA4 matrix cores.  The line-search candidate
+// scoring is GEMM-shaped per node: (deg x K neighbor rows) @ (K x 16
+// clamped candidate rows)^T, mapped onto v_mfma_f32_16x16x32_bf16 /
+// v_mfma_f32_16x16x4_f32 tiles of 16 edges x 16 candidates.
+//
+// A and B fragments are filled with the SAME lane->k mapping
+// (k = kbase + (lane>>4)*8 + i for bf16, kbase + (lane>>4) for fp32), so
+// the contraction pairs A[r][k] with B[k][c] correctly under ANY internal
+// k-slot permutation; only the C/D register mapping is layout-sensitive:
+//   col = lane & 15, row = (lane >> 4) * 4 + reg_idx   (ISA §10, probed
+//   on-device by mfma_probe / tests/test_gpu.py).
+
+typedef short bf16x8 __attribute__((ext_vector_type(8)));  // 8 bf16, 4 VGPR
+typedef float f32x4v __attribute__((ext_vector_type(4)));  // MFMA acc
+
+__device__ __forceinline__ bf16x8 as_bf16x8(uint4 v) {
+  union {
+    uint4 u;
+    bf16x8 b;
+  } c;
+  c.u = v;
+  return c.b;
+}
+
 // ------------------------------------------------------------------- K1
 //
 // One 256-thread block per LOCAL node (launch order = degree-descending so
@@ -751,33 +777,18 @@ extern "C" __global__ void __launch_bounds__(BLOCK) k4_llh_only_bf16(
 // LDS: gacc/grad K floats + fu_s K floats (+1 KB statics) -> K <= 8192
 // fp32 via the NSLOT template (the K=5000 / K<=500 headline configs).
 
+// Phase A as a shared device function (used by kf_fused_t and kf_mfma_t):
+// on return gacc = the FINAL gradient row (also written to HBM), fu_s = a
+// copy of F_u, *s_llh_u = llh[u] (also written to HBM).  Caller must
+// __syncthreads() after phase A inside this function's tail (done here).
 template <int NSLOT>
-__global__ void __launch_bounds__(BLOCK, 3) kf_fused_t(
-    const float* __restrict__ F, const long long* __restrict__ indptr,
-    const int* __restrict__ indices, const float* __restrict__ sumF,
-    const int* __restrict__ order, float* __restrict__ grad,
-    double* __restrict__ llh, const float* __restrict__ ladder,
-    float* __restrict__ best, int n_local, int K, int n_ladder, float alpha,
-    float min_p, float max_p, float min_f, float max_f) {
-  const int u = order[blockIdx.x];
-  const long long e0 = indptr[u];
-  const long long e1 = indptr[u + 1];
+__device__ __forceinline__ void kf_phase_a(
+    const float* __restrict__ F, const int* __restrict__ indices,
+    const float* __restrict__ sumF, int u, long long e0, long long e1, int K,
+    float min_p, float max_p, float* __restrict__ gacc,
+    float* __restrict__ fu_s, float* red, double* s_llh_u,
+    float* __restrict__ grad, double* __restrict__ llh) {
   const int tid = threadIdx.x;
-  const int lane = tid & (WAVE - 1);
-  const int wid = tid >> 6;
-
-  __shared__ __attribute__((aligned(16))) float s_lad[MAX_LS];
-  __shared__ __attribute__((aligned(16))) double acc_llh[NWAVE][MAX_LS];
-  __shared__ __attribute__((aligned(16))) float acc_nt[NWAVE][MAX_LS];
-  __shared__ __attribute__((aligned(16))) float red[NWAVE];
-  __shared__ double s_llh_u;
-  extern __shared__ __attribute__((aligned(16))) char smem[];
-  float* gacc = reinterpret_cast<float*>(smem);  // K floats: grad after A
-  float* fu_s = gacc + K;                        // K floats: fu copy
-
-  if (tid < MAX_LS) s_lad[tid] = (tid < n_ladder) ? ladder[tid] : 0.f;
-
-  // ---------------- phase A: fused gradient + LLH (K1 structure)
   const float* __restrict__ fu_g = F + (size_t)u * K;
   float4 fu4[NSLOT];
 #pragma unroll
@@ -853,9 +864,106 @@ __global__ void __launch_bounds__(BLOCK, 3) kf_fused_t(
   if (tid == 0) {
     const double l = llh_acc + (double)(-fs) + (double)ff;
     llh[u] = l;
-    s_llh_u = l;
+    *s_llh_u = l;
   }
   __syncthreads();  // gacc = final grad, fu_s, s_llh_u visible
+}
+
+// Shared tail: per-candidate node terms, Armijo threshold and step pick.
+// acc_llh[wid][j] must already hold each wave's candidate-j edge-term
+// partial (the two phase-B variants fill it in different lane layouts but
+// the SAME [wave][candidate] indexing).
+__device__ __forceinline__ void kf_tail(
+    const float* __restrict__ fu, const float* __restrict__ gu,
+    const float* __restrict__ sumF, const float* s_lad,
+    double (*acc_llh)[MAX_LS], float (*acc_nt)[MAX_LS], float* red,
+    const double* s_llh_u, float* __restrict__ best, int u, int K,
+    int n_ladder, float alpha, float min_f, float max_f) {
+  const int tid = threadIdx.x;
+  const int lane = tid & (WAVE - 1);
+  const int wid = tid >> 6;
+  const int jmine = (((lane >> 5) & 1) << 3) | (((lane >> 4) & 1) << 2) |
+                    (((lane >> 3) & 1) << 1) | ((lane >> 2) & 1);
+  float s[MAX_LS];
+#pragma unroll
+  for (int j = 0; j < MAX_LS; ++j) s[j] = s_lad[j];
+  const v2f lo2 = {min_f, min_f}, hi2 = {max_f, max_f};
+
+  v2f accn2[MAX_LS];
+#pragma unroll
+  for (int j = 0; j < MAX_LS; ++j) accn2[j] = v2f{0.f, 0.f};
+  float p_gg = 0.f;
+  for (int k = tid * 4; k < K; k += BLOCK * 4) {
+    const float4 a = ld4(fu + k);
+    const float4 g = ld4(gu + k);
+    const float4 sf = ld4(sumF + k);
+    const v2f a0 = {a.x, a.y}, a1 = {a.z, a.w};
+    const v2f g0 = {g.x, g.y}, g1 = {g.z, g.w};
+    const v2f d0 = {a.x - sf.x, a.y - sf.y};
+    const v2f d1 = {a.z - sf.z, a.w - sf.w};
+    p_gg = dot4(g, g, p_gg);
+#pragma unroll
+    for (int j = 0; j < MAX_LS; ++j) {
+      const v2f sj = {s[j], s[j]};
+      v2f t = accn2[j];
+      t = __builtin_elementwise_fma(pk_clamp_fma(sj, g0, a0, lo2, hi2), d0,
+                                    t);
+      t = __builtin_elementwise_fma(pk_clamp_fma(sj, g1, a1, lo2, hi2), d1,
+                                    t);
+      accn2[j] = t;
+    }
+  }
+  float accn[MAX_LS];
+#pragma unroll
+  for (int j = 0; j < MAX_LS; ++j) accn[j] = accn2[j].x + accn2[j].y;
+  wave_reduce16(accn, lane);
+  if ((lane & 3) == 0) acc_nt[wid][jmine] = accn[0];
+  const float gg = block_allreduce_sum(p_gg, red);  // ends in __syncthreads
+
+  if (wid == 0) {
+    bool ok = false;
+    if (lane < MAX_LS) {
+      double trial = 0.0;
+#pragma unroll
+      for (int wv = 0; wv < NWAVE; ++wv)
+        trial += acc_llh[wv][lane] + (double)acc_nt[wv][lane];
+      ok = (lane < n_ladder) &&
+           (trial >= *s_llh_u + (double)(alpha * s_lad[lane] * gg));
+    }
+    const unsigned long long bal = __ballot(ok);
+    if (lane == 0)
+      best[u] = bal ? s_lad[__ffsll((unsigned long long)bal) - 1] : 0.f;
+  }
+}
+
+template <int NSLOT>
+__global__ void __launch_bounds__(BLOCK, 3) kf_fused_t(
+    const float* __restrict__ F, const long long* __restrict__ indptr,
+    const int* __restrict__ indices, const float* __restrict__ sumF,
+    const int* __restrict__ order, float* __restrict__ grad,
+    double* __restrict__ llh, const float* __restrict__ ladder,
+    float* __restrict__ best, int n_local, int K, int n_ladder, float alpha,
+    float min_p, float max_p, float min_f, float max_f) {
+  const int u = order[blockIdx.x];
+  const long long e0 = indptr[u];
+  const long long e1 = indptr[u + 1];
+  const int tid = threadIdx.x;
+  const int lane = tid & (WAVE - 1);
+  const int wid = tid >> 6;
+
+  __shared__ __attribute__((aligned(16))) float s_lad[MAX_LS];
+  __shared__ __attribute__((aligned(16))) double acc_llh[NWAVE][MAX_LS];
+  __shared__ __attribute__((aligned(16))) float acc_nt[NWAVE][MAX_LS];
+  __shared__ __attribute__((aligned(16))) float red[NWAVE];
+  __shared__ double s_llh_u;
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  float* gacc = reinterpret_cast<float*>(smem);  // K floats: grad after A
+  float* fu_s = gacc + K;                        // K floats: fu copy
+
+  if (tid < MAX_LS) s_lad[tid] = (tid < n_ladder) ? ladder[tid] : 0.f;
+
+  kf_phase_a<NSLOT>(F, indices, sumF, u, e0, e1, K, min_p, max_p, gacc, fu_s,
+                    red, &s_llh_u, grad, llh);
 
   // ---------------- phase B: 16-candidate line search (K2 structure)
   const float* __restrict__ fu = fu_s;
@@ -912,62 +1020,36 @@ __global__ void __launch_bounds__(BLOCK, 3) kf_fused_t(
       llh_mine += (double)log1pf(-p) + (double)x;
     }
   }
+  if ((lane & 3) == 0) acc_llh[wid][jmine] = llh_mine;
 
-  v2f accn2[MAX_LS];
-#pragma unroll
-  for (int j = 0; j < MAX_LS; ++j) accn2[j] = v2f{0.f, 0.f};
-  float p_gg = 0.f;
-  for (int k = tid * 4; k < K; k += BLOCK * 4) {
-    const float4 a = ld4(fu + k);
-    const float4 g = ld4(gu + k);
-    const float4 sf = ld4(sumF + k);
-    const v2f a0 = {a.x, a.y}, a1 = {a.z, a.w};
-    const v2f g0 = {g.x, g.y}, g1 = {g.z, g.w};
-    const v2f d0 = {a.x - sf.x, a.y - sf.y};
-    const v2f d1 = {a.z - sf.z, a.w - sf.w};
-    p_gg = dot4(g, g, p_gg);
-#pragma unroll
-    for (int j = 0; j < MAX_LS; ++j) {
-      const v2f sj = {s[j], s[j]};
-      v2f t = accn2[j];
-      t = __builtin_elementwise_fma(pk_clamp_fma(sj, g0, a0, lo2, hi2), d0,
-                                    t);
-      t = __builtin_elementwise_fma(pk_clamp_fma(sj, g1, a1, lo2, hi2), d1,
-                                    t);
-      accn2[j] = t;
-    }
-  }
-  float accn[MAX_LS];
-#pragma unroll
-  for (int j = 0; j < MAX_LS; ++j) accn[j] = accn2[j].x + accn2[j].y;
-  wave_reduce16(accn, lane);
-  if ((lane & 3) == 0) {
-    acc_nt[wid][jmine] = accn[0];
-    acc_llh[wid][jmine] = llh_mine;
-  }
-  const float gg = block_allreduce_sum(p_gg, red);  // ends in __syncthreads
-
-  if (wid == 0) {
-    bool ok = false;
-    if (lane < MAX_LS) {
-      double trial = 0.0;
-#pragma unroll
-      for (int wv = 0; wv < NWAVE; ++wv)
-        trial += acc_llh[wv][lane] + (double)acc_nt[wv][lane];
-      ok = (lane < n_ladder) &&
-           (trial >= s_llh_u + (double)(alpha * s_lad[lane] * gg));
-    }
-    const unsigned long long bal = __ballot(ok);
-    if (lane == 0)
-      best[u] = bal ? s_lad[__ffsll((unsigned long long)bal) - 1] : 0.f;
-  }
+  kf_tail(fu, gu, sumF, s_lad, acc_llh, acc_nt, red, &s_llh_u, best, u, K,
+          n_ladder, alpha, min_f, max_f);
 }
 
-// bf16 fused K1+K2 (same structure as kf_fused_t): gacc fp32 K*4B +
-// fu raw bf16 K*2B LDS -> covers K <= 16384 via the NSLOT template.
+// ------------------------------------------------------------ KF (MFMA)
+//
+// MFMA phase-B variant for HIGH-DEGREE nodes (the degree-descending launch
+// order makes them a prefix; the Python side splits at deg >= 16).  The
+// candidate scoring is run as a GEMM: per 16-edge tile, score[e][j] =
+// sum_k fv[e][k] * clamp(fu[k] + s_j*g[k]) via v_mfma_f32_16x16x4_f32
+// (exact fp32, ISA §12.10).  Candidate fragments are built once per
+// 64-element k-chunk per wave and amortized over up to TCAP edge tiles,
+// with accumulators resident in VGPRs across k-chunks.  vs the direct
+// kernel this removes BOTH per-(edge,candidate,k) VALU FMAs and the
+// per-edge 17-shuffle reductions — the measured issue-port bottleneck
+// (profiles/r01_kernel_opt_log.md: VALUBusy 80% at the occupancy cap) —
+// leaving fv streaming as the floor.
+//
+// Work split: the K dimension is quartered across the 4 waves (every wave
+// computes partial scores for EVERY edge tile on its own k-quarter, then
+// the quarters combine in a 4 KB LDS buffer per tile).  A tile-per-wave
+// split would idle 3 of the 4 waves (= 3 of 4 SIMDs/CU) for every node
+// with deg < 128 — most of the deg>=16 population on power-law graphs.
+#define KF_TG 4  // edge tiles per accumulator group (4 x f32x4v VGPRs each)
+
 template <int NSLOT>
-__global__ void __launch_bounds__(BLOCK, 3) kf_fused_bf16_t(
-    const u32* __restrict__ F, const long long* __restrict__ indptr,
+__global__ void __launch_bounds__(BLOCK, 3) kf_mfma_t(
+    const float* __restrict__ F, const long long* __restrict__ indptr,
     const int* __restrict__ indices, const float* __restrict__ sumF,
     const int* __restrict__ order, float* __restrict__ grad,
     double* __restrict__ llh, const float* __restrict__ ladder,
@@ -986,12 +1068,126 @@ __global__ void __launch_bounds__(BLOCK, 3) kf_fused_bf16_t(
   __shared__ __attribute__((aligned(16))) float red[NWAVE];
   __shared__ double s_llh_u;
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  float* gacc = reinterpret_cast<float*>(smem);               // K floats
-  u32* fu_s = reinterpret_cast<u32*>(smem + (size_t)K * 4);   // K/2 u32
+  float* gacc = reinterpret_cast<float*>(smem);  // K floats: grad after A
+  float* fu_s = gacc + K;                        // K floats: fu copy
 
   if (tid < MAX_LS) s_lad[tid] = (tid < n_ladder) ? ladder[tid] : 0.f;
 
-  // ---------------- phase A (k1_grad_llh_bf16_t structure)
+  kf_phase_a<NSLOT>(F, indices, sumF, u, e0, e1, K, min_p, max_p, gacc, fu_s,
+                    red, &s_llh_u, grad, llh);
+
+  // ---------------- phase B: MFMA 16-edge x 16-candidate tiles
+  __shared__ __attribute__((aligned(16))) float redt[NWAVE][WAVE][4];
+  const int deg = (int)(e1 - e0);
+  const int ntiles = (deg + 15) >> 4;
+  const int arow = lane & 15;   // A row (edge-in-tile) and C column select
+  const int kgrp = lane >> 4;   // k-group 0..3 (one float of 4 per step)
+  const float s_j = s_lad[arow];  // B column j == lane & 15
+  // K is partitioned into 16-element windows (one 4-step MFMA burst each,
+  // fed from ONE float4 A-load per lane); each wave owns a contiguous
+  // quarter of the windows.  K % 4 == 0, so a window's float4 at
+  // w*16 + kgrp*4 is either fully inside [0,K) or fully outside.
+  const int nwin = (K + 15) >> 4;
+  const int wq = (nwin + NWAVE - 1) / NWAVE;
+  const int ww0 = wid * wq;
+  const int ww1 = min(nwin, ww0 + wq);
+  constexpr int WC = 4;  // windows per candidate-fragment burst (64 k)
+  double llh_j = 0.0;
+
+  for (int tbase = 0; tbase < ntiles; tbase += KF_TG) {
+    const int nt = min(KF_TG, ntiles - tbase);
+    const float* aptr[KF_TG];
+#pragma unroll
+    for (int t = 0; t < KF_TG; ++t) {
+      const long long e = e0 + (long long)(tbase + t) * 16 + arow;
+      // padded rows get NO pointer: a clamped duplicate of the last edge
+      // row cost ~3x extra load traffic at mean degree 5.5
+      aptr[t] = (e < e1) ? F + (size_t)indices[e] * K : nullptr;
+    }
+    f32x4v acc[KF_TG];
+#pragma unroll
+    for (int t = 0; t < KF_TG; ++t) acc[t] = f32x4v{0.f, 0.f, 0.f, 0.f};
+
+    // Per window w, MFMA step i=0..3 uses the k-assignment
+    //   k(lane, i) = w*16 + (lane>>4)*4 + i
+    // — a bijection onto the window shared by A and B, so ONE float4
+    // A-load per lane feeds 4 MFMA steps (a scalar per-step gather was
+    // 4x the load instructions and lost to the direct kernel).
+    for (int wb = ww0; wb < ww1; wb += WC) {
+      float bfrag[WC * 4];
+#pragma unroll
+      for (int c = 0; c < WC * 4; ++c) {
+        const int k = (wb + (c >> 2)) * 16 + kgrp * 4 + (c & 3);
+        bfrag[c] = (wb + (c >> 2) < ww1 && k < K)
+                       ? __builtin_amdgcn_fmed3f(
+                             fmaf(s_j, gacc[k], fu_s[k]), min_f, max_f)
+                       : 0.f;
+      }
+#pragma unroll
+      for (int t = 0; t < KF_TG; ++t) {
+        if (t >= nt) break;
+#pragma unroll
+        for (int wi = 0; wi < WC; ++wi) {
+          const int k4 = (wb + wi) * 16 + kgrp * 4;
+          const float4 a4 = (wb + wi < ww1 && k4 + 3 < K && aptr[t])
+                                ? ld4(aptr[t] + k4)
+                                : float4{0.f, 0.f, 0.f, 0.f};
+          acc[t] = __builtin_amdgcn_mfma_f32_16x16x4f32(a4.x, bfrag[wi * 4],
+                                                        acc[t], 0, 0, 0);
+          acc[t] = __builtin_amdgcn_mfma_f32_16x16x4f32(
+              a4.y, bfrag[wi * 4 + 1], acc[t], 0, 0, 0);
+          acc[t] = __builtin_amdgcn_mfma_f32_16x16x4f32(
+              a4.z, bfrag[wi * 4 + 2], acc[t], 0, 0, 0);
+          acc[t] = __builtin_amdgcn_mfma_f32_16x16x4f32(
+              a4.w, bfrag[wi * 4 + 3], acc[t], 0, 0, 0);
+        }
+      }
+    }
+    // combine the four k-quarters per tile and run the LLH epilogue on a
+    // rotating wave (spreads the exp/log work across SIMDs)
+#pragma unroll
+    for (int t = 0; t < KF_TG; ++t) {
+      if (t >= nt) break;
+      __syncthreads();  // redt free / previous tile consumed
+      *reinterpret_cast<float4*>(&redt[wid][lane][0]) =
+          float4{acc[t][0], acc[t][1], acc[t][2], acc[t][3]};
+      __syncthreads();
+      if (wid == ((tbase + t) & (NWAVE - 1))) {
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          const int erow = (tbase + t) * 16 + kgrp * 4 + r;
+          if (erow < deg) {
+            const float x = redt[0][lane][r] + redt[1][lane][r] +
+                            redt[2][lane][r] + redt[3][lane][r];
+            const float p = clamp_p(__expf(-x), min_p, max_p);
+            llh_j += (double)log1pf(-p) + (double)x;
+          }
+        }
+      }
+    }
+  }
+  // fold the 4 row-groups: lanes l, l^16, l^32, l^48 share candidate l&15
+  llh_j += __shfl_xor(llh_j, 16, WAVE);
+  llh_j += __shfl_xor(llh_j, 32, WAVE);
+  if (lane < MAX_LS) acc_llh[wid][lane] = llh_j;
+  __syncthreads();  // all epilogue reads done before kf_tail reuses LDS
+
+  kf_tail(fu_s, gacc, sumF, s_lad, acc_llh, acc_nt, red, &s_llh_u, best, u,
+          K, n_ladder, alpha, min_f, max_f);
+}
+
+// bf16 fused K1+K2 (same structure as kf_fused_t): gacc fp32 K*4B +
+// fu raw bf16 K*2B LDS -> covers K <= 16384 via the NSLOT template.
+// bf16 phase A as a shared device function (kf_fused_bf16_t /
+// kf_mfma_bf16_t): gacc = final fp32 grad, fu_s = raw bf16 F_u copy.
+template <int NSLOT>
+__device__ __forceinline__ void kf_phase_a_bf16(
+    const u32* __restrict__ F, const int* __restrict__ indices,
+    const float* __restrict__ sumF, int u, long long e0, long long e1, int K,
+    float min_p, float max_p, float* __restrict__ gacc, u32* __restrict__ fu_s,
+    float* red, double* s_llh_u, float* __restrict__ grad,
+    double* __restrict__ llh) {
+  const int tid = threadIdx.x;
   const u32* __restrict__ fu_g = F + (size_t)u * (K / 2);
   f32x8 fu8[NSLOT];
 #pragma unroll
@@ -1098,9 +1294,107 @@ __global__ void __launch_bounds__(BLOCK, 3) kf_fused_bf16_t(
   if (tid == 0) {
     const double l = llh_acc + (double)(-fs) + (double)ff;
     llh[u] = l;
-    s_llh_u = l;
+    *s_llh_u = l;
   }
   __syncthreads();
+}
+
+// bf16 shared tail: node terms + Armijo pick (fu raw bf16 from LDS).
+__device__ __forceinline__ void kf_tail_bf16(
+    const u32* __restrict__ fu, const float* __restrict__ gu,
+    const float* __restrict__ sumF, const float* s_lad,
+    double (*acc_llh)[MAX_LS], float (*acc_nt)[MAX_LS], float* red,
+    const double* s_llh_u, float* __restrict__ best, int u, int K,
+    int n_ladder, float alpha, float min_f, float max_f) {
+  const int tid = threadIdx.x;
+  const int lane = tid & (WAVE - 1);
+  const int wid = tid >> 6;
+  const int jmine = (((lane >> 5) & 1) << 3) | (((lane >> 4) & 1) << 2) |
+                    (((lane >> 3) & 1) << 1) | ((lane >> 2) & 1);
+  float s[MAX_LS];
+#pragma unroll
+  for (int j = 0; j < MAX_LS; ++j) s[j] = s_lad[j];
+  const v2f lo2 = {min_f, min_f}, hi2 = {max_f, max_f};
+
+  v2f accn2[MAX_LS];
+#pragma unroll
+  for (int j = 0; j < MAX_LS; ++j) accn2[j] = v2f{0.f, 0.f};
+  float p_gg = 0.f;
+  for (int k = tid * 8; k < K; k += BLOCK * 8) {
+    const uint4 araw = *reinterpret_cast<const uint4*>(fu + k / 2);
+    const v2f a0 = bf2(araw.x), a1 = bf2(araw.y), a2 = bf2(araw.z),
+              a3 = bf2(araw.w);
+    const float4 gA = ld4(gu + k);
+    const float4 gB = ld4(gu + k + 4);
+    const v2f g0 = {gA.x, gA.y}, g1 = {gA.z, gA.w};
+    const v2f g2 = {gB.x, gB.y}, g3 = {gB.z, gB.w};
+    const float4 sA = ld4(sumF + k);
+    const float4 sB = ld4(sumF + k + 4);
+    const v2f d0 = a0 - v2f{sA.x, sA.y}, d1 = a1 - v2f{sA.z, sA.w};
+    const v2f d2 = a2 - v2f{sB.x, sB.y}, d3 = a3 - v2f{sB.z, sB.w};
+    p_gg = dot4(gA, gA, dot4(gB, gB, p_gg));
+#pragma unroll
+    for (int j = 0; j < MAX_LS; ++j) {
+      const v2f sj = {s[j], s[j]};
+      v2f t = accn2[j];
+      t = __builtin_elementwise_fma(pk_clamp_fma(sj, g0, a0, lo2, hi2), d0, t);
+      t = __builtin_elementwise_fma(pk_clamp_fma(sj, g1, a1, lo2, hi2), d1, t);
+      t = __builtin_elementwise_fma(pk_clamp_fma(sj, g2, a2, lo2, hi2), d2, t);
+      t = __builtin_elementwise_fma(pk_clamp_fma(sj, g3, a3, lo2, hi2), d3, t);
+      accn2[j] = t;
+    }
+  }
+  float accn[MAX_LS];
+#pragma unroll
+  for (int j = 0; j < MAX_LS; ++j) accn[j] = accn2[j].x + accn2[j].y;
+  wave_reduce16(accn, lane);
+  if ((lane & 3) == 0) acc_nt[wid][jmine] = accn[0];
+  const float gg = block_allreduce_sum(p_gg, red);  // ends in __syncthreads
+
+  if (wid == 0) {
+    bool ok = false;
+    if (lane < MAX_LS) {
+      double trial = 0.0;
+#pragma unroll
+      for (int wv = 0; wv < NWAVE; ++wv)
+        trial += acc_llh[wv][lane] + (double)acc_nt[wv][lane];
+      ok = (lane < n_ladder) &&
+           (trial >= *s_llh_u + (double)(alpha * s_lad[lane] * gg));
+    }
+    const unsigned long long bal = __ballot(ok);
+    if (lane == 0)
+      best[u] = bal ? s_lad[__ffsll((unsigned long long)bal) - 1] : 0.f;
+  }
+}
+
+template <int NSLOT>
+__global__ void __launch_bounds__(BLOCK, 3) kf_fused_bf16_t(
+    const u32* __restrict__ F, const long long* __restrict__ indptr,
+    const int* __restrict__ indices, const float* __restrict__ sumF,
+    const int* __restrict__ order, float* __restrict__ grad,
+    double* __restrict__ llh, const float* __restrict__ ladder,
+    float* __restrict__ best, int n_local, int K, int n_ladder, float alpha,
+    float min_p, float max_p, float min_f, float max_f) {
+  const int u = order[blockIdx.x];
+  const long long e0 = indptr[u];
+  const long long e1 = indptr[u + 1];
+  const int tid = threadIdx.x;
+  const int lane = tid & (WAVE - 1);
+  const int wid = tid >> 6;
+
+  __shared__ __attribute__((aligned(16))) float s_lad[MAX_LS];
+  __shared__ __attribute__((aligned(16))) double acc_llh[NWAVE][MAX_LS];
+  __shared__ __attribute__((aligned(16))) float acc_nt[NWAVE][MAX_LS];
+  __shared__ __attribute__((aligned(16))) float red[NWAVE];
+  __shared__ double s_llh_u;
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  float* gacc = reinterpret_cast<float*>(smem);               // K floats
+  u32* fu_s = reinterpret_cast<u32*>(smem + (size_t)K * 4);   // K/2 u32
+
+  if (tid < MAX_LS) s_lad[tid] = (tid < n_ladder) ? ladder[tid] : 0.f;
+
+  kf_phase_a_bf16<NSLOT>(F, indices, sumF, u, e0, e1, K, min_p, max_p, gacc,
+                         fu_s, red, &s_llh_u, grad, llh);
 
   // ---------------- phase B (k2_ls_v3_bf16 structure; fu/g from LDS)
   const u32* __restrict__ fu = fu_s;
@@ -1155,59 +1449,142 @@ __global__ void __launch_bounds__(BLOCK, 3) kf_fused_bf16_t(
       llh_mine += (double)log1pf(-p) + (double)x;
     }
   }
+  if ((lane & 3) == 0) acc_llh[wid][jmine] = llh_mine;
 
-  v2f accn2[MAX_LS];
+  kf_tail_bf16(fu, gu, sumF, s_lad, acc_llh, acc_nt, red, &s_llh_u, best, u,
+               K, n_ladder, alpha, min_f, max_f);
+}
+
+// bf16 MFMA phase-B variant (see kf_mfma_t): v_mfma_f32_16x16x32_bf16,
+// 8192 MACs/instruction — the A fragment is the raw bf16 fv row (16B
+// uint4 per lane), the B fragment is the clamped candidate row built in
+// fp32 from LDS fu/grad and packed to bf16 (RNE, same rounding as the K3
+// bf16 commit).  Candidate fragments are built once per 256-element
+// k-chunk per wave and reused across KF_TCAP edge tiles.
+template <int NSLOT>
+__global__ void __launch_bounds__(BLOCK, 3) kf_mfma_bf16_t(
+    const u32* __restrict__ F, const long long* __restrict__ indptr,
+    const int* __restrict__ indices, const float* __restrict__ sumF,
+    const int* __restrict__ order, float* __restrict__ grad,
+    double* __restrict__ llh, const float* __restrict__ ladder,
+    float* __restrict__ best, int n_local, int K, int n_ladder, float alpha,
+    float min_p, float max_p, float min_f, float max_f) {
+  const int u = order[blockIdx.x];
+  const long long e0 = indptr[u];
+  const long long e1 = indptr[u + 1];
+  const int tid = threadIdx.x;
+  const int lane = tid & (WAVE - 1);
+  const int wid = tid >> 6;
+
+  __shared__ __attribute__((aligned(16))) float s_lad[MAX_LS];
+  __shared__ __attribute__((aligned(16))) double acc_llh[NWAVE][MAX_LS];
+  __shared__ __attribute__((aligned(16))) float acc_nt[NWAVE][MAX_LS];
+  __shared__ __attribute__((aligned(16))) float red[NWAVE];
+  __shared__ double s_llh_u;
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  float* gacc = reinterpret_cast<float*>(smem);               // K floats
+  u32* fu_s = reinterpret_cast<u32*>(smem + (size_t)K * 4);   // K/2 u32
+
+  if (tid < MAX_LS) s_lad[tid] = (tid < n_ladder) ? ladder[tid] : 0.f;
+
+  kf_phase_a_bf16<NSLOT>(F, indices, sumF, u, e0, e1, K, min_p, max_p, gacc,
+                         fu_s, red, &s_llh_u, grad, llh);
+
+  // ---------------- phase B: MFMA 16-edge x 16-candidate tiles
+  // (k-quartered across waves; see kf_mfma_t)
+  __shared__ __attribute__((aligned(16))) float redt[NWAVE][WAVE][4];
+  const int deg = (int)(e1 - e0);
+  const int ntiles = (deg + 15) >> 4;
+  const int arow = lane & 15;   // A row (edge) / B col (candidate) select
+  const int kgrp = lane >> 4;   // k-group 0..3 (8 contiguous bf16 each)
+  const float s_j = s_lad[arow];
+  const int nkstep = (K + 31) >> 5;  // 32-k MFMA steps (K % 8 == 0)
+  const int sq = (nkstep + NWAVE - 1) / NWAVE;
+  const int sw0 = wid * sq;
+  const int sw1 = min(nkstep, sw0 + sq);
+  constexpr int KC = 8;  // candidate fragments built per burst (256 k)
+  double llh_j = 0.0;
+
+  for (int tbase = 0; tbase < ntiles; tbase += KF_TG) {
+    const int nt = min(KF_TG, ntiles - tbase);
+    const u32* aptr[KF_TG];
 #pragma unroll
-  for (int j = 0; j < MAX_LS; ++j) accn2[j] = v2f{0.f, 0.f};
-  float p_gg = 0.f;
-  for (int k = tid * 8; k < K; k += BLOCK * 8) {
-    const uint4 araw = *reinterpret_cast<const uint4*>(fu + k / 2);
-    const v2f a0 = bf2(araw.x), a1 = bf2(araw.y), a2 = bf2(araw.z),
-              a3 = bf2(araw.w);
-    const float4 gA = ld4(gu + k);
-    const float4 gB = ld4(gu + k + 4);
-    const v2f g0 = {gA.x, gA.y}, g1 = {gA.z, gA.w};
-    const v2f g2 = {gB.x, gB.y}, g3 = {gB.z, gB.w};
-    const float4 sA = ld4(sumF + k);
-    const float4 sB = ld4(sumF + k + 4);
-    const v2f d0 = a0 - v2f{sA.x, sA.y}, d1 = a1 - v2f{sA.z, sA.w};
-    const v2f d2 = a2 - v2f{sB.x, sB.y}, d3 = a3 - v2f{sB.z, sB.w};
-    p_gg = dot4(gA, gA, dot4(gB, gB, p_gg));
+    for (int t = 0; t < KF_TG; ++t) {
+      const long long e = e0 + (long long)(tbase + t) * 16 + arow;
+      aptr[t] = (e < e1) ? F + (size_t)indices[e] * (K / 2) : nullptr;
+    }
+    f32x4v acc[KF_TG];
 #pragma unroll
-    for (int j = 0; j < MAX_LS; ++j) {
-      const v2f sj = {s[j], s[j]};
-      v2f t = accn2[j];
-      t = __builtin_elementwise_fma(pk_clamp_fma(sj, g0, a0, lo2, hi2), d0, t);
-      t = __builtin_elementwise_fma(pk_clamp_fma(sj, g1, a1, lo2, hi2), d1, t);
-      t = __builtin_elementwise_fma(pk_clamp_fma(sj, g2, a2, lo2, hi2), d2, t);
-      t = __builtin_elementwise_fma(pk_clamp_fma(sj, g3, a3, lo2, hi2), d3, t);
-      accn2[j] = t;
+    for (int t = 0; t < KF_TG; ++t) acc[t] = f32x4v{0.f, 0.f, 0.f, 0.f};
+
+    for (int sb = sw0; sb < sw1; sb += KC) {
+      bf16x8 bfrag[KC];
+#pragma unroll
+      for (int c = 0; c < KC; ++c) {
+        const int k = (sb + c) * 32 + kgrp * 8;
+        if (sb + c < sw1 && k < K) {
+          const uint4 araw = *reinterpret_cast<const uint4*>(fu_s + k / 2);
+          const float4 gA = ld4(gacc + k);
+          const float4 gB = ld4(gacc + k + 4);
+          const v2f a0 = bf2(araw.x), a1 = bf2(araw.y), a2 = bf2(araw.z),
+                    a3 = bf2(araw.w);
+          const v2f lo2 = {min_f, min_f}, hi2 = {max_f, max_f};
+          const v2f sj = {s_j, s_j};
+          const v2f c0 = pk_clamp_fma(sj, v2f{gA.x, gA.y}, a0, lo2, hi2);
+          const v2f c1 = pk_clamp_fma(sj, v2f{gA.z, gA.w}, a1, lo2, hi2);
+          const v2f c2 = pk_clamp_fma(sj, v2f{gB.x, gB.y}, a2, lo2, hi2);
+          const v2f c3 = pk_clamp_fma(sj, v2f{gB.z, gB.w}, a3, lo2, hi2);
+          bfrag[c] = as_bf16x8(
+              uint4{pack_bf16_rne(c0.x, c0.y), pack_bf16_rne(c1.x, c1.y),
+                    pack_bf16_rne(c2.x, c2.y), pack_bf16_rne(c3.x, c3.y)});
+        } else {
+          bfrag[c] = as_bf16x8(uint4{0u, 0u, 0u, 0u});
+        }
+      }
+#pragma unroll
+      for (int t = 0; t < KF_TG; ++t) {
+        if (t >= nt) break;
+#pragma unroll
+        for (int c = 0; c < KC; ++c) {
+          const int k = (sb + c) * 32 + kgrp * 8;
+          const bf16x8 afrag =
+              (sb + c < sw1 && k < K && aptr[t])
+                  ? as_bf16x8(*reinterpret_cast<const uint4*>(aptr[t] +
+                                                              k / 2))
+                  : as_bf16x8(uint4{0u, 0u, 0u, 0u});
+          acc[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag, bfrag[c],
+                                                           acc[t], 0, 0, 0);
+        }
+      }
+    }
+#pragma unroll
+    for (int t = 0; t < KF_TG; ++t) {
+      if (t >= nt) break;
+      __syncthreads();  // redt free / previous tile consumed
+      *reinterpret_cast<float4*>(&redt[wid][lane][0]) =
+          float4{acc[t][0], acc[t][1], acc[t][2], acc[t][3]};
+      __syncthreads();
+      if (wid == ((tbase + t) & (NWAVE - 1))) {
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          const int erow = (tbase + t) * 16 + kgrp * 4 + r;
+          if (erow < deg) {
+            const float x = redt[0][lane][r] + redt[1][lane][r] +
+                            redt[2][lane][r] + redt[3][lane][r];
+            const float p = clamp_p(__expf(-x), min_p, max_p);
+            llh_j += (double)log1pf(-p) + (double)x;
+          }
+        }
+      }
     }
   }
-  float accn[MAX_LS];
-#pragma unroll
-  for (int j = 0; j < MAX_LS; ++j) accn[j] = accn2[j].x + accn2[j].y;
-  wave_reduce16(accn, lane);
-  if ((lane & 3) == 0) {
-    acc_nt[wid][jmine] = accn[0];
-    acc_llh[wid][jmine] = llh_mine;
-  }
-  const float gg = block_allreduce_sum(p_gg, red);  // ends in __syncthreads
+  llh_j += __shfl_xor(llh_j, 16, WAVE);
+  llh_j += __shfl_xor(llh_j, 32, WAVE);
+  if (lane < MAX_LS) acc_llh[wid][lane] = llh_j;
+  __syncthreads();  // all epilogue reads done before kf_tail runs
 
-  if (wid == 0) {
-    bool ok = false;
-    if (lane < MAX_LS) {
-      double trial = 0.0;
-#pragma unroll
-      for (int wv = 0; wv < NWAVE; ++wv)
-        trial += acc_llh[wv][lane] + (double)acc_nt[wv][lane];
-      ok = (lane < n_ladder) &&
-           (trial >= s_llh_u + (double)(alpha * s_lad[lane] * gg));
-    }
-    const unsigned long long bal = __ballot(ok);
-    if (lane == 0)
-      best[u] = bal ? s_lad[__ffsll((unsigned long long)bal) - 1] : 0.f;
-  }
+  kf_tail_bf16(fu_s, gacc, sumF, s_lad, acc_llh, acc_nt, red, &s_llh_u, best,
+               u, K, n_ladder, alpha, min_f, max_f);
 }
 
 // K2 tiled variant for K too large to stage whole rows (fp32 K > ~20k,
@@ -1911,5 +2288,124 @@ extern "C" void launch_kf_bf16(const void* F, const long long* indptr,
     KFB_CASE(8);
   }
 #undef KFB_CASE
+  HIP_CHECK(hipGetLastError());
+}
+
+extern "C" void launch_kf_mfma(const float* F, const long long* indptr,
+                               const int* indices, const float* sumF,
+                               const int* order, float* grad, double* llh,
+                               const float* ladder, float* best, int n_local,
+                               int K, int n_ladder, float alpha, float min_p,
+                               float max_p, float min_f, float max_f,
+                               hipStream_t stream) {
+  if (n_local == 0) return;
+  if (n_ladder > 16) throw std::runtime_error("ladder length > 16 unsupported");
+  if (K > 8192) throw std::runtime_error("fused kernel requires K <= 8192");
+  const size_t lds = (size_t)K * 8;  // gacc + fu_s
+#define KFM_CASE(NS)                                                         \
+  do {                                                                       \
+    allow_large_lds((const void*)&kf_mfma_t<NS>, lds);                       \
+    hipLaunchKernelGGL((kf_mfma_t<NS>), dim3(n_local), dim3(256), lds,       \
+                       stream, F, indptr, indices, sumF, order, grad, llh,   \
+                       ladder, best, n_local, K, n_ladder, alpha, min_p,     \
+                       max_p, min_f, max_f);                                 \
+  } while (0)
+  if (K <= 1024) {
+    KFM_CASE(1);
+  } else if (K <= 2048) {
+    KFM_CASE(2);
+  } else if (K <= 4096) {
+    KFM_CASE(4);
+  } else {
+    KFM_CASE(8);
+  }
+#undef KFM_CASE
+  HIP_CHECK(hipGetLastError());
+}
+
+extern "C" void launch_kf_mfma_bf16(const void* F, const long long* indptr,
+                                    const int* indices, const float* sumF,
+                                    const int* order, float* grad,
+                                    double* llh, const float* ladder,
+                                    float* best, int n_local, int K,
+                                    int n_ladder, float alpha, float min_p,
+                                    float max_p, float min_f, float max_f,
+                                    hipStream_t stream) {
+  if (n_local == 0) return;
+  if (n_ladder > 16) throw std::runtime_error("ladder length > 16 unsupported");
+  if (K > 16384)
+    throw std::runtime_error("bf16 fused kernel requires K <= 16384");
+  const size_t lds = (size_t)K * 6;  // gacc fp32 + fu raw bf16
+  const u32* Fb = reinterpret_cast<const u32*>(F);
+#define KFMB_CASE(NS)                                                        \
+  do {                                                                       \
+    allow_large_lds((const void*)&kf_mfma_bf16_t<NS>, lds);                  \
+    hipLaunchKernelGGL((kf_mfma_bf16_t<NS>), dim3(n_local), dim3(256), lds,  \
+                       stream, Fb, indptr, indices, sumF, order, grad, llh,  \
+                       ladder, best, n_local, K, n_ladder, alpha, min_p,     \
+                       max_p, min_f, max_f);                                 \
+  } while (0)
+  if (K <= 2048) {
+    KFMB_CASE(1);
+  } else if (K <= 4096) {
+    KFMB_CASE(2);
+  } else if (K <= 8192) {
+    KFMB_CASE(4);
+  } else {
+    KFMB_CASE(8);
+  }
+#undef KFMB_CASE
+  HIP_CHECK(hipGetLastError());
+}
+
+// ------------------------------------------------- MFMA layout probes
+//
+// On-device verification of the assumed C/D register mapping
+// (col = lane&15, row = (lane>>4)*4 + reg): computes ONE 16x16 tile
+// D = A @ B with B supplied column-major, written out per the assumed
+// mapping; tests/test_gpu.py compares against a torch matmul with
+// ASYMMETRIC inputs (a swapped mapping produces D^T and fails loudly).
+
+extern "C" __global__ void mfma_probe_bf16_k(const u32* __restrict__ A,
+                                             const u32* __restrict__ Bc,
+                                             float* __restrict__ D) {
+  // A: 16 rows x 32 k bf16 row-major (16 u32/row); Bc[c][k] = B[k][c].
+  const int lane = threadIdx.x;
+  const int r = lane & 15, kg = lane >> 4;
+  const bf16x8 a =
+      as_bf16x8(*reinterpret_cast<const uint4*>(A + r * 16 + kg * 4));
+  const bf16x8 b =
+      as_bf16x8(*reinterpret_cast<const uint4*>(Bc + r * 16 + kg * 4));
+  f32x4v c = {0.f, 0.f, 0.f, 0.f};
+  c = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, c, 0, 0, 0);
+#pragma unroll
+  for (int i = 0; i < 4; ++i) D[(kg * 4 + i) * 16 + r] = c[i];
+}
+
+extern "C" __global__ void mfma_probe_f32_k(const float* __restrict__ A,
+                                            const float* __restrict__ Bc,
+                                            float* __restrict__ D) {
+  // A: 16 rows x 4 k fp32 row-major; Bc[c][k] = B[k][c].
+  const int lane = threadIdx.x;
+  const int r = lane & 15, kg = lane >> 4;
+  f32x4v c = {0.f, 0.f, 0.f, 0.f};
+  c = __builtin_amdgcn_mfma_f32_16x16x4f32(A[r * 4 + kg], Bc[r * 4 + kg], c,
+                                           0, 0, 0);
+#pragma unroll
+  for (int i = 0; i < 4; ++i) D[(kg * 4 + i) * 16 + r] = c[i];
+}
+
+extern "C" void launch_mfma_probe_bf16(const void* A, const void* Bc,
+                                       float* D, hipStream_t stream) {
+  hipLaunchKernelGGL(mfma_probe_bf16_k, dim3(1), dim3(64), 0, stream,
+                     reinterpret_cast<const u32*>(A),
+                     reinterpret_cast<const u32*>(Bc), D);
+  HIP_CHECK(hipGetLastError());
+}
+
+extern "C" void launch_mfma_probe_f32(const float* A, const float* Bc,
+                                      float* D, hipStream_t stream) {
+  hipLaunchKernelGGL(mfma_probe_f32_k, dim3(1), dim3(64), 0, stream, A, Bc,
+                     D);
   HIP_CHECK(hipGetLastError());
 }

@@ -39,6 +39,19 @@ extern "C" void launch_kf_bf16(const void*, const long long*, const int*,
                                const float*, const int*, float*, double*,
                                const float*, float*, int, int, int, float,
                                float, float, float, float, hipStream_t);
+extern "C" void launch_kf_mfma(const float*, const long long*, const int*,
+                               const float*, const int*, float*, double*,
+                               const float*, float*, int, int, int, float,
+                               float, float, float, float, hipStream_t);
+extern "C" void launch_kf_mfma_bf16(const void*, const long long*, const int*,
+                                    const float*, const int*, float*, double*,
+                                    const float*, float*, int, int, int,
+                                    float, float, float, float, float,
+                                    hipStream_t);
+extern "C" void launch_mfma_probe_bf16(const void*, const void*, float*,
+                                       hipStream_t);
+extern "C" void launch_mfma_probe_f32(const float*, const float*, float*,
+                                      hipStream_t);
 
 namespace {
 
@@ -180,11 +193,15 @@ void apply_step(torch::Tensor F_local, torch::Tensor grad,
   }
 }
 
+// n_mfma: the first n_mfma entries of `order` (degree-descending, so the
+// high-degree prefix) run the MFMA phase-B kernel; the rest the direct
+// one.  The two launches cover disjoint nodes.
 void fused_grad_ls(torch::Tensor F, torch::Tensor indptr,
                    torch::Tensor indices, torch::Tensor sumF,
                    torch::Tensor order, torch::Tensor grad, torch::Tensor llh,
                    torch::Tensor ladder, torch::Tensor best, double alpha,
-                   double min_p, double max_p, double min_f, double max_f) {
+                   double min_p, double max_p, double min_f, double max_f,
+                   int64_t n_mfma) {
   CHECK_F(F);
   CHECK_IN(indptr, torch::kInt64);
   CHECK_IN(indices, torch::kInt32);
@@ -200,24 +217,56 @@ void fused_grad_ls(torch::Tensor F, torch::Tensor indptr,
   TORCH_CHECK(grad.size(0) == n_local && grad.size(1) == K);
   const auto ip =
       reinterpret_cast<const long long*>(indptr.data_ptr<int64_t>());
+  TORCH_CHECK(n_mfma >= 0 && n_mfma <= n_blocks, "bad n_mfma");
+  const int n_hi = (int)n_mfma;
+  const int n_lo = n_blocks - n_hi;
+  const int* ord = order.data_ptr<int>();
   if (is_bf16(F)) {
     TORCH_CHECK(K % 8 == 0 && K <= 16384,
                 "bf16 fused kernel: K padded to 8, K <= 16384");
+    launch_kf_mfma_bf16(F.data_ptr(), ip, indices.data_ptr<int>(),
+                        sumF.data_ptr<float>(), ord, grad.data_ptr<float>(),
+                        llh.data_ptr<double>(), ladder.data_ptr<float>(),
+                        best.data_ptr<float>(), n_hi, K, (int)ladder.size(0),
+                        (float)alpha, (float)min_p, (float)max_p,
+                        (float)min_f, (float)max_f, current_stream());
     launch_kf_bf16(F.data_ptr(), ip, indices.data_ptr<int>(),
-                   sumF.data_ptr<float>(), order.data_ptr<int>(),
+                   sumF.data_ptr<float>(), ord + n_hi,
                    grad.data_ptr<float>(), llh.data_ptr<double>(),
-                   ladder.data_ptr<float>(), best.data_ptr<float>(), n_blocks,
+                   ladder.data_ptr<float>(), best.data_ptr<float>(), n_lo,
                    K, (int)ladder.size(0), (float)alpha, (float)min_p,
                    (float)max_p, (float)min_f, (float)max_f,
                    current_stream());
   } else {
     TORCH_CHECK(K % 4 == 0 && K <= 8192, "fused kernel: K padded, K <= 8192");
+    launch_kf_mfma(F.data_ptr<float>(), ip, indices.data_ptr<int>(),
+                   sumF.data_ptr<float>(), ord, grad.data_ptr<float>(),
+                   llh.data_ptr<double>(), ladder.data_ptr<float>(),
+                   best.data_ptr<float>(), n_hi, K, (int)ladder.size(0),
+                   (float)alpha, (float)min_p, (float)max_p, (float)min_f,
+                   (float)max_f, current_stream());
     launch_kf(F.data_ptr<float>(), ip, indices.data_ptr<int>(),
-              sumF.data_ptr<float>(), order.data_ptr<int>(),
-              grad.data_ptr<float>(), llh.data_ptr<double>(),
-              ladder.data_ptr<float>(), best.data_ptr<float>(), n_blocks, K,
-              (int)ladder.size(0), (float)alpha, (float)min_p, (float)max_p,
-              (float)min_f, (float)max_f, current_stream());
+              sumF.data_ptr<float>(), ord + n_hi, grad.data_ptr<float>(),
+              llh.data_ptr<double>(), ladder.data_ptr<float>(),
+              best.data_ptr<float>(), n_lo, K, (int)ladder.size(0),
+              (float)alpha, (float)min_p, (float)max_p, (float)min_f,
+              (float)max_f, current_stream());
+  }
+}
+
+// On-device MFMA C/D-layout probe: D = A @ B with Bc = B column-major.
+void mfma_probe(torch::Tensor A, torch::Tensor Bc, torch::Tensor D) {
+  TORCH_CHECK(D.scalar_type() == torch::kFloat32 && D.is_contiguous());
+  TORCH_CHECK(A.is_contiguous() && Bc.is_contiguous());
+  TORCH_CHECK(A.scalar_type() == Bc.scalar_type());
+  if (is_bf16(A)) {
+    TORCH_CHECK(A.size(0) == 16 && A.size(1) == 32);
+    launch_mfma_probe_bf16(A.data_ptr(), Bc.data_ptr(),
+                           D.data_ptr<float>(), current_stream());
+  } else {
+    TORCH_CHECK(A.size(0) == 16 && A.size(1) == 4);
+    launch_mfma_probe_f32(A.data_ptr<float>(), Bc.data_ptr<float>(),
+                          D.data_ptr<float>(), current_stream());
   }
 }
 
@@ -246,5 +295,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("conductance", &conductance,
         "K5: ego-net conductance per node (CDNA4)");
   m.def("fused_grad_ls", &fused_grad_ls,
-        "KF: fused K1 gradient+LLH and K2 line search, one pass per node");
+        "KF: fused K1 gradient+LLH and K2 line search, one pass per node "
+        "(MFMA tiles for the high-degree prefix)");
+  m.def("mfma_probe", &mfma_probe,
+        "MFMA C/D layout probe: one 16x16 D = A @ Bc^T tile");
 }

@@ -7,6 +7,7 @@
 #include <algorithm>
 #include <cstdio>
 #include <cstdlib>
+#include <cstring>
 #include <vector>
 
 extern "C" void launch_k1(const float*, const long long*, const int*,
@@ -21,6 +22,23 @@ extern "C" void launch_k3(float*, const float*, const float*, int, int, float,
 extern "C" void launch_k4(const float*, const long long*, const int*,
                           const float*, const int*, double*, int, int, float,
                           float, hipStream_t);
+extern "C" void launch_kf(const float*, const long long*, const int*,
+                          const float*, const int*, float*, double*,
+                          const float*, float*, int, int, int, float, float,
+                          float, float, float, hipStream_t);
+extern "C" void launch_kf_mfma(const float*, const long long*, const int*,
+                               const float*, const int*, float*, double*,
+                               const float*, float*, int, int, int, float,
+                               float, float, float, float, hipStream_t);
+extern "C" void launch_kf_bf16(const void*, const long long*, const int*,
+                               const float*, const int*, float*, double*,
+                               const float*, float*, int, int, int, float,
+                               float, float, float, float, hipStream_t);
+extern "C" void launch_kf_mfma_bf16(const void*, const long long*, const int*,
+                                    const float*, const int*, float*, double*,
+                                    const float*, float*, int, int, int,
+                                    float, float, float, float, float,
+                                    hipStream_t);
 
 #define HIP_CHECK(x)                                                \
   do {                                                              \
@@ -117,5 +135,74 @@ int main(int argc, char** argv) {
   time3("k3", [&] {
     launch_k3(dF, dG, dBest, N, K, 0.f, 1000.f, 0);
   });
+
+  // fused K1+K2 with the MFMA phase-B prefix at several degree splits
+  auto split_at = [&](long long thr) {
+    int n = 0;
+    while (n < N && indptr[order[n] + 1] - indptr[order[n]] >= thr) ++n;
+    return n;
+  };
+  auto kf_split = [&](int n_hi) {
+    launch_kf_mfma(dF, dIp, dIdx, dSum, dOrd, dG, dLlh, dLad, dBest, n_hi, K,
+                   16, 0.05f, 1e-4f, 0.9999f, 0.f, 1000.f, 0);
+    launch_kf(dF, dIp, dIdx, dSum, dOrd + n_hi, dG, dLlh, dLad, dBest,
+              N - n_hi, K, 16, 0.05f, 1e-4f, 0.9999f, 0.f, 1000.f, 0);
+  };
+  const int thrs[] = {0, 8, 16, 32, 64, 1 << 30};
+  for (int thr : thrs) {
+    const int n_hi = thr == 0 ? N : (thr == (1 << 30) ? 0 : split_at(thr));
+    char name[32];
+    snprintf(name, sizeof name, "kf@%-4d", thr == (1 << 30) ? -1 : thr);
+    printf("  (n_mfma=%d)\n", n_hi);
+    time3(name, [&] { kf_split(n_hi); });
+  }
+
+  // bf16 storage path (K padded to 8 assumed by the callers; K=5000 ok
+  // since kernels only need K%8==0 for uint4 rows — pad here)
+  if (K % 8 == 0 || true) {
+    const int Kb = (K + 7) & ~7;
+    unsigned int* dFb;
+    HIP_CHECK(hipMalloc(&dFb, (size_t)N * Kb * 2));
+    // pack dF (fp32) into bf16 pairs on host once (slow but simple)
+    {
+      std::vector<float> row(Kb, 0.f);
+      std::vector<unsigned int> rowb(Kb / 2);
+      std::vector<unsigned int> all((size_t)N * Kb / 2);
+      std::vector<float> hFall((size_t)N * K);
+      HIP_CHECK(hipMemcpy(hFall.data(), dF, (size_t)N * K * 4,
+                          hipMemcpyDeviceToHost));
+      for (int u = 0; u < N; ++u) {
+        for (int k = 0; k < K; ++k) row[k] = hFall[(size_t)u * K + k];
+        for (int k = K; k < Kb; ++k) row[k] = 0.f;
+        for (int k = 0; k < Kb; k += 2) {
+          unsigned int l, h;
+          memcpy(&l, &row[k], 4);
+          memcpy(&h, &row[k + 1], 4);
+          l += 0x7fffu + ((l >> 16) & 1u);
+          h += 0x7fffu + ((h >> 16) & 1u);
+          rowb[k / 2] = (l >> 16) | (h & 0xffff0000u);
+        }
+        memcpy(&all[(size_t)u * Kb / 2], rowb.data(), Kb / 2 * 4);
+      }
+      HIP_CHECK(hipMemcpy(dFb, all.data(), (size_t)N * Kb / 2 * 4,
+                          hipMemcpyHostToDevice));
+    }
+    auto kfb_split = [&](int n_hi) {
+      launch_kf_mfma_bf16(dFb, dIp, dIdx, dSum, dOrd, dG, dLlh, dLad, dBest,
+                          n_hi, Kb, 16, 0.05f, 1e-4f, 0.9999f, 0.f, 1000.f,
+                          0);
+      launch_kf_bf16(dFb, dIp, dIdx, dSum, dOrd + n_hi, dG, dLlh, dLad,
+                     dBest, N - n_hi, Kb, 16, 0.05f, 1e-4f, 0.9999f, 0.f,
+                     1000.f, 0);
+    };
+    for (int thr : thrs) {
+      const int n_hi = thr == 0 ? N : (thr == (1 << 30) ? 0 : split_at(thr));
+      char name[32];
+      snprintf(name, sizeof name, "kfb@%-4d", thr == (1 << 30) ? -1 : thr);
+      printf("  (n_mfma=%d)\n", n_hi);
+      time3(name, [&] { kfb_split(n_hi); });
+    }
+    HIP_CHECK(hipFree(dFb));
+  }
   return 0;
 }

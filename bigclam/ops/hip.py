@@ -79,11 +79,15 @@ def fused_grad_ls(
     order: torch.Tensor,
     cfg: BigClamConfig,
     out=None,
+    n_mfma: int = 0,
 ):
     """KF: fused K1 gradient+LLH and K2 line search in one per-node pass.
 
     Returns (grad [n,K], llh [n] f64, best_step [n]).  ``order`` may be a
-    node subset (halo overlap) writing into shared ``out`` buffers."""
+    node subset (halo overlap) writing into shared ``out`` buffers.
+    ``n_mfma``: the first n_mfma entries of ``order`` (the high-degree
+    prefix of the degree-descending launch order) run the MFMA phase-B
+    kernel; the rest the direct one."""
     ext = ensure_loaded()
     n_local = len(indptr) - 1
     if out is None:
@@ -98,6 +102,7 @@ def fused_grad_ls(
         F, indptr, indices, sumF, order, grad, llh,
         _ladder(cfg, F.device), best,
         cfg.alpha, cfg.min_p, cfg.max_p, cfg.min_f, cfg.max_f,
+        int(n_mfma),
     )
     return grad, llh, best
 

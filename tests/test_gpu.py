@@ -245,3 +245,95 @@ def test_fused_bf16_matches_separate_kernels():
     torch.testing.assert_close(grad_f, grad_s, rtol=0, atol=0)
     torch.testing.assert_close(llh_f, llh_s, rtol=0, atol=0)
     torch.testing.assert_close(best_f, best_s, rtol=0, atol=0)
+
+
+def test_mfma_probe_layout():
+    """On-device check of the assumed MFMA C/D register mapping
+    (col = lane&15, row = (lane>>4)*4 + reg) with ASYMMETRIC inputs:
+    a swapped mapping produces D^T and fails."""
+    from bigclam.ops.hip import ensure_loaded
+
+    ext = ensure_loaded()
+    torch.manual_seed(0)
+    # bf16 16x16x32
+    A = (torch.rand(16, 32, device="cuda") * 0.5).bfloat16()
+    Bc = (torch.rand(16, 32, device="cuda") * 0.5).bfloat16()
+    Bc[0, :] += 3.0  # make D clearly asymmetric
+    D = torch.empty(16, 16, device="cuda", dtype=torch.float32)
+    ext.mfma_probe(A, Bc, D)
+    Dref = A.float() @ Bc.float().T
+    torch.testing.assert_close(D, Dref, rtol=1e-5, atol=1e-5)
+    # fp32 16x16x4
+    A32 = torch.rand(16, 4, device="cuda") * 0.5
+    Bc32 = torch.rand(16, 4, device="cuda") * 0.5
+    Bc32[0, :] += 3.0
+    D32 = torch.empty(16, 16, device="cuda", dtype=torch.float32)
+    ext.mfma_probe(A32, Bc32, D32)
+    torch.testing.assert_close(D32, A32 @ Bc32.T, rtol=1e-6, atol=1e-6)
+
+
+def test_fused_mfma_matches_direct_fp32():
+    """kf_mfma_t (MFMA phase B) vs kf_fused_t (direct phase B) with EVERY
+    node forced down the MFMA path (exercises partial edge tiles: the
+    R-MAT graph has degrees from 1 to hundreds).  Phase A is shared code,
+    so grad/llh must be bitwise; the chosen step can flip on borderline
+    Armijo accepts (different fp32 summation order of the trial dots)."""
+    from bigclam.ops import hip as hip_ops
+
+    g = rmat_graph(10, 7.0, seed=53)
+    cfg, st = _mkstate(g, 192, seed=17)
+    assert st.fused_ok
+    args = (st.F, st.indptr, st.indices, st.sumF, st.order, cfg)
+    grad_d, llh_d, best_d = hip_ops.fused_grad_ls(*args, n_mfma=0)
+    grad_m, llh_m, best_m = hip_ops.fused_grad_ls(
+        *args, n_mfma=int(st.order.numel())
+    )
+    torch.testing.assert_close(grad_m, grad_d, rtol=0, atol=0)
+    torch.testing.assert_close(llh_m, llh_d, rtol=0, atol=0)
+    agree = (best_m == best_d).float().mean().item()
+    assert agree > 0.98, f"only {agree:.3f} of best-steps agree"
+    rbest = ref_ops.linesearch(
+        st.F, st.indptr, st.indices, st.sumF, grad_m, llh_m, cfg,
+        n_local=st.n_local,
+    )
+    agree_r = (best_m == rbest).float().mean().item()
+    assert agree_r > 0.98, f"only {agree_r:.3f} agree with torch reference"
+
+
+def test_fused_mfma_matches_direct_bf16():
+    """bf16 MFMA phase B vs direct bf16 phase B.  The MFMA path rounds the
+    clamped candidate rows to bf16 (the direct path keeps them fp32), so
+    the step pick tolerance is a bit looser; grad/llh (shared phase A)
+    stay bitwise."""
+    from bigclam.ops import hip as hip_ops
+
+    g = rmat_graph(10, 7.0, seed=54)
+    cfg, st = _mkstate_dtype(g, 192, "bf16", seed=18)
+    assert st.fused_ok
+    args = (st.F, st.indptr, st.indices, st.sumF, st.order, cfg)
+    grad_d, llh_d, best_d = hip_ops.fused_grad_ls(*args, n_mfma=0)
+    grad_m, llh_m, best_m = hip_ops.fused_grad_ls(
+        *args, n_mfma=int(st.order.numel())
+    )
+    torch.testing.assert_close(grad_m, grad_d, rtol=0, atol=0)
+    torch.testing.assert_close(llh_m, llh_d, rtol=0, atol=0)
+    agree = (best_m == best_d).float().mean().item()
+    assert agree > 0.95, f"only {agree:.3f} of best-steps agree"
+
+
+def test_fused_mfma_split_dispatch_fp32(monkeypatch):
+    """Split dispatch (deg>=16 prefix on MFMA, rest direct) == all-direct
+    on grad/llh and nearly everywhere on the step pick."""
+    from bigclam.ops import hip as hip_ops
+
+    monkeypatch.setenv("BIGCLAM_MFMA_DEG", "16")
+    g = rmat_graph(10, 7.0, seed=55)
+    cfg, st = _mkstate(g, 192, seed=19)
+    assert st.n_mfma > 0 and st.n_mfma < st.order.numel()
+    args = (st.F, st.indptr, st.indices, st.sumF, st.order, cfg)
+    grad_d, llh_d, best_d = hip_ops.fused_grad_ls(*args, n_mfma=0)
+    grad_s, llh_s, best_s = hip_ops.fused_grad_ls(*args, n_mfma=st.n_mfma)
+    torch.testing.assert_close(grad_s, grad_d, rtol=0, atol=0)
+    torch.testing.assert_close(llh_s, llh_d, rtol=0, atol=0)
+    agree = (best_s == best_d).float().mean().item()
+    assert agree > 0.98, f"only {agree:.3f} of best-steps agree"
