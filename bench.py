@@ -40,7 +40,7 @@ def main():
     ap.add_argument("--k", type=int, default=5000)
     ap.add_argument("--nodes", type=int, default=334863)  # com-Amazon N
     ap.add_argument("--edges", type=int, default=925872)  # com-Amazon E
-    ap.add_argument("--dtype", choices=["fp32", "bf16"], default="fp32")
+    ap.add_argument("--dtype", choices=["fp32", "bf16"], default="bf16")
     ap.add_argument("--graph-seed", type=int, default=42)
     ap.add_argument("--locality", type=float, default=0.7)
     ap.add_argument("--ls-steps", type=int, default=15,

@@ -1946,6 +1946,59 @@ extern "C" __global__ void __launch_bounds__(BLOCK) k3_apply_step_bf16(
   }
 }
 
+// K3 + column-sum fused (bf16): projected commit AND the per-column fp32
+// partial sums in ONE pass over F — the separate flow re-read F (plus a
+// torch fp32 materialization) just to refresh sumF, ~3.5 ms/sweep at the
+// com-Amazon K=5000 config.  Grid: (row stripes) x (2048-element
+// k-chunks); each thread owns 8 columns of the chunk and accumulates the
+// POST-ROUNDING bf16 values; per-stripe partials go to
+// partials[stripe][K] and a deterministic stage-2 sum (torch, [S,K] ->
+// [K]) produces sumF.  Rows with step 0 skip the grad read and the F
+// write but still contribute to the column sums.
+#define K3CS_ROWS 512   // rows per stripe
+#define K3CS_KCH 2048   // k per chunk (256 threads x 8)
+
+extern "C" __global__ void __launch_bounds__(BLOCK) k3_colsum_bf16(
+    u32* __restrict__ F, const float* __restrict__ grad,
+    const float* __restrict__ steps, float* __restrict__ partials,
+    int n_local, int K, float min_f, float max_f) {
+  const int r0 = blockIdx.x * K3CS_ROWS;
+  const int r1 = min(n_local, r0 + K3CS_ROWS);
+  const int k = blockIdx.y * K3CS_KCH + threadIdx.x * 8;
+  if (k >= K) return;
+  float c0 = 0.f, c1 = 0.f, c2 = 0.f, c3 = 0.f;
+  float c4 = 0.f, c5 = 0.f, c6 = 0.f, c7 = 0.f;
+  for (int u = r0; u < r1; ++u) {
+    u32* __restrict__ fu = F + (size_t)u * (K / 2) + k / 2;
+    const float s = steps[u];
+    uint4 o = *reinterpret_cast<uint4*>(fu);
+    if (s > 0.f) {
+      const float4 g0 = ld4(grad + (size_t)u * K + k);
+      const float4 g1 = ld4(grad + (size_t)u * K + k + 4);
+      const v2f a0 = bf2(o.x), a1 = bf2(o.y), a2 = bf2(o.z), a3 = bf2(o.w);
+      o.x = pack_bf16_rne(
+          fminf(fmaxf(fmaf(s, g0.x, a0.x), min_f), max_f),
+          fminf(fmaxf(fmaf(s, g0.y, a0.y), min_f), max_f));
+      o.y = pack_bf16_rne(
+          fminf(fmaxf(fmaf(s, g0.z, a1.x), min_f), max_f),
+          fminf(fmaxf(fmaf(s, g0.w, a1.y), min_f), max_f));
+      o.z = pack_bf16_rne(
+          fminf(fmaxf(fmaf(s, g1.x, a2.x), min_f), max_f),
+          fminf(fmaxf(fmaf(s, g1.y, a2.y), min_f), max_f));
+      o.w = pack_bf16_rne(
+          fminf(fmaxf(fmaf(s, g1.z, a3.x), min_f), max_f),
+          fminf(fmaxf(fmaf(s, g1.w, a3.y), min_f), max_f));
+      *reinterpret_cast<uint4*>(fu) = o;
+    }
+    const v2f b0 = bf2(o.x), b1 = bf2(o.y), b2 = bf2(o.z), b3 = bf2(o.w);
+    c0 += b0.x; c1 += b0.y; c2 += b1.x; c3 += b1.y;
+    c4 += b2.x; c5 += b2.y; c6 += b3.x; c7 += b3.y;
+  }
+  float* __restrict__ out = partials + (size_t)blockIdx.x * K + k;
+  *reinterpret_cast<float4*>(out) = float4{c0, c1, c2, c3};
+  *reinterpret_cast<float4*>(out + 4) = float4{c4, c5, c6, c7};
+}
+
 // ------------------------------------------------------------------- K5
 //
 // Ego-net conductance per node (replaces codes/bigclamv3-7.scala:39-54;
@@ -2288,6 +2341,19 @@ extern "C" void launch_kf_bf16(const void* F, const long long* indptr,
     KFB_CASE(8);
   }
 #undef KFB_CASE
+  HIP_CHECK(hipGetLastError());
+}
+
+extern "C" void launch_k3_colsum_bf16(void* F, const float* grad,
+                                      const float* steps, float* partials,
+                                      int n_local, int K, float min_f,
+                                      float max_f, hipStream_t stream) {
+  if (n_local == 0) return;
+  const int ns = (n_local + K3CS_ROWS - 1) / K3CS_ROWS;
+  const int nk = (K + K3CS_KCH - 1) / K3CS_KCH;
+  hipLaunchKernelGGL(k3_colsum_bf16, dim3(ns, nk), dim3(256), 0, stream,
+                     reinterpret_cast<u32*>(F), grad, steps, partials,
+                     n_local, K, min_f, max_f);
   HIP_CHECK(hipGetLastError());
 }
 

@@ -39,6 +39,9 @@ extern "C" void launch_kf_bf16(const void*, const long long*, const int*,
                                const float*, const int*, float*, double*,
                                const float*, float*, int, int, int, float,
                                float, float, float, float, hipStream_t);
+extern "C" void launch_k3_colsum_bf16(void*, const float*, const float*,
+                                      float*, int, int, float, float,
+                                      hipStream_t);
 extern "C" void launch_kf_mfma(const float*, const long long*, const int*,
                                const float*, const int*, float*, double*,
                                const float*, float*, int, int, int, float,
@@ -254,6 +257,31 @@ void fused_grad_ls(torch::Tensor F, torch::Tensor indptr,
   }
 }
 
+// K3 + column partial sums fused (bf16 storage): commits F in place and
+// writes per-stripe fp32 column partials; sumF = partials.sum(0) on the
+// caller side (deterministic stage-2).
+void apply_step_colsum(torch::Tensor F_local, torch::Tensor grad,
+                       torch::Tensor steps, torch::Tensor partials,
+                       double min_f, double max_f) {
+  TORCH_CHECK(F_local.scalar_type() == torch::kBFloat16 &&
+                  F_local.is_contiguous(),
+              "apply_step_colsum is the bf16 path");
+  CHECK_IN(grad, torch::kFloat32);
+  CHECK_IN(steps, torch::kFloat32);
+  CHECK_IN(partials, torch::kFloat32);
+  const int n_local = (int)F_local.size(0);
+  const int K = (int)F_local.size(1);
+  TORCH_CHECK(K % 8 == 0, "bf16 K must be padded to a multiple of 8");
+  TORCH_CHECK(grad.size(0) == n_local && grad.size(1) == K);
+  TORCH_CHECK(steps.size(0) == n_local);
+  TORCH_CHECK(partials.size(0) == (n_local + 511) / 512 &&
+              partials.size(1) == K);
+  launch_k3_colsum_bf16(F_local.data_ptr(), grad.data_ptr<float>(),
+                        steps.data_ptr<float>(), partials.data_ptr<float>(),
+                        n_local, K, (float)min_f, (float)max_f,
+                        current_stream());
+}
+
 // On-device MFMA C/D-layout probe: D = A @ B with Bc = B column-major.
 void mfma_probe(torch::Tensor A, torch::Tensor Bc, torch::Tensor D) {
   TORCH_CHECK(D.scalar_type() == torch::kFloat32 && D.is_contiguous());
@@ -297,6 +325,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("fused_grad_ls", &fused_grad_ls,
         "KF: fused K1 gradient+LLH and K2 line search, one pass per node "
         "(MFMA tiles for the high-degree prefix)");
+  m.def("apply_step_colsum", &apply_step_colsum,
+        "K3+colsum fused (bf16): commit F and emit per-stripe column sums");
   m.def("mfma_probe", &mfma_probe,
         "MFMA C/D layout probe: one 16x16 D = A @ Bc^T tile");
 }

@@ -22,6 +22,16 @@ extern "C" void launch_k3(float*, const float*, const float*, int, int, float,
 extern "C" void launch_k4(const float*, const long long*, const int*,
                           const float*, const int*, double*, int, int, float,
                           float, hipStream_t);
+extern "C" void launch_k1_bf16(const void*, const long long*, const int*,
+                               const float*, const int*, float*, double*,
+                               int, int, float, float, hipStream_t);
+extern "C" void launch_k2_bf16(const void*, const long long*, const int*,
+                               const float*, const float*, const double*,
+                               const int*, const float*, float*, int, int,
+                               int, float, float, float, float, float,
+                               hipStream_t);
+extern "C" void launch_k3_bf16(void*, const float*, const float*, int, int,
+                               float, float, hipStream_t);
 extern "C" void launch_kf(const float*, const long long*, const int*,
                           const float*, const int*, float*, double*,
                           const float*, float*, int, int, int, float, float,
@@ -202,6 +212,19 @@ int main(int argc, char** argv) {
       printf("  (n_mfma=%d)\n", n_hi);
       time3(name, [&] { kfb_split(n_hi); });
     }
+    // bf16 separate kernels (phase A of the fused kernel == k1 structure,
+    // so k1b approximates the fused kernel's phase-A share)
+    time3("k1b", [&] {
+      launch_k1_bf16(dFb, dIp, dIdx, dSum, dOrd, dG, dLlh, N, Kb, 1e-4f,
+                     0.9999f, 0);
+    });
+    time3("k2b", [&] {
+      launch_k2_bf16(dFb, dIp, dIdx, dSum, dG, dLlh, dOrd, dLad, dBest, N,
+                     Kb, 16, 0.05f, 1e-4f, 0.9999f, 0.f, 1000.f, 0);
+    });
+    time3("k3b", [&] {
+      launch_k3_bf16(dFb, dG, dBest, N, Kb, 0.f, 1000.f, 0);
+    });
     HIP_CHECK(hipFree(dFb));
   }
   return 0;

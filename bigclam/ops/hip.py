@@ -151,6 +151,23 @@ def apply_step(
     ext.apply_step(F_local, grad, steps, cfg.min_f, cfg.max_f)
 
 
+def apply_step_colsum(
+    F_local: torch.Tensor,
+    grad: torch.Tensor,
+    steps: torch.Tensor,
+    partials: torch.Tensor,
+    cfg: BigClamConfig,
+) -> torch.Tensor:
+    """K3+colsum fused (bf16 storage): commits F in place and returns the
+    exact fp32 column sums of the updated shard (one pass over F instead
+    of commit + fp32 materialize + torch reduce)."""
+    ext = ensure_loaded()
+    ext.apply_step_colsum(
+        F_local, grad, steps, partials, cfg.min_f, cfg.max_f
+    )
+    return partials.sum(dim=0)
+
+
 def conductance_full_graph(graph, device) -> "torch.Tensor":
     """K5: ego-net conductance of every node of the FULL graph on GPU.
 

@@ -337,3 +337,26 @@ def test_fused_mfma_split_dispatch_fp32(monkeypatch):
     torch.testing.assert_close(llh_s, llh_d, rtol=0, atol=0)
     agree = (best_s == best_d).float().mean().item()
     assert agree > 0.98, f"only {agree:.3f} of best-steps agree"
+
+
+def test_apply_step_colsum_bf16():
+    """Fused K3+colsum == separate K3 then torch column sum: F bitwise,
+    sumF to fp32 reduction-order tolerance."""
+    from bigclam.ops.hip import ensure_loaded
+
+    ext = ensure_loaded()
+    g = rmat_graph(10, 7.0, seed=56)
+    cfg, st = _mkstate_dtype(g, 200, "bf16", seed=20)
+    grad, llh, best = st.fused_grad_ls_overlap(None)
+    F_a = st.F_local.clone()
+    F_b = st.F_local.clone()
+    ext.apply_step(F_a, grad, best, cfg.min_f, cfg.max_f)
+    sum_a = F_a.float().sum(dim=0)
+    ns = (F_b.shape[0] + 511) // 512
+    partials = torch.empty(
+        ns, F_b.shape[1], device="cuda", dtype=torch.float32
+    )
+    ext.apply_step_colsum(F_b, grad, best, partials, cfg.min_f, cfg.max_f)
+    sum_b = partials.sum(dim=0)
+    assert torch.equal(F_a, F_b)
+    torch.testing.assert_close(sum_b, sum_a, rtol=1e-5, atol=1e-2)
