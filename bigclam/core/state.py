@@ -202,7 +202,11 @@ class ShardState:
             return False
         if self.storage_dtype == torch.float32:
             return self.kp <= 8192
-        return self.kp <= 16384
+        if self.kp <= 16384:
+            return True
+        # beyond the direct kernel's K cap, the MFMA kernel (fu via LDS
+        # only, K*6 bytes <= 160 KB) must cover EVERY node
+        return self.kp <= 26000 and self.n_mfma == int(self.order.numel())
 
     def fused_grad_ls_overlap(
         self, halo_work

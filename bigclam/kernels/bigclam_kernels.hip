@@ -38,6 +38,8 @@
 #define NWAVE (BLOCK / WAVE)
 #define MAX_LS 16  // ladder length (ls_steps + 1); reference uses 16
 
+typedef float v2f __attribute__((ext_vector_type(2)));  // v_pk_* pairs
+
 // ---------------------------------------------------------------- reductions
 
 __device__ __forceinline__ float wave_allreduce_sum(float v) {
@@ -110,11 +112,9 @@ __device__ __forceinline__ float4 ld4(const float* p) {
   return *reinterpret_cast<const float4*>(p);
 }
 
-// 2-wide fp32 vectors lowered to CDNA packed-math (v_pk_fma_f32 /
-// v_pk_max_f32 / v_pk_min_f32): one instruction per pair, doubling VALU
-// throughput in the K2 inner loop.
-typedef float v2f __attribute__((ext_vector_type(2)));
-
+// 2-wide fp32 vectors (v2f, defined with the reduction helpers) lower to
+// CDNA packed-math (v_pk_fma_f32): one instruction per pair, doubling
+// VALU throughput in the K2 inner loop.
 __device__ __forceinline__ v2f pk_clamp_fma(v2f s, v2f g, v2f a, v2f lo,
                                             v2f hi) {
   // v_pk_fma_f32 + two v_med3_f32 (med3 = whole clamp in ONE VALU op;
@@ -148,12 +148,15 @@ struct f32x8 {
   float4 a, b;
 };
 
-__device__ __forceinline__ f32x8 ld8bf(const u32* p) {
-  const uint4 u = *reinterpret_cast<const uint4*>(p);
+__device__ __forceinline__ f32x8 ld8bf_u(uint4 u) {
   f32x8 r;
   r.a = float4{bf_lo(u.x), bf_hi(u.x), bf_lo(u.y), bf_hi(u.y)};
   r.b = float4{bf_lo(u.z), bf_hi(u.z), bf_lo(u.w), bf_hi(u.w)};
   return r;
+}
+
+__device__ __forceinline__ f32x8 ld8bf(const u32* p) {
+  return ld8bf_u(*reinterpret_cast<const uint4*>(p));
 }
 
 __device__ __forceinline__ u32 pack_bf16_rne(float lo, float hi) {
@@ -1299,6 +1302,94 @@ __device__ __forceinline__ void kf_phase_a_bf16(
   __syncthreads();
 }
 
+// bf16 phase A without register-resident fu/fv (K beyond the NSLOT
+// register budget, e.g. the com-Amazon K=25000 config): fu lives in LDS
+// only and the weighted accumulate re-reads fv from global (L2-hot, the
+// block just streamed it for the dot).  Covers K up to the LDS cap
+// (gacc fp32 + fu bf16 = K*6 bytes <= 160 KB -> K <= 26000 padded).
+__device__ __forceinline__ void kf_phase_a_bf16_lds(
+    const u32* __restrict__ F, const int* __restrict__ indices,
+    const float* __restrict__ sumF, int u, long long e0, long long e1, int K,
+    float min_p, float max_p, float* __restrict__ gacc, u32* __restrict__ fu_s,
+    float* red, double* s_llh_u, float* __restrict__ grad,
+    double* __restrict__ llh) {
+  const int tid = threadIdx.x;
+  const u32* __restrict__ fu_g = F + (size_t)u * (K / 2);
+  for (int k = tid * 8; k < K; k += BLOCK * 8) {
+    *reinterpret_cast<uint4*>(fu_s + k / 2) =
+        *reinterpret_cast<const uint4*>(fu_g + k / 2);
+    *reinterpret_cast<float4*>(gacc + k) = float4{0.f, 0.f, 0.f, 0.f};
+    *reinterpret_cast<float4*>(gacc + k + 4) = float4{0.f, 0.f, 0.f, 0.f};
+  }
+  __syncthreads();
+
+  double llh_acc = 0.0;
+#pragma clang loop unroll(disable)
+  for (long long e = e0; e < e1; ++e) {
+    const u32* __restrict__ fv = F + (size_t)indices[e] * (K / 2);
+    float part = 0.f;
+    for (int k = tid * 8; k < K; k += BLOCK * 8) {
+      const f32x8 a = ld8bf(fu_s + k / 2);
+      const f32x8 b = ld8bf(fv + k / 2);
+      part = dot8(a, b, part);
+    }
+    const float x = block_allreduce_sum(part, red);
+    const float p = clamp_p(__expf(-x), min_p, max_p);
+    const float w = 1.f / (1.f - p);
+    if (tid == 0) llh_acc += (double)log1pf(-p) + (double)x;
+    for (int k = tid * 8; k < K; k += BLOCK * 8) {
+      const f32x8 b = ld8bf(fv + k / 2);  // L2 re-read
+      float4 g0 = ld4(gacc + k);
+      float4 g1 = ld4(gacc + k + 4);
+      g0.x = fmaf(w, b.a.x, g0.x);
+      g0.y = fmaf(w, b.a.y, g0.y);
+      g0.z = fmaf(w, b.a.z, g0.z);
+      g0.w = fmaf(w, b.a.w, g0.w);
+      g1.x = fmaf(w, b.b.x, g1.x);
+      g1.y = fmaf(w, b.b.y, g1.y);
+      g1.z = fmaf(w, b.b.z, g1.z);
+      g1.w = fmaf(w, b.b.w, g1.w);
+      *reinterpret_cast<float4*>(gacc + k) = g0;
+      *reinterpret_cast<float4*>(gacc + k + 4) = g1;
+    }
+  }
+
+  float p_fs = 0.f, p_ff = 0.f;
+  for (int k = tid * 8; k < K; k += BLOCK * 8) {
+    const f32x8 a = ld8bf(fu_s + k / 2);
+    const float4 s0 = ld4(sumF + k);
+    const float4 s1 = ld4(sumF + k + 4);
+    p_fs = dot4(a.a, s0, p_fs);
+    p_fs = dot4(a.b, s1, p_fs);
+    p_ff = dot8(a, a, p_ff);
+  }
+  const float fs = block_allreduce_sum(p_fs, red);
+  const float ff = block_allreduce_sum(p_ff, red);
+
+  float* __restrict__ gout = grad + (size_t)u * K;
+  for (int k = tid * 8; k < K; k += BLOCK * 8) {
+    const f32x8 a = ld8bf(fu_s + k / 2);
+    const float4 g0a = ld4(gacc + k);
+    const float4 g1a = ld4(gacc + k + 4);
+    const float4 s0 = ld4(sumF + k);
+    const float4 s1 = ld4(sumF + k + 4);
+    const float4 g0 = float4{g0a.x - s0.x + a.a.x, g0a.y - s0.y + a.a.y,
+                             g0a.z - s0.z + a.a.z, g0a.w - s0.w + a.a.w};
+    const float4 g1 = float4{g1a.x - s1.x + a.b.x, g1a.y - s1.y + a.b.y,
+                             g1a.z - s1.z + a.b.z, g1a.w - s1.w + a.b.w};
+    *reinterpret_cast<float4*>(gacc + k) = g0;
+    *reinterpret_cast<float4*>(gacc + k + 4) = g1;
+    *reinterpret_cast<float4*>(gout + k) = g0;
+    *reinterpret_cast<float4*>(gout + k + 4) = g1;
+  }
+  if (tid == 0) {
+    const double l = llh_acc + (double)(-fs) + (double)ff;
+    llh[u] = l;
+    *s_llh_u = l;
+  }
+  __syncthreads();
+}
+
 // bf16 shared tail: node terms + Armijo pick (fu raw bf16 from LDS).
 __device__ __forceinline__ void kf_tail_bf16(
     const u32* __restrict__ fu, const float* __restrict__ gu,
@@ -1487,8 +1578,13 @@ __global__ void __launch_bounds__(BLOCK, 3) kf_mfma_bf16_t(
 
   if (tid < MAX_LS) s_lad[tid] = (tid < n_ladder) ? ladder[tid] : 0.f;
 
-  kf_phase_a_bf16<NSLOT>(F, indices, sumF, u, e0, e1, K, min_p, max_p, gacc,
-                         fu_s, red, &s_llh_u, grad, llh);
+  if constexpr (NSLOT == 0) {  // large-K path: fu in LDS only
+    kf_phase_a_bf16_lds(F, indices, sumF, u, e0, e1, K, min_p, max_p, gacc,
+                        fu_s, red, &s_llh_u, grad, llh);
+  } else {
+    kf_phase_a_bf16<NSLOT>(F, indices, sumF, u, e0, e1, K, min_p, max_p,
+                           gacc, fu_s, red, &s_llh_u, grad, llh);
+  }
 
   // ---------------- phase B: MFMA 16-edge x 16-candidate tiles
   // (k-quartered across waves; see kf_mfma_t)
@@ -2399,8 +2495,8 @@ extern "C" void launch_kf_mfma_bf16(const void* F, const long long* indptr,
                                     hipStream_t stream) {
   if (n_local == 0) return;
   if (n_ladder > 16) throw std::runtime_error("ladder length > 16 unsupported");
-  if (K > 16384)
-    throw std::runtime_error("bf16 fused kernel requires K <= 16384");
+  if (K > 26000)
+    throw std::runtime_error("bf16 MFMA fused kernel requires K <= 26000");
   const size_t lds = (size_t)K * 6;  // gacc fp32 + fu raw bf16
   const u32* Fb = reinterpret_cast<const u32*>(F);
 #define KFMB_CASE(NS)                                                        \
@@ -2417,8 +2513,10 @@ extern "C" void launch_kf_mfma_bf16(const void* F, const long long* indptr,
     KFMB_CASE(2);
   } else if (K <= 8192) {
     KFMB_CASE(4);
-  } else {
+  } else if (K <= 16384) {
     KFMB_CASE(8);
+  } else {
+    KFMB_CASE(0);  // fu via LDS only (see kf_phase_a_bf16_lds)
   }
 #undef KFMB_CASE
   HIP_CHECK(hipGetLastError());

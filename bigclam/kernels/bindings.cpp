@@ -225,8 +225,9 @@ void fused_grad_ls(torch::Tensor F, torch::Tensor indptr,
   const int n_lo = n_blocks - n_hi;
   const int* ord = order.data_ptr<int>();
   if (is_bf16(F)) {
-    TORCH_CHECK(K % 8 == 0 && K <= 16384,
-                "bf16 fused kernel: K padded to 8, K <= 16384");
+    TORCH_CHECK(K % 8 == 0 && K <= 26000,
+                "bf16 fused kernel: K padded to 8, K <= 26000 (the direct "
+                "kernel additionally requires K <= 16384 for its share)");
     launch_kf_mfma_bf16(F.data_ptr(), ip, indices.data_ptr<int>(),
                         sumF.data_ptr<float>(), ord, grad.data_ptr<float>(),
                         llh.data_ptr<double>(), ladder.data_ptr<float>(),

@@ -360,3 +360,63 @@ def test_apply_step_colsum_bf16():
     sum_b = partials.sum(dim=0)
     assert torch.equal(F_a, F_b)
     torch.testing.assert_close(sum_b, sum_a, rtol=1e-5, atol=1e-2)
+
+
+def test_fused_mfma_split_order_launches_match_full_bf16():
+    """Two subset launches (the ws>1 interior/boundary halo-overlap call
+    pattern) with per-subset MFMA prefixes == one full launch — grad/llh
+    bitwise, step picks identical (same kernel per node either way)."""
+    from bigclam.ops import hip as hip_ops
+
+    g = rmat_graph(10, 7.0, seed=57)
+    cfg, st = _mkstate_dtype(g, 192, "bf16", seed=21)
+    n = st.n_local
+    full = hip_ops.fused_grad_ls(
+        st.F, st.indptr, st.indices, st.sumF, st.order, cfg,
+        n_mfma=int(st.order.numel()),
+    )
+    # split the degree-sorted order into two interleaved subsets, each
+    # still degree-descending (as order_interior/order_boundary are)
+    o_np = st.order.cpu().numpy()
+    a = torch.from_numpy(o_np[::2].copy()).cuda()
+    b = torch.from_numpy(o_np[1::2].copy()).cuda()
+    grad = torch.empty(n, st.F.shape[1], device="cuda", dtype=torch.float32)
+    llh = torch.empty(n, device="cuda", dtype=torch.float64)
+    best = torch.empty(n, device="cuda", dtype=torch.float32)
+    out = (grad, llh, best)
+    hip_ops.fused_grad_ls(
+        st.F, st.indptr, st.indices, st.sumF, a, cfg, out=out,
+        n_mfma=int(a.numel()),
+    )
+    hip_ops.fused_grad_ls(
+        st.F, st.indptr, st.indices, st.sumF, b, cfg, out=out,
+        n_mfma=int(b.numel()),
+    )
+    torch.testing.assert_close(grad, full[0], rtol=0, atol=0)
+    torch.testing.assert_close(llh, full[1], rtol=0, atol=0)
+    torch.testing.assert_close(best, full[2], rtol=0, atol=0)
+
+
+def test_fused_mfma_large_k_bf16():
+    """K=17000 > the direct kernel's 16384 cap exercises the LDS-only
+    phase A (kf_phase_a_bf16_lds) + MFMA phase B; compared against the
+    torch reference ops (fp32 math on the same bf16 F)."""
+    from bigclam.ops import hip as hip_ops
+
+    g = rmat_graph(8, 6.0, seed=58)  # ~250 nodes: keep K=17k cheap
+    cfg, st = _mkstate_dtype(g, 17000, "bf16", seed=22, scale=0.02)
+    assert st.fused_ok and st.n_mfma == int(st.order.numel())
+    grad, llh, best = st.fused_grad_ls_overlap(None)
+    rgrad, rllh = ref_ops.edge_grad_llh(
+        st.F, st.indptr, st.indices, st.sumF, cfg, n_local=st.n_local
+    )
+    torch.testing.assert_close(
+        grad, rgrad, rtol=0.05, atol=0.05 * float(rgrad.abs().mean())
+    )
+    torch.testing.assert_close(llh, rllh, rtol=1e-4, atol=1.0)
+    rbest = ref_ops.linesearch(
+        st.F, st.indptr, st.indices, st.sumF, grad, llh, cfg,
+        n_local=st.n_local,
+    )
+    agree = (best == rbest).float().mean().item()
+    assert agree > 0.95, f"only {agree:.3f} of best-steps agree"
