@@ -171,7 +171,25 @@ class Trainer:
         res = FitResult()
         llh_old = 0.0
         carry, llh0 = self.prologue()
-        self.metrics.log({"sweep": -1, "llh": llh0, "note": "initial"})
+        # per-sweep collective volume (static per topology — SURVEY §5
+        # observability: per-collective bytes): halo p2p both directions,
+        # the 1xK sumF all-reduce and the scalar LLH all-reduce
+        st = self.state
+        esize = st.F.element_size()
+        self.metrics.log(
+            {
+                "sweep": -1,
+                "llh": llh0,
+                "note": "initial",
+                "world_size": self.world_size,
+                "comm_bytes_per_sweep": {
+                    "halo_send": int(st.shard.plan.total_send) * st.kp * esize,
+                    "halo_recv": int(st.shard.n_halo) * st.kp * esize,
+                    "sumF_allreduce": st.kp * 4,
+                    "llh_allreduce": 8,
+                },
+            }
+        )
         for i in range(cfg.max_sweeps):
             timer = PhaseTimer(sync=True)
             timer.start("sweep")
