@@ -91,8 +91,15 @@ class ShardState:
         # CSR (profiles/r01_kernel_opt_log.md): bf16 MFMA wins for EVERY
         # degree (16.4 vs 22.6 ms even with tile padding at mean degree
         # 5.5), so the bf16 default routes all nodes to it.
-        default_thr = "1" if self.storage_dtype == torch.bfloat16 else "0"
-        thr = int(os.environ.get("BIGCLAM_MFMA_DEG", default_thr))
+        # K-dependent (measured): at kp <= 8192 (NSLOT<=4, spill-free)
+        # MFMA-all wins; at 8192 < kp <= 16384 the NSLOT=8 template
+        # spills 20 B/lane and the direct kernel measures faster
+        # (com-Youtube K=8385: 149 vs 159 ms); above 16384 only the MFMA
+        # kernel covers the shape.
+        mfma_all = self.storage_dtype == torch.bfloat16 and (
+            self.kp <= 8192 or self.kp > 16384
+        )
+        thr = int(os.environ.get("BIGCLAM_MFMA_DEG", "1" if mfma_all else "0"))
         if thr > 0:
             self.n_mfma = int((deg[order_np] >= thr).sum())
             self.n_mfma_interior = int((deg[interior] >= thr).sum())

@@ -1598,20 +1598,23 @@ __global__ void __launch_bounds__(BLOCK, 3) kf_mfma_bf16_t(
   const int sq = (nkstep + NWAVE - 1) / NWAVE;
   const int sw0 = wid * sq;
   const int sw1 = min(nkstep, sw0 + sq);
-  constexpr int KC = 8;  // candidate fragments built per burst (256 k)
+  // NSLOT=8 keeps 64 VGPRs of fu resident through phase B: halve the
+  // candidate burst and accumulator group there to stay spill-free
+  constexpr int KC = (NSLOT > 4) ? 4 : 8;  // candidate frags per burst
+  constexpr int TG = (NSLOT > 4) ? 2 : KF_TG;  // acc tiles per group
   double llh_j = 0.0;
 
-  for (int tbase = 0; tbase < ntiles; tbase += KF_TG) {
-    const int nt = min(KF_TG, ntiles - tbase);
-    const u32* aptr[KF_TG];
+  for (int tbase = 0; tbase < ntiles; tbase += TG) {
+    const int nt = min(TG, ntiles - tbase);
+    const u32* aptr[TG];
 #pragma unroll
-    for (int t = 0; t < KF_TG; ++t) {
+    for (int t = 0; t < TG; ++t) {
       const long long e = e0 + (long long)(tbase + t) * 16 + arow;
       aptr[t] = (e < e1) ? F + (size_t)indices[e] * (K / 2) : nullptr;
     }
-    f32x4v acc[KF_TG];
+    f32x4v acc[TG];
 #pragma unroll
-    for (int t = 0; t < KF_TG; ++t) acc[t] = f32x4v{0.f, 0.f, 0.f, 0.f};
+    for (int t = 0; t < TG; ++t) acc[t] = f32x4v{0.f, 0.f, 0.f, 0.f};
 
     for (int sb = sw0; sb < sw1; sb += KC) {
       bf16x8 bfrag[KC];
@@ -1638,7 +1641,7 @@ __global__ void __launch_bounds__(BLOCK, 3) kf_mfma_bf16_t(
         }
       }
 #pragma unroll
-      for (int t = 0; t < KF_TG; ++t) {
+      for (int t = 0; t < TG; ++t) {
         if (t >= nt) break;
 #pragma unroll
         for (int c = 0; c < KC; ++c) {
@@ -1654,7 +1657,7 @@ __global__ void __launch_bounds__(BLOCK, 3) kf_mfma_bf16_t(
       }
     }
 #pragma unroll
-    for (int t = 0; t < KF_TG; ++t) {
+    for (int t = 0; t < TG; ++t) {
       if (t >= nt) break;
       __syncthreads();  // redt free / previous tile consumed
       *reinterpret_cast<float4*>(&redt[wid][lane][0]) =

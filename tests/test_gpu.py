@@ -420,3 +420,22 @@ def test_fused_mfma_large_k_bf16():
     )
     agree = (best == rbest).float().mean().item()
     assert agree > 0.95, f"only {agree:.3f} of best-steps agree"
+
+
+def test_fused_mfma_matches_direct_bf16_nslot8():
+    """K=8500 exercises the NSLOT=8 template pair (the 8192<K<=16384
+    regime where the DIRECT kernel is the default)."""
+    from bigclam.ops import hip as hip_ops
+
+    g = rmat_graph(8, 6.0, seed=59)
+    cfg, st = _mkstate_dtype(g, 8500, "bf16", seed=23, scale=0.05)
+    assert st.fused_ok and st.n_mfma == 0  # direct is the default here
+    args = (st.F, st.indptr, st.indices, st.sumF, st.order, cfg)
+    grad_d, llh_d, best_d = hip_ops.fused_grad_ls(*args, n_mfma=0)
+    grad_m, llh_m, best_m = hip_ops.fused_grad_ls(
+        *args, n_mfma=int(st.order.numel())
+    )
+    torch.testing.assert_close(grad_m, grad_d, rtol=0, atol=0)
+    torch.testing.assert_close(llh_m, llh_d, rtol=0, atol=0)
+    agree = (best_m == best_d).float().mean().item()
+    assert agree > 0.95, f"only {agree:.3f} of best-steps agree"
