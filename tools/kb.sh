@@ -1,0 +1,3 @@
+#!/bin/bash
+cd /root/repo
+exec ./bigclam/kernels/kbench "${1:-5000}"
