@@ -1551,7 +1551,7 @@ __global__ void __launch_bounds__(BLOCK, 3) kf_fused_bf16_t(
 // uint4 per lane), the B fragment is the clamped candidate row built in
 // fp32 from LDS fu/grad and packed to bf16 (RNE, same rounding as the K3
 // bf16 commit).  Candidate fragments are built once per 256-element
-// k-chunk per wave and reused across KF_TCAP edge tiles.
+// k-chunk per wave and reused across the accumulator tile group.
 template <int NSLOT>
 __global__ void __launch_bounds__(BLOCK, 3) kf_mfma_bf16_t(
     const u32* __restrict__ F, const long long* __restrict__ indptr,

@@ -93,3 +93,35 @@ def test_bf16_storage_path(small_graph):
     np.testing.assert_allclose(
         grad_b.numpy(), grad_f.numpy(), rtol=0.05, atol=0.05
     )
+
+
+def test_mfma_dispatch_prefix_invariant(monkeypatch):
+    """The kernel dispatch assumes each launch-order list is degree-
+    descending so the deg>=threshold nodes form a PREFIX of length
+    n_mfma; verify on sharded states (interior/boundary included)."""
+    import numpy as np
+    import torch
+
+    from bigclam.config import BigClamConfig
+    from bigclam.core.shard import make_shard
+    from bigclam.core.state import ShardState
+    from bigclam.io import rmat_graph
+
+    monkeypatch.setenv("BIGCLAM_MFMA_DEG", "8")
+    g = rmat_graph(9, 6.0, seed=77)
+    for ws, rank in [(1, 0), (3, 1)]:
+        shard = make_shard(g, rank, ws)
+        st = ShardState(
+            shard, BigClamConfig(k=16, device="cpu"),
+            device=torch.device("cpu"),
+        )
+        deg = shard.degrees()
+        for order, n_hi in [
+            (st.order, st.n_mfma),
+            (st.order_interior, st.n_mfma_interior),
+            (st.order_boundary, st.n_mfma_boundary),
+        ]:
+            o = order.cpu().numpy()
+            assert np.all(deg[o[:n_hi]] >= 8)
+            if n_hi < len(o):
+                assert np.all(deg[o[n_hi:]] < 8)
