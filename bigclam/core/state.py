@@ -132,6 +132,20 @@ class ShardState:
         )
         if self.use_hip:
             _hip_ops().ensure_loaded()  # fail loudly if the .so is missing
+            # fail at construction, not at the first kernel launch, when
+            # the requested K exceeds what the bf16 kernels cover
+            if self.storage_dtype == torch.bfloat16 and self.kp > 16384:
+                if self.kp > 26000:
+                    raise ValueError(
+                        f"bf16 GPU path covers K <= 26000 (got padded "
+                        f"K={self.kp}); use dtype=fp32 for larger K"
+                    )
+                if self.n_mfma != int(self.order.numel()):
+                    raise ValueError(
+                        "bf16 with 16384 < K <= 26000 requires the MFMA "
+                        "fused kernel on every node; unset BIGCLAM_MFMA_DEG "
+                        "or set it to 1"
+                    )
 
     # ------------------------------------------------------------------ util
     @property
