@@ -9,7 +9,14 @@
 //   K2 linesearch     — all 16 Armijo candidates per node in one edge pass
 //                       (replaces the cartesian fan-out, scala:153-163)
 //   K3 apply_step     — projected commit (scala:89-92, 171)
+//   K3CS k3_colsum    — bf16 commit + fp32 column partial sums, one pass
 //   K4 llh_only       — per-node local LLH (scala:106-120 / 177-200)
+//   K5 k5_conductance — ego-net conductance seed ranking (scala:39-54)
+//   KF kf_fused/kf_mfma — K1+K2 fused per node; the MFMA variants run
+//                       the 16-candidate scoring as (deg x K)@(K x 16)
+//                       GEMM tiles on v_mfma_f32_16x16x32_bf16 /
+//                       v_mfma_f32_16x16x4_f32 (dispatch is K-aware,
+//                       see launch_kf_mfma* and core/state.py)
 //
 // Design notes (measured rationale in profiles/r01_kernel_opt_log.md):
 //  * K1 is BLOCK-per-node: per edge the workgroup does one cooperative dot

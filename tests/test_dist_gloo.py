@@ -1,4 +1,4 @@
-"""Multi-process shard correctness on CPU (gloo, world_size 2-3).
+"""Multi-process shard correctness on CPU (gloo, world_size 2-4).
 
 The halo-exchange + allreduce path must produce results identical (to fp32
 tolerance) to the single-shard run — RCCL on GPU exercises the exact same
@@ -57,7 +57,7 @@ def _worker(rank, world_size, port, out_dir, n_sweeps):
         dist.destroy_process_group()
 
 
-@pytest.mark.parametrize("world_size", [2, 3])
+@pytest.mark.parametrize("world_size", [2, 3, 4])
 def test_sharded_equals_single(world_size, tmp_path):
     n_sweeps = 3
     llh1, F1 = _single_run(n_sweeps)
