@@ -100,7 +100,13 @@ class ShardState:
             self.kp <= 8192 or self.kp > 16384
         )
         thr = int(os.environ.get("BIGCLAM_MFMA_DEG", "1" if mfma_all else "0"))
-        if thr > 0:
+        if thr == 1:
+            # "all nodes": include any degree-0 rows too (the MFMA kernel
+            # handles them; the direct kernel rejects K > 16384)
+            self.n_mfma = int(order_np.size)
+            self.n_mfma_interior = int(interior.size)
+            self.n_mfma_boundary = int(bnd.size)
+        elif thr > 0:
             self.n_mfma = int((deg[order_np] >= thr).sum())
             self.n_mfma_interior = int((deg[interior] >= thr).sum())
             self.n_mfma_boundary = int((deg[bnd] >= thr).sum())
