@@ -61,6 +61,9 @@ extern "C" void launch_kf_mfma_bf16(const void*, const long long*, const int*,
 
 int main(int argc, char** argv) {
   int K = argc > 1 ? atoi(argv[1]) : 5000;
+  // mode "kfb0": ONLY the bf16 all-MFMA fused timing (for rocprofv3
+  // --pmc runs, where every extra dispatch multiplies collection passes)
+  const bool only_kfb0 = argc > 2 && strcmp(argv[2], "kfb0") == 0;
 
   FILE* fp = fopen("gpurun_out_graph.bin", "rb");
   if (!fp) {
@@ -132,17 +135,17 @@ int main(int argc, char** argv) {
     fflush(stdout);
   };
 
-  time3("k1", [&] {
+  if (!only_kfb0) time3("k1", [&] {
     launch_k1(dF, dIp, dIdx, dSum, dOrd, dG, dLlh, N, K, 1e-4f, 0.9999f, 0);
   });
-  time3("k4", [&] {
+  if (!only_kfb0) time3("k4", [&] {
     launch_k4(dF, dIp, dIdx, dSum, dOrd, dLlh, N, K, 1e-4f, 0.9999f, 0);
   });
-  time3("k2", [&] {
+  if (!only_kfb0) time3("k2", [&] {
     launch_k2(dF, dIp, dIdx, dSum, dG, dLlh, dOrd, dLad, dBest, N, K, 16,
               0.05f, 1e-4f, 0.9999f, 0.f, 1000.f, 0);
   });
-  time3("k3", [&] {
+  if (!only_kfb0) time3("k3", [&] {
     launch_k3(dF, dG, dBest, N, K, 0.f, 1000.f, 0);
   });
 
@@ -159,7 +162,7 @@ int main(int argc, char** argv) {
               N - n_hi, K, 16, 0.05f, 1e-4f, 0.9999f, 0.f, 1000.f, 0);
   };
   const int thrs[] = {0, 8, 16, 32, 64, 1 << 30};
-  if (K <= 8192) {  // fp32 fused-kernel K cap
+  if (K <= 8192 && !only_kfb0) {  // fp32 fused-kernel K cap
     for (int thr : thrs) {
       const int n_hi = thr == 0 ? N : (thr == (1 << 30) ? 0 : split_at(thr));
       char name[32];
@@ -208,6 +211,7 @@ int main(int argc, char** argv) {
                      1000.f, 0);
     };
     for (int thr : thrs) {
+      if (only_kfb0 && thr != 0) continue;
       const int n_hi = thr == 0 ? N : (thr == (1 << 30) ? 0 : split_at(thr));
       char name[32];
       snprintf(name, sizeof name, "kfb@%-4d", thr == (1 << 30) ? -1 : thr);
@@ -216,15 +220,15 @@ int main(int argc, char** argv) {
     }
     // bf16 separate kernels (phase A of the fused kernel == k1 structure,
     // so k1b approximates the fused kernel's phase-A share)
-    time3("k1b", [&] {
+    if (!only_kfb0) time3("k1b", [&] {
       launch_k1_bf16(dFb, dIp, dIdx, dSum, dOrd, dG, dLlh, N, Kb, 1e-4f,
                      0.9999f, 0);
     });
-    time3("k2b", [&] {
+    if (!only_kfb0) time3("k2b", [&] {
       launch_k2_bf16(dFb, dIp, dIdx, dSum, dG, dLlh, dOrd, dLad, dBest, N,
                      Kb, 16, 0.05f, 1e-4f, 0.9999f, 0.f, 1000.f, 0);
     });
-    time3("k3b", [&] {
+    if (!only_kfb0) time3("k3b", [&] {
       launch_k3_bf16(dFb, dG, dBest, N, Kb, 0.f, 1000.f, 0);
     });
     HIP_CHECK(hipFree(dFb));

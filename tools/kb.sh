@@ -1,3 +1,3 @@
 #!/bin/bash
 cd /root/repo
-exec ./bigclam/kernels/kbench "${1:-5000}"
+exec ./bigclam/kernels/kbench "${1:-5000}" "${2:-}"
