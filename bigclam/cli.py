@@ -14,6 +14,7 @@ from __future__ import annotations
 import argparse
 import dataclasses
 import json
+import os
 import sys
 
 import numpy as np
@@ -105,6 +106,24 @@ def main(argv=None):
     p_ext.add_argument("checkpoint_dir")
     p_ext.add_argument("edgelist")
     p_ext.add_argument("--out", required=True)
+
+    sub.add_parser(
+        "bench",
+        add_help=False,
+        description="flagship benchmark; all flags forwarded to bench.py",
+    )
+
+    if argv is None:
+        argv = sys.argv[1:]
+    if argv and argv[0] == "bench":
+        # delegate to the repo-root bench harness (same contract the
+        # scaling driver uses); works installed or from a checkout
+        import runpy
+
+        sys.argv = ["bench.py"] + list(argv[1:])
+        here = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+        runpy.run_path(os.path.join(here, "bench.py"), run_name="__main__")
+        return 0
 
     args = ap.parse_args(argv)
     rank = comm.init_distributed()

@@ -210,6 +210,12 @@ class Trainer:
                     "llh": llh,
                     "rel_change": abs(1.0 - llh / llh_old) if llh_old else None,
                     "accepted_frac": float((steps > 0).float().mean().item()),
+                    # F-row density: rows densify early and sparsify as
+                    # the fit converges (input to a sparsity-adaptive
+                    # candidate path; reference v3 keeps rows sparse)
+                    "f_nnz_frac": float(
+                        (self.state.F_local != 0).float().mean().item()
+                    ),
                     "step_hist": hist.tolist(),
                     "sweep_s": dt,
                     "edges_per_s": self.graph.num_directed_edges / dt if dt else None,
