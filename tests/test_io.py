@@ -40,6 +40,25 @@ def test_parse_file(tmp_path):
     assert g.num_edges == 3
 
 
+def test_parse_file_messy(tmp_path):
+    """Real-world edge-list quirks: CRLF line endings, repeated tabs and
+    spaces, leading whitespace, blank lines, no trailing newline."""
+    p = tmp_path / "messy.txt"
+    p.write_bytes(
+        b"# header\r\n"
+        b"\r\n"
+        b"1\t\t2\r\n"
+        b"  2   3\n"
+        b"\n"
+        b"3\t 1"
+    )
+    arr = parse_edge_array(str(p))
+    assert arr.shape == (3, 2)
+    g = load_graph(str(p))
+    assert g.num_nodes == 3
+    assert g.num_edges == 3
+
+
 def test_rmat_properties():
     g = rmat_graph(10, 8.0, seed=3)
     assert g.num_nodes <= 1024
