@@ -86,16 +86,14 @@ class ShardState:
             bnd = order_np[:0]
         # MFMA split: each launch list is degree-descending, so the nodes
         # taking the MFMA phase-B kernel (deg >= threshold) are a prefix;
-        # precompute the split points.  BIGCLAM_MFMA_DEG=0 disables (all
-        # nodes on the direct kernel).  Measured on the com-Amazon-shaped
-        # CSR (profiles/r01_kernel_opt_log.md): bf16 MFMA wins for EVERY
-        # degree (16.4 vs 22.6 ms even with tile padding at mean degree
-        # 5.5), so the bf16 default routes all nodes to it.
-        # K-dependent (measured): at kp <= 8192 (NSLOT<=4, spill-free)
-        # MFMA-all wins; at 8192 < kp <= 16384 the NSLOT=8 template
-        # spills 20 B/lane and the direct kernel measures faster
-        # (com-Youtube K=8385: 149 vs 159 ms); above 16384 only the MFMA
-        # kernel covers the shape.
+        # precompute the split points.  BIGCLAM_MFMA_DEG overrides
+        # (0 = all nodes on the direct kernel).  Default is K-dependent,
+        # measured in profiles/r01_kernel_opt_log.md: at kp <= 8192
+        # (NSLOT<=4, spill-free) MFMA-all wins at EVERY degree (16.4 vs
+        # 22.6 ms on the com-Amazon CSR, mean degree 5.5); at
+        # 8192 < kp <= 16384 the NSLOT=8 template spills 20 B/lane and
+        # the direct kernel measures faster (com-Youtube K=8385: 149 vs
+        # 159 ms); above 16384 only the MFMA kernel covers the shape.
         mfma_all = self.storage_dtype == torch.bfloat16 and (
             self.kp <= 8192 or self.kp > 16384
         )
