@@ -103,3 +103,20 @@ def test_cli_select_k(tmp_path):
     assert r.returncode == 0, r.stderr
     res = json.loads(r.stdout.strip().splitlines()[-1])
     assert "grid" in res and res["grid"][0] == 2
+
+
+def test_cli_bench_delegates(capfd):
+    """`python -m bigclam bench` forwards to the repo-root harness and
+    prints the driver-contract JSON line."""
+    import json as _json
+
+    from bigclam.cli import main
+
+    main([
+        "bench", "--steps", "1", "--warmup", "0", "--nodes", "400",
+        "--edges", "1200", "--k", "16",
+    ])
+    out = capfd.readouterr().out.strip().splitlines()[-1]
+    rec = _json.loads(out)
+    assert rec["metric"] == "edges/sec per grad iter"
+    assert rec["steps"] == 1 and rec["n_gpus"] == 1
