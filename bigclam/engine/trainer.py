@@ -213,9 +213,17 @@ class Trainer:
                     # F-row density: rows densify early and sparsify as
                     # the fit converges (input to a sparsity-adaptive
                     # candidate path; reference v3 keeps rows sparse)
-                    "f_nnz_frac": float(
+                    "f_nnz_frac": (nnz := float(
                         (self.state.F_local != 0).float().mean().item()
-                    ),
+                    )),
+                    # all-zero F is an ABSORBING state (grad == -sumF == 0,
+                    # llh stuck at the x=0 floor): surface it.  Observed
+                    # with the indicator seed init at large K (e.g.
+                    # Email-Enron K=500) — the reference's init has the
+                    # same property; --init random recovers.
+                    **({"warning": "F collapsed to all-zero (absorbing "
+                        "state); try --init random or smaller K"}
+                       if nnz == 0.0 else {}),
                     "step_hist": hist.tolist(),
                     "sweep_s": dt,
                     "edges_per_s": self.graph.num_directed_edges / dt if dt else None,
