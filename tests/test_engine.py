@@ -150,3 +150,33 @@ def test_select_k_exhausted_grid_returns_last_k():
     )
     out = select_k(g, cfg, init="random")
     assert out["k"] == out["grid"][-1] != 0
+
+
+def test_collapse_warning_logged():
+    """All-zero F is an absorbing state; the trainer surfaces it."""
+    import torch
+
+    from bigclam.config import BigClamConfig
+    from bigclam.engine.trainer import Trainer
+    from bigclam.io import rmat_graph
+    from bigclam.utils.metrics import MetricsLogger
+
+    class Capture(MetricsLogger):
+        def __init__(self):
+            super().__init__(rank=0, quiet=True)
+            self.records = []
+
+        def log(self, record):
+            self.records.append(dict(record))
+
+    g = rmat_graph(8, 5.0, seed=4)
+    cfg = BigClamConfig(k=16, device="cpu", max_sweeps=2, seed=0)
+    cap = Capture()
+    tr = Trainer(g, cfg, rank=0, world_size=1, device=torch.device("cpu"),
+                 metrics=cap)
+    tr.state.set_local_F(torch.zeros(g.num_nodes, cfg.k))
+    tr.fit(skip_init=True)
+    warns = [r for r in cap.records if "warning" in r]
+    assert warns and "absorbing" in warns[0]["warning"]
+    assert all(r.get("f_nnz_frac", 1.0) == 0.0 for r in cap.records
+               if "f_nnz_frac" in r)
