@@ -180,3 +180,15 @@ def test_collapse_warning_logged():
     assert warns and "absorbing" in warns[0]["warning"]
     assert all(r.get("f_nnz_frac", 1.0) == 0.0 for r in cap.records
                if "f_nnz_frac" in r)
+
+
+def test_k_grid_matches_reference_repl_output():
+    """codes/bigclam4-7.scala:268 contains a pasted REPL OUTPUT of the
+    reference's own K grid (Array(50, 54, ..., 200)); k_grid reproduces
+    it exactly at (min=50, max=200, div=15)."""
+    from bigclam.config import k_grid
+
+    assert k_grid(50, 200, 15) == [
+        50, 54, 59, 64, 70, 76, 83, 91, 99, 108, 118, 129, 141, 154,
+        168, 184, 200,
+    ]
