@@ -138,3 +138,25 @@ def test_fit_real_enron_smoke():
     assert len(res.llh_history) == 2
     assert all(l == l for l in res.llh_history)  # finite
     assert res.llh_history[-1] > res.llh_history[0]
+
+
+@_pytest.mark.skipif(
+    not os.path.isdir(_REF_DATA), reason="reference datasets not mounted"
+)
+def test_fit_facebook_k25_baseline_config():
+    """BASELINE.json config #1: facebook_combined K=25 on CPU (gloo ws=1
+    plumbing config) — converging fit on the real fixture."""
+    import numpy as np
+    import torch
+
+    from bigclam.config import BigClamConfig
+    from bigclam.engine.trainer import Trainer
+    from bigclam.io import load_graph
+
+    g = load_graph(os.path.join(_REF_DATA, "facebook_combined.txt"))
+    assert g.num_nodes == 4039 and g.num_edges == 88234
+    cfg = BigClamConfig(k=25, device="cpu", max_sweeps=6, seed=2)
+    tr = Trainer(g, cfg, rank=0, world_size=1, device=torch.device("cpu"))
+    res = tr.fit(init="random")
+    assert np.isfinite(res.llh)
+    assert res.llh > res.llh_history[0]
