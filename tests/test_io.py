@@ -160,3 +160,41 @@ def test_fit_facebook_k25_baseline_config():
     res = tr.fit(init="random")
     assert np.isfinite(res.llh)
     assert res.llh > res.llh_history[0]
+
+
+def test_build_graph_properties_random():
+    """Property test over random edge lists: symmetric CSR, sorted rows,
+    no self-loops/duplicates, degree sum == 2E."""
+    from hypothesis import given, settings
+    from hypothesis import strategies as st
+
+    from bigclam.io.edgelist import build_graph
+
+    @settings(max_examples=60, deadline=None)
+    @given(
+        st.lists(
+            st.tuples(st.integers(0, 40), st.integers(0, 40)),
+            min_size=1,
+            max_size=200,
+        )
+    )
+    def check(pairs):
+        import numpy as np
+
+        edges = np.array(pairs, dtype=np.int64)
+        g = build_graph(edges)
+        und = {tuple(sorted(p)) for p in pairs if p[0] != p[1]}
+        if not und:
+            return
+        assert g.num_edges == len(und)
+        assert len(g.indices) == 2 * len(und)
+        indptr, idx = g.indptr, g.indices
+        for u in range(g.num_nodes):
+            row = idx[indptr[u]:indptr[u + 1]]
+            assert np.all(np.diff(row) > 0), "sorted, no duplicates"
+            assert u not in row, "no self-loops"
+            for v in row:  # symmetry
+                vr = idx[indptr[v]:indptr[v + 1]]
+                assert u in vr
+
+    check()
