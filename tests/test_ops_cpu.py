@@ -125,3 +125,30 @@ def test_mfma_dispatch_prefix_invariant(monkeypatch):
             assert np.all(deg[o[:n_hi]] >= 8)
             if n_hi < len(o):
                 assert np.all(deg[o[n_hi:]] < 8)
+
+
+def test_reference_ops_extreme_values_finite():
+    """Numerical extremes: F rows at the clamp bound (1000) make
+    x = Fu.Fv overflow exp(-x) to 0 -> p clamps to min_p; all outputs
+    stay finite and llh uses the clamped p (no -inf)."""
+    import numpy as np
+    import torch
+
+    from bigclam.config import BigClamConfig
+    from bigclam.io import rmat_graph
+    from bigclam.ops import reference as ref_ops
+
+    g = rmat_graph(6, 4.0, seed=8)
+    cfg = BigClamConfig(k=8, device="cpu")
+    F = torch.full((g.num_nodes, 8), 1000.0)
+    sumF = F.sum(0)
+    indptr = torch.from_numpy(g.indptr)
+    indices = torch.from_numpy(g.indices.astype("int64")).int()
+    grad, llh = ref_ops.edge_grad_llh(F, indptr, indices, sumF, cfg)
+    assert torch.isfinite(grad).all()
+    assert torch.isfinite(llh).all()
+    best = ref_ops.linesearch(F, indptr, indices, sumF, grad, llh, cfg)
+    assert torch.isfinite(best).all()
+    newF, _ = ref_ops.apply_step(F, grad, best, cfg)
+    assert torch.isfinite(newF).all()
+    assert (newF <= cfg.max_f).all() and (newF >= cfg.min_f).all()
