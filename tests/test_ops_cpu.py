@@ -152,3 +152,31 @@ def test_reference_ops_extreme_values_finite():
     newF, _ = ref_ops.apply_step(F, grad, best, cfg)
     assert torch.isfinite(newF).all()
     assert (newF <= cfg.max_f).all() and (newF >= cfg.min_f).all()
+
+
+def test_mfma_dispatch_defaults_by_k():
+    """The measured K-aware dispatch defaults (state.py): bf16 routes ALL
+    nodes to the MFMA kernel at kp <= 8192 and above 16384, none in the
+    spill regime between; fp32 defaults to the direct kernel."""
+    import torch
+
+    from bigclam.config import BigClamConfig
+    from bigclam.core.shard import make_shard
+    from bigclam.core.state import ShardState
+    from bigclam.io import rmat_graph
+
+    g = rmat_graph(7, 4.0, seed=6)
+    for dtype, k, expect_all in [
+        ("bf16", 5000, True),
+        ("bf16", 8500, False),
+        ("bf16", 17000, True),
+        ("fp32", 5000, False),
+    ]:
+        st = ShardState(
+            make_shard(g, 0, 1),
+            BigClamConfig(k=k, device="cpu", dtype=dtype),
+            device=torch.device("cpu"),
+        )
+        n = int(st.order.numel())
+        assert (st.n_mfma == n) if expect_all else (st.n_mfma == 0), (
+            dtype, k, st.n_mfma)
