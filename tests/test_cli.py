@@ -118,5 +118,13 @@ def test_cli_bench_delegates(capfd):
     ])
     out = capfd.readouterr().out.strip().splitlines()[-1]
     rec = _json.loads(out)
+    # the full driver contract (BASELINE metric + required fields)
     assert rec["metric"] == "edges/sec per grad iter"
     assert rec["steps"] == 1 and rec["n_gpus"] == 1
+    assert rec["higher_is_better"] is True
+    assert rec["scaling"] == "strong"
+    assert rec["unit"] == "edges/s"
+    assert rec["dtype"] in ("bf16", "fp32")
+    assert rec["data"] == "synthetic"
+    assert rec["value"] > 0 and rec["ms_per_step"] > 0
+    assert "parallelism" in rec["config"] and "k" in rec["config"]
