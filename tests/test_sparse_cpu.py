@@ -1,0 +1,78 @@
+"""Vectorized sparse CPU sweep == dense torch reference (full sweep)."""
+import numpy as np
+import pytest
+import scipy.sparse as sp
+import torch
+
+from bigclam.config import BigClamConfig
+from bigclam.engine.trainer import Trainer
+from bigclam.io import rmat_graph
+from bigclam.ops import reference as ref_ops
+from bigclam.ops.sparse_cpu import sparse_sweep
+
+
+def _partially_sparse_F(g, k=48, sweeps=12):
+    cfg = BigClamConfig(k=k, device="cpu", max_sweeps=sweeps, seed=9)
+    tr = Trainer(g, cfg, rank=0, world_size=1, device=torch.device("cpu"))
+    tr.init_F("seed")
+    carry, _ = tr.prologue()
+    for _ in range(sweeps):
+        carry, _, _ = tr.pipelined_sweep(carry)
+    return cfg, tr.state.F_local_k.float().numpy().copy()
+
+
+def test_sparse_sweep_matches_dense():
+    g = rmat_graph(8, 5.0, seed=13)
+    cfg, F = _partially_sparse_F(g)
+    Ft = torch.from_numpy(F)
+    sumF = Ft.sum(0)
+    indptr = torch.from_numpy(g.indptr)
+    indices = torch.from_numpy(g.indices.astype(np.int64)).int()
+    grad, llh = ref_ops.edge_grad_llh(Ft, indptr, indices, sumF, cfg)
+    best_d = ref_ops.linesearch(Ft, indptr, indices, sumF, grad, llh, cfg)
+    newF_d, _ = ref_ops.apply_step(Ft, grad, best_d, cfg)
+
+    Fs = sp.csr_matrix(F)
+    newF_s, best_s, llh_s = sparse_sweep(
+        Fs, g.indptr, g.indices, sumF.numpy(), cfg
+    )
+
+    np.testing.assert_allclose(llh_s, llh.numpy(), rtol=1e-5, atol=1e-4)
+    bd = best_d.numpy()
+    agree = best_s == bd
+    # disagreements only below the beta^5 noise rung (see
+    # tests/test_sparse_proto.py for the rationale)
+    noise = cfg.beta ** 5
+    bad = ~agree & ((best_s > noise) | (bd > noise))
+    assert not bad.any(), np.flatnonzero(bad)[:10]
+    # at this partially-converged state most nodes sit at the threshold
+    # on the deep rungs, so exact agreement is only ~50% — every single
+    # disagreement is below the noise rung (asserted above)
+    assert agree.mean() > 0.3
+    # committed rows equal where the picks agree
+    dense_new = newF_s.toarray()
+    rows = np.flatnonzero(agree)
+    np.testing.assert_allclose(
+        dense_new[rows], newF_d.numpy()[rows], rtol=1e-4, atol=2e-4
+    )
+
+
+def test_sparse_sweep_trajectory():
+    """Several consecutive sparse sweeps track the dense engine's LLH."""
+    g = rmat_graph(8, 5.0, seed=14)
+    cfg, F = _partially_sparse_F(g, sweeps=8)
+    Fs = sp.csr_matrix(F)
+    Fd = torch.from_numpy(F.copy())
+    indptr = torch.from_numpy(g.indptr)
+    indices = torch.from_numpy(g.indices.astype(np.int64)).int()
+    for _ in range(4):
+        # dense step
+        sumF_d = Fd.sum(0)
+        grad, llh_d = ref_ops.edge_grad_llh(Fd, indptr, indices, sumF_d, cfg)
+        bd = ref_ops.linesearch(Fd, indptr, indices, sumF_d, grad, llh_d, cfg)
+        Fd, _ = ref_ops.apply_step(Fd, grad, bd, cfg)
+        # sparse step
+        sumF_s = np.asarray(Fs.sum(axis=0)).ravel()
+        Fs, bs, llh_s = sparse_sweep(Fs, g.indptr, g.indices, sumF_s, cfg)
+        rel = abs(llh_s.sum() - float(llh_d.sum())) / abs(float(llh_d.sum()))
+        assert rel < 1e-4, rel
