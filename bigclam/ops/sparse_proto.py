@@ -14,9 +14,10 @@ All clamps give exact zeros, so the sets are exact, not thresholded —
 tests/test_sparse_proto.py asserts both the set-containment claims and
 value equality against the dense torch reference (ops/reference.py).
 
-This is the algorithmic contract for the next round's GPU path (rows as
-index/value segments, the candidate scoring as MFMA tiles over the
-gathered fv[:, S_u] panel); kept NumPy-simple here on purpose.
+This is the readable algorithmic contract; ops/sparse_cpu.py is the
+vectorized full-sweep realization (same math, scipy CSR), and the next
+round's GPU path follows the same shape (rows as index/value segments,
+candidate scoring as MFMA tiles over the gathered fv[:, S_u] panel).
 """
 from __future__ import annotations
 

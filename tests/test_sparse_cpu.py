@@ -1,6 +1,5 @@
 """Vectorized sparse CPU sweep == dense torch reference (full sweep)."""
 import numpy as np
-import pytest
 import scipy.sparse as sp
 import torch
 
