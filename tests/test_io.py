@@ -198,3 +198,28 @@ def test_build_graph_properties_random():
                 assert u in vr
 
     check()
+
+
+def test_shaped_graph_cross_process_deterministic():
+    """bench.py relies on every torchrun rank generating the IDENTICAL
+    graph from the seed; verify determinism across fresh processes."""
+    import hashlib
+    import subprocess
+    import sys
+
+    code = (
+        "from bigclam.io import shaped_graph\n"
+        "import hashlib\n"
+        "g = shaped_graph(3000, 9000, locality=0.7, seed=42)\n"
+        "h = hashlib.sha256()\n"
+        "h.update(g.indptr.tobytes()); h.update(g.indices.tobytes())\n"
+        "print(h.hexdigest())\n"
+    )
+    outs = {
+        subprocess.run(
+            [sys.executable, "-c", code], capture_output=True, text=True,
+            check=True,
+        ).stdout.strip()
+        for _ in range(2)
+    }
+    assert len(outs) == 1, outs
