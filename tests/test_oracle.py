@@ -121,3 +121,52 @@ def test_conductance_guards(tiny_graph):
         # reference formula: cut/min(volS, volT) — can exceed 1 when the
         # complement volume volT is small; guards give 0.0 / 1.0 exactly.
         assert c >= 0.0 and np.isfinite(c)
+
+
+def test_reference_matches_oracle_random_graphs():
+    """Hypothesis fuzz: the torch fp32 reference matches the float64
+    NumPy oracle on random tiny graphs and random F (SURVEY §4 unit net,
+    broadened beyond the fixed fixtures)."""
+    from hypothesis import given, settings
+    from hypothesis import strategies as st
+
+    import numpy as np
+    import torch
+
+    from bigclam.config import BigClamConfig
+    from bigclam.io.edgelist import build_graph
+    from bigclam.ops import reference as ref_ops
+    import oracle
+
+    @settings(max_examples=25, deadline=None)
+    @given(
+        st.lists(
+            st.tuples(st.integers(0, 12), st.integers(0, 12)),
+            min_size=3, max_size=40,
+        ).filter(lambda ps: any(a != b for a, b in ps)),
+        st.integers(0, 2 ** 31 - 1),
+    )
+    def check(pairs, seed):
+        g = build_graph(np.array(pairs, dtype=np.int64))
+        k = 5
+        rng = np.random.default_rng(seed)
+        F = (rng.random((g.num_nodes, k)) * 0.6).astype(np.float32)
+        cfg = BigClamConfig(k=k, device="cpu")
+        Ft = torch.from_numpy(F)
+        sumF = Ft.sum(0)
+        grad, llh = ref_ops.edge_grad_llh(
+            Ft, torch.from_numpy(g.indptr),
+            torch.from_numpy(g.indices.astype(np.int64)).int(), sumF, cfg,
+        )
+        F64 = F.astype(np.float64)
+        s64 = F64.sum(0)
+        for u in range(g.num_nodes):
+            og, ol = oracle.node_grad_llh(
+                F64, s64, g.indptr, g.indices, u
+            )
+            np.testing.assert_allclose(
+                grad[u].numpy(), og, rtol=5e-4, atol=5e-4
+            )
+            assert abs(llh[u].item() - ol) < 1e-4 * max(1.0, abs(ol))
+
+    check()
