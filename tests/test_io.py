@@ -155,7 +155,7 @@ def test_fit_facebook_k25_baseline_config():
 
     g = load_graph(os.path.join(_REF_DATA, "facebook_combined.txt"))
     assert g.num_nodes == 4039 and g.num_edges == 88234
-    cfg = BigClamConfig(k=25, device="cpu", max_sweeps=6, seed=2)
+    cfg = BigClamConfig(k=25, device="cpu", max_sweeps=4, seed=2)
     tr = Trainer(g, cfg, rank=0, world_size=1, device=torch.device("cpu"))
     res = tr.fit(init="random")
     assert np.isfinite(res.llh)
