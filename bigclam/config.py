@@ -68,6 +68,15 @@ class BigClamConfig:
     #: output path for community assignments
     out: Optional[str] = None
 
+    def __post_init__(self):
+        if not 0 <= self.ls_steps <= 15:
+            raise ValueError(
+                "ls_steps must be in [0, 15] (ladder of at most 16 "
+                f"candidates, the GPU kernels' MAX_LS); got {self.ls_steps}"
+            )
+        if self.dtype not in ("fp32", "bf16"):
+            raise ValueError(f"dtype must be fp32 or bf16, got {self.dtype!r}")
+
     def ladder(self) -> list:
         """The Armijo candidate-step ladder, largest first.
 
