@@ -5,6 +5,10 @@ import torch
 
 def pytest_configure(config):
     config.addinivalue_line("markers", "gpu: requires an MI355X GPU (run via gpurun)")
+    # safety net for the driver's -x -q runs: no single test may hang
+    # the round's build/test tier (pytest-timeout is in the image)
+    if config.getoption("--timeout", None) in (None, 0):
+        config.option.timeout = 600
 
 
 def pytest_collection_modifyitems(config, items):
