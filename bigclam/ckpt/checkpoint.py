@@ -71,11 +71,38 @@ def load_full_F(dirpath: str) -> np.ndarray:
     return np.concatenate(parts, axis=0)
 
 
-def resume(dirpath: str, trainer) -> int:
-    """Load checkpoint rows into trainer's shard (any world size).
-    Returns the sweep number to continue from."""
+def load_F_slice(dirpath: str, start: int, stop: int) -> np.ndarray:
+    """Global rows [start, stop) as fp32, reading only the shard files that
+    overlap (memory-mapped) — never materializes the full N×K matrix."""
     meta = load_meta(dirpath)
-    full = load_full_F(dirpath)  # simple + correct; re-shard by slicing
+    bounds = meta["bounds"]
+    out = np.empty((stop - start, meta["k"]), dtype=np.float32)
+    for r in range(meta["world_size"]):
+        rs, re_ = int(bounds[r]), int(bounds[r + 1])
+        lo, hi = max(start, rs), min(stop, re_)
+        if lo >= hi:
+            continue
+        arr = np.load(
+            os.path.join(dirpath, f"F_rank{r}.npy"), mmap_mode="r"
+        )[lo - rs : hi - rs]
+        if meta["dtype"] == "bf16":
+            arr = (
+                torch.from_numpy(np.ascontiguousarray(arr))
+                .view(torch.bfloat16)
+                .float()
+                .numpy()
+            )
+        out[lo - start : hi - start] = arr
+    return out
+
+
+def resume(dirpath: str, trainer) -> tuple:
+    """Load checkpoint rows into trainer's shard (any world size; each rank
+    reads only its own slice).  Returns ``(sweep, llh)`` so the training
+    loop can continue with the saved objective as ``llh_old`` (same
+    convergence behavior as an uninterrupted run)."""
+    meta = load_meta(dirpath)
     s = trainer.shard
-    trainer.state.set_local_F(torch.from_numpy(full[s.start : s.stop]))
-    return int(meta["sweep"])
+    sl = load_F_slice(dirpath, s.start, s.stop)
+    trainer.state.set_local_F(torch.from_numpy(sl))
+    return int(meta["sweep"]), float(meta["llh"])

@@ -122,7 +122,15 @@ def main(argv=None):
 
         sys.argv = ["bench.py"] + list(argv[1:])
         here = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
-        runpy.run_path(os.path.join(here, "bench.py"), run_name="__main__")
+        bench = os.path.join(here, "bench.py")
+        if not os.path.exists(bench):  # installed wheel: no repo-root bench.py
+            print(
+                "bigclam bench requires a repo checkout (bench.py at the "
+                "repo root); run it from the source tree",
+                file=sys.stderr,
+            )
+            return 2
+        runpy.run_path(bench, run_name="__main__")
         return 0
 
     args = ap.parse_args(argv)
@@ -152,9 +160,9 @@ def main(argv=None):
     if getattr(args, "resume", None):
         from .ckpt.checkpoint import resume as ckpt_resume
 
-        sweep0 = ckpt_resume(args.resume, tr)
-        metrics.log({"note": "resumed", "from_sweep": sweep0})
-        res = tr.fit(skip_init=True)
+        sweep0, llh0 = ckpt_resume(args.resume, tr)
+        metrics.log({"note": "resumed", "from_sweep": sweep0, "llh": llh0})
+        res = tr.fit(skip_init=True, llh_old=llh0, sweep0=sweep0)
     else:
         res = tr.fit(init=args.init)
     F = tr.gather_F()

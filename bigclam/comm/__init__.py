@@ -1,5 +1,6 @@
 from .dist import (
     all_reduce_,
+    backend,
     all_to_all,
     barrier,
     get_rank,
@@ -10,6 +11,7 @@ from .dist import (
 
 __all__ = [
     "all_reduce_",
+    "backend",
     "all_to_all",
     "barrier",
     "get_rank",

@@ -43,13 +43,28 @@ def init_distributed(backend: str = None, timeout_s: int = 600) -> int:
     if backend is None:
         backend = "nccl" if torch.cuda.is_available() else "gloo"
     os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
-    os.environ.setdefault("MASTER_PORT", "29571")
+    if "MASTER_PORT" not in os.environ:
+        # no rendezvous info (manual launch without torchrun): pick a port
+        # deterministically from job-identifying state every rank shares —
+        # cwd + world size — so concurrent jobs in different directories
+        # don't collide on a fixed hardcoded port.
+        import hashlib
+
+        h = hashlib.sha1(
+            f"{os.getcwd()}:{world}".encode()
+        ).digest()
+        os.environ["MASTER_PORT"] = str(20000 + int.from_bytes(h[:2], "big") % 20000)
     dist.init_process_group(
         backend=backend, timeout=datetime.timedelta(seconds=timeout_s)
     )
     if backend == "nccl":
         torch.cuda.set_device(int(os.environ.get("LOCAL_RANK", "0")))
     return dist.get_rank()
+
+
+def backend() -> str:
+    """Active process-group backend name ('' when not distributed)."""
+    return str(dist.get_backend()) if is_distributed() else ""
 
 
 def all_reduce_(t: torch.Tensor) -> torch.Tensor:

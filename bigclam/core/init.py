@@ -145,11 +145,27 @@ def seed_init_local_F(
             F[s - start, c] = 1.0
     n_pad = k - len(seeds)
     if n_pad > 0:
-        rng = np.random.default_rng(rng_seed)
-        pad = rng.integers(
-            0, 2, size=(graph.num_nodes, n_pad), dtype=np.int8
-        ).astype(np.float32)
-        F[:, len(seeds) :] = pad[start:stop]
+        # Bernoulli(0.5) pad columns, generated per global row-block with a
+        # block-keyed RNG so every rank's slice is identical regardless of
+        # sharding — and only the local rows are ever materialized (the
+        # full-[N, n_pad] array this replaces is ~150 GB/rank at the
+        # N=15M, K=10000 config).
+        block = 8192
+        b0 = start // block
+        b1 = (stop - 1) // block if stop > start else b0
+        for b in range(b0, b1 + 1):
+            rng = np.random.default_rng(((rng_seed + 1) << 20) + b)
+            rows = rng.integers(
+                0,
+                2,
+                size=(min(block, graph.num_nodes - b * block), n_pad),
+                dtype=np.int8,
+            )
+            lo = max(start, b * block)
+            hi = min(stop, b * block + rows.shape[0])
+            F[lo - start : hi - start, len(seeds) :] = rows[
+                lo - b * block : hi - b * block
+            ]
     return F
 
 

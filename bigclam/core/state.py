@@ -28,6 +28,21 @@ def _hip_ops():
     return hip_ops
 
 
+class _CastOnWait:
+    """Work-handle wrapper that copies the fp32 staging buffer into the
+    bf16 halo section after the collective completes (gloo bf16 path)."""
+
+    def __init__(self, work, src32: torch.Tensor, dst: torch.Tensor):
+        self._work = work
+        self._src32 = src32
+        self._dst = dst
+
+    def wait(self):
+        if self._work is not None:
+            self._work.wait()
+        self._dst.copy_(self._src32)
+
+
 class ShardState:
     def __init__(
         self,
@@ -64,13 +79,13 @@ class ShardState:
         # all_to_all is in flight (halo fractions are large on power-law
         # graphs: Email-Enron ws=8 halo = 1.95x N).
         if shard.world_size > 1 and shard.n_halo > 0:
-            row_has_halo = np.zeros(shard.n_local, dtype=bool)
             remote = shard.indices >= shard.n_local
-            if remote.any():
-                seg = np.add.reduceat(
-                    remote, shard.indptr[:-1].clip(max=max(len(remote) - 1, 0))
-                )
-                row_has_halo = (seg > 0) & (np.diff(shard.indptr) > 0)
+            # exact segment sums even for empty rows: cumulative count of
+            # remote neighbors, differenced at row boundaries (reduceat
+            # with clipped indptr misattributes the last edge when
+            # trailing degree-0 rows exist)
+            cs = np.concatenate([[0], np.cumsum(remote)])
+            row_has_halo = (cs[shard.indptr[1:]] - cs[shard.indptr[:-1]]) > 0
             bnd = order_np[row_has_halo[order_np]]
             interior = order_np[~row_has_halo[order_np]]
             self.order_interior = torch.from_numpy(
@@ -196,8 +211,22 @@ class ShardState:
         if self.shard.world_size == 1 or self.shard.n_halo == 0:
             return None
         send = self.F_local.index_select(0, self.send_idx)
-        self._halo_send = send  # keep alive while the collective is in flight
         recv = self.F[self.n_local :]
+        if self.storage_dtype == torch.bfloat16 and comm.backend() == "gloo":
+            # gloo has no bf16 collectives: route the exchange through fp32
+            # (RCCL takes bf16 natively — no cast on the GPU path)
+            send = send.float()
+            recv32 = torch.empty_like(recv, dtype=torch.float32)
+            self._halo_send = send
+            work = comm.all_to_all(
+                recv32, send, self.recv_splits, self.send_splits,
+                async_op=async_op,
+            )
+            if async_op:
+                return _CastOnWait(work, recv32, recv)
+            recv.copy_(recv32)
+            return None
+        self._halo_send = send  # keep alive while the collective is in flight
         return comm.all_to_all(
             recv, send, self.recv_splits, self.send_splits, async_op=async_op
         )

@@ -164,12 +164,21 @@ class Trainer:
         comm.all_reduce_(total)
         return (grad, steps_next), float(total.item()), steps
 
-    def fit(self, init: str = "seed", skip_init: bool = False) -> FitResult:
+    def fit(
+        self,
+        init: str = "seed",
+        skip_init: bool = False,
+        llh_old: float = 0.0,
+        sweep0: int = 0,
+    ) -> FitResult:
+        """Run the convergence loop.  ``llh_old``/``sweep0`` continue a
+        resumed fit with the checkpointed objective and sweep numbering —
+        the resumed trajectory is identical to an uninterrupted run
+        (tests/test_ckpt.py)."""
         cfg = self.cfg
         if not skip_init:
             self.init_F(init)
         res = FitResult()
-        llh_old = 0.0
         carry, llh0 = self.prologue()
         # per-sweep collective volume (static per topology — SURVEY §5
         # observability: per-collective bytes): halo p2p both directions,
@@ -190,7 +199,7 @@ class Trainer:
                 },
             }
         )
-        for i in range(cfg.max_sweeps):
+        for i in range(sweep0, sweep0 + cfg.max_sweeps):
             timer = PhaseTimer(sync=True)
             timer.start("sweep")
             carry, llh, steps = self.pipelined_sweep(carry)
@@ -250,22 +259,27 @@ class Trainer:
 
     # -------------------------------------------------------------- gather F
     def gather_F(self) -> Optional[torch.Tensor]:
-        """Gather the full F to rank 0 (CPU) for extraction/output."""
+        """Gather the full F to rank 0 (CPU) for extraction/output.
+
+        Point-to-point sends to rank 0 only — O(N·K) traffic and host
+        memory on rank 0, nothing on the other ranks (the round-1
+        broadcast-to-all was O(ws·N·K) everywhere).  Prefer
+        ``engine.extract.extract_communities_sharded`` where possible —
+        it never materializes N×K at all.
+        """
         local = self.state.F_local_k.contiguous().float()
         if self.world_size == 1:
             return local.cpu()
         import torch.distributed as dist
 
         dev = local.device  # nccl needs device tensors, gloo wants CPU
-        gathered: List[torch.Tensor] = []
-        for r in range(self.world_size):
+        if self.rank != 0:
+            dist.send(local, dst=0)
+            return None
+        gathered: List[torch.Tensor] = [local.cpu()]
+        for r in range(1, self.world_size):
             n_r = int(self.bounds[r + 1] - self.bounds[r])
-            buf = (
-                local.contiguous()
-                if r == self.rank
-                else torch.empty(n_r, self.cfg.k, dtype=torch.float32, device=dev)
-            )
-            dist.broadcast(buf, src=r)
-            if self.rank == 0:
-                gathered.append(buf.cpu())
-        return torch.cat(gathered, dim=0) if self.rank == 0 else None
+            buf = torch.empty(n_r, self.cfg.k, dtype=torch.float32, device=dev)
+            dist.recv(buf, src=r)
+            gathered.append(buf.cpu())
+        return torch.cat(gathered, dim=0)

@@ -75,6 +75,121 @@ def test_sharded_equals_single(world_size, tmp_path):
     np.testing.assert_allclose(F_w, F1, rtol=1e-5, atol=1e-6)
 
 
+def _single_fit(n_sweeps, dtype):
+    g = _graph()
+    cfg = BigClamConfig(
+        k=3, device="cpu", seed=5, dtype=dtype, max_sweeps=n_sweeps, tol=0.0
+    )
+    tr = Trainer(g, cfg, rank=0, world_size=1, device=torch.device("cpu"))
+    res = tr.fit(init="random")
+    return res.llh_history, tr.state.F_local_k.float().numpy().copy()
+
+
+def _worker_fit(rank, world_size, port, out_dir, n_sweeps, dtype):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    os.environ["RANK"] = str(rank)
+    os.environ["WORLD_SIZE"] = str(world_size)
+    import torch.distributed as dist
+
+    dist.init_process_group("gloo", rank=rank, world_size=world_size)
+    try:
+        g = _graph()
+        cfg = BigClamConfig(
+            k=3, device="cpu", seed=5, dtype=dtype, max_sweeps=n_sweeps,
+            tol=0.0,
+        )
+        tr = Trainer(g, cfg, device=torch.device("cpu"))
+        res = tr.fit(init="random")
+        F = tr.gather_F()
+        if rank == 0:
+            np.save(os.path.join(out_dir, "F.npy"), F.numpy())
+            with open(os.path.join(out_dir, "llh.json"), "w") as f:
+                json.dump(res.llh_history, f)
+    finally:
+        dist.destroy_process_group()
+
+
+@pytest.mark.parametrize(
+    "world_size,dtype",
+    [(8, "fp32"), (2, "bf16"), (3, "bf16")],
+    ids=["ws8-fp32", "ws2-bf16", "ws3-bf16"],
+)
+def test_full_fit_sharded_equals_single(world_size, dtype, tmp_path):
+    """The full pipelined fit (prologue + async halo overlap + fused path)
+    at ws=8, and the bf16 halo exchange over gloo (fp32-staged cast),
+    match the single-shard run — VERDICT r01 next-round #4."""
+    n_sweeps = 3
+    llh1, F1 = _single_fit(n_sweeps, dtype)
+    port = 29650 + world_size + (100 if dtype == "bf16" else 0)
+    mp.spawn(
+        _worker_fit,
+        args=(world_size, port, str(tmp_path), n_sweeps, dtype),
+        nprocs=world_size,
+        join=True,
+    )
+    llh_w = json.load(open(tmp_path / "llh.json"))
+    F_w = np.load(tmp_path / "F.npy")
+    for a, b in zip(llh1, llh_w):
+        assert abs(a - b) < 1e-6 * max(1.0, abs(a)), (llh1, llh_w)
+    np.testing.assert_allclose(F_w, F1, rtol=1e-5, atol=1e-6)
+
+
+def test_torchrun_bench_cpu(tmp_path):
+    """The exact harness path the scaling driver uses — torchrun with 2
+    ranks over gloo on CPU — runs bench.py end to end and emits the
+    contract JSON line."""
+    import subprocess
+
+    root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    cmd = [
+        sys.executable, "-m", "torch.distributed.run",
+        "--nnodes=1", "--nproc-per-node", "2",
+        "--master-addr", "127.0.0.1", "--master-port", "29777",
+        os.path.join(root, "bench.py"),
+        "--gpus", "2", "--steps", "2", "--warmup", "1",
+        "--k", "8", "--nodes", "300", "--edges", "900", "--dtype", "fp32",
+    ]
+    out = subprocess.run(
+        cmd, capture_output=True, text=True, timeout=300, cwd=root
+    )
+    assert out.returncode == 0, out.stderr[-2000:]
+    line = [l for l in out.stdout.splitlines() if l.startswith("{")][-1]
+    rec = json.loads(line)
+    assert rec["n_gpus"] == 2
+    assert rec["value"] > 0 and np.isfinite(rec["ms_per_step"])
+
+
+def test_boundary_split_trailing_empty_rows():
+    """Interior/boundary split stays exact when trailing degree-0 local
+    rows exist (the reduceat-with-clipped-indptr version
+    classified a boundary row as interior, racing the async halo)."""
+    from bigclam.core.shard import GraphShard, HaloPlan
+    from bigclam.core.state import ShardState
+
+    shard = GraphShard(
+        rank=0,
+        world_size=2,
+        start=0,
+        stop=3,
+        n_total=6,
+        indptr=np.array([0, 1, 3, 3], dtype=np.int64),
+        indices=np.array([0, 1, 5], dtype=np.int32),
+        halo_globals=np.array([5], dtype=np.int64),
+        plan=HaloPlan(
+            send_idx=[np.empty(0, dtype=np.int64), np.empty(0, dtype=np.int64)],
+            recv_counts=[0, 1],
+        ),
+        num_edges_global=2,
+    )
+    st = ShardState(shard, BigClamConfig(k=3, device="cpu"),
+                    device=torch.device("cpu"))
+    # row 1's last neighbor is halo row 5 (local index 3 >= n_local=3)
+    assert 1 in st.order_boundary.tolist()
+    assert set(st.order_interior.tolist()) | set(st.order_boundary.tolist()) == {0, 1, 2}
+    assert set(st.order_interior.tolist()) & set(st.order_boundary.tolist()) == set()
+
+
 def test_halo_plan_consistency():
     """Every rank's send list matches the peers' recv expectations."""
     g = _graph()
