@@ -175,3 +175,23 @@ def conductance(indptr, indices, u, total_degree):
     if vol_t == 0:
         return 1.0
     return cut / min(vol_s, vol_t)
+
+
+def armijo_margin_f64(graph, F, grad_u, u, s, cfg):
+    """fp64 Armijo acceptance margin of step ``s`` for node ``u``:
+    llh_u(clamp(F_u + s·grad_u)) − llh_u(F_u) − α·s·‖grad_u‖².
+    A margin within fp32 noise of zero means accept/reject is a tie and
+    fp32 implementations may legitimately disagree on the pick."""
+    Fd = np.asarray(F, dtype=np.float64)
+    sumF = Fd.sum(0)
+    gu = np.asarray(grad_u, dtype=np.float64)
+    nbrs = graph.indices[graph.indptr[u] : graph.indptr[u + 1]]
+
+    def llh_u(fu):
+        x = Fd[nbrs] @ fu
+        p = np.clip(np.exp(-x), cfg.min_p, cfg.max_p)
+        sf = sumF - Fd[u] + fu
+        return float(np.sum(np.log(1 - p) + x) - fu @ sf + fu @ fu)
+
+    fu2 = np.clip(Fd[u] + s * gu, cfg.min_f, cfg.max_f)
+    return llh_u(fu2) - llh_u(Fd[u]) - cfg.alpha * s * (gu @ gu)

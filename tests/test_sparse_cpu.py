@@ -45,23 +45,12 @@ def test_sparse_sweep_matches_dense():
     # within fp32 noise of zero (observed: +1e-6 on |llh|~23)
     noise = cfg.beta ** 5
     bad = ~agree & ((best_s > noise) | (bd > noise))
-    Fd64 = F.astype(np.float64)
-    sumF64 = Fd64.sum(0)
-    g64 = grad.numpy().astype(np.float64)
+    from .oracle import armijo_margin_f64
+
     for u in np.flatnonzero(bad):
         s = max(best_s[u], bd[u])
-        nbrs = g.indices[g.indptr[u] : g.indptr[u + 1]]
-
-        def llh_u(fu):
-            x = Fd64[nbrs] @ fu
-            p = np.clip(np.exp(-x), cfg.min_p, cfg.max_p)
-            sf = sumF64 - Fd64[u] + fu
-            return float(np.sum(np.log(1 - p) + x) - fu @ sf + fu @ fu)
-
-        base = llh_u(Fd64[u])
-        fu2 = np.clip(Fd64[u] + s * g64[u], cfg.min_f, cfg.max_f)
-        margin = llh_u(fu2) - base - cfg.alpha * s * (g64[u] @ g64[u])
-        assert abs(margin) < 5e-5 * max(1.0, abs(base)), (u, s, margin)
+        margin = armijo_margin_f64(g, F, grad[u].numpy(), int(u), s, cfg)
+        assert abs(margin) < 5e-5 * max(1.0, abs(float(llh[u]))), (u, s, margin)
     # at this partially-converged state most nodes sit at the threshold
     # on the deep rungs, so exact agreement is only ~50% — every single
     # disagreement is below the noise rung (asserted above)
