@@ -45,7 +45,7 @@ def test_sparse_sweep_matches_dense():
     # within fp32 noise of zero (observed: +1e-6 on |llh|~23)
     noise = cfg.beta ** 5
     bad = ~agree & ((best_s > noise) | (bd > noise))
-    from .oracle import armijo_margin_f64
+    from oracle import armijo_margin_f64
 
     for u in np.flatnonzero(bad):
         s = max(best_s[u], bd[u])

@@ -74,7 +74,7 @@ def test_sparse_sweep_matches_dense_reference():
             if max(b, rb) > noise_rung:
                 # above the noise rung a disagreement must be a genuine
                 # fp32 Armijo tie: fp64 margin within fp32 noise of zero
-                from .oracle import armijo_margin_f64
+                from oracle import armijo_margin_f64
 
                 m = armijo_margin_f64(g, F, grad[u].numpy(), int(u),
                                       max(b, rb), cfg)
