@@ -165,15 +165,22 @@ def main(argv=None):
         res = tr.fit(skip_init=True, llh_old=llh0, sweep0=sweep0)
     else:
         res = tr.fit(init=args.init)
-    F = tr.gather_F()
     if cfg.checkpoint_dir:  # every rank writes its own shard file
         from .ckpt.checkpoint import save_shard_checkpoint
 
         save_shard_checkpoint(cfg.checkpoint_dir, tr, res.sweeps, res.llh)
+    if cfg.out:
+        # sharded K7: per-rank device-side threshold, compacted lists to
+        # rank 0 — no N×K gather
+        from .engine.extract import (
+            extract_communities_sharded,
+            write_membership_pairs,
+        )
+
+        pairs = extract_communities_sharded(tr)
+        if rank == 0:
+            write_membership_pairs(cfg.out, pairs[0], pairs[1], g.raw_ids)
     if rank == 0:
-        if cfg.out:
-            members = extract_communities(F, g.num_edges)
-            write_communities(cfg.out, members, g.raw_ids)
         print(
             json.dumps(
                 {

@@ -2187,6 +2187,98 @@ extern "C" __global__ void __launch_bounds__(BLOCK) k5_conductance(
   }
 }
 
+// ------------------------------------------------------------------- K7
+//
+// Community extraction (codes/Bigclamv2.scala:223-230): node u belongs to
+// community c iff F[u,c] >= delta; a row whose max is below delta (but
+// nonzero) falls back to its argmax columns (ties included); all-zero rows
+// get no membership (documented deviation, engine/extract.py).  Runs as
+// two deterministic passes with NO atomics: the COUNT pass writes per-row
+// membership counts (host prefix-sums them into offsets), the FILL pass
+// writes row u's community ids ascending at offsets[u].  Each thread owns
+// a CONTIGUOUS column chunk, so the block-exclusive-scan write offsets
+// preserve ascending-c order.  One block per row; HBM-bound (2 row reads).
+
+template <bool BF16, bool FILL>
+__global__ void __launch_bounds__(BLOCK) k7_membership(
+    const void* __restrict__ Fv, int n, int K, int ldF, float delta,
+    const long long* __restrict__ offsets, int* __restrict__ counts,
+    int* __restrict__ comms) {
+  __shared__ float redf[NWAVE];
+  __shared__ int scan[BLOCK];
+  const int u = blockIdx.x;
+  if (u >= n) return;
+  const float* Ff =
+      BF16 ? nullptr : reinterpret_cast<const float*>(Fv) + (size_t)u * ldF;
+  const u32* Fb = BF16
+                      ? reinterpret_cast<const u32*>(Fv) + (size_t)u * (ldF / 2)
+                      : nullptr;
+  const int chunk = (K + BLOCK - 1) / BLOCK;
+  const int c0 = min((int)threadIdx.x * chunk, K);
+  const int c1 = min(c0 + chunk, K);
+  float mymax = 0.f;
+  float myabove = 0.f;
+  for (int c = c0; c < c1; ++c) {
+    const float f = BF16 ? ((c & 1) ? bf_hi(Fb[c >> 1]) : bf_lo(Fb[c >> 1]))
+                         : Ff[c];
+    mymax = fmaxf(mymax, f);
+    myabove += (f >= delta) ? 1.f : 0.f;
+  }
+  // block max
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int wid = threadIdx.x >> 6;
+  float m = mymax;
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1)
+    m = fmaxf(m, __shfl_xor(m, off, WAVE));
+  if (lane == 0) redf[wid] = m;
+  __syncthreads();
+  float fmax = redf[0];
+#pragma unroll
+  for (int w = 1; w < NWAVE; ++w) fmax = fmaxf(fmax, redf[w]);
+  __syncthreads();
+  const float total_above = block_allreduce_sum(myabove, redf);
+  // per-thread membership count under the row's mode
+  int cnt = 0;
+  if (total_above > 0.f) {
+    cnt = (int)myabove;
+  } else if (fmax > 0.f) {
+    for (int c = c0; c < c1; ++c) {
+      const float f = BF16 ? ((c & 1) ? bf_hi(Fb[c >> 1]) : bf_lo(Fb[c >> 1]))
+                           : Ff[c];
+      cnt += (f == fmax);
+    }
+  }
+  // block inclusive scan of the 256 per-thread counts (Hillis-Steele)
+  scan[threadIdx.x] = cnt;
+  __syncthreads();
+#pragma unroll
+  for (int off = 1; off < BLOCK; off <<= 1) {
+    const int v = (threadIdx.x >= off) ? scan[threadIdx.x - off] : 0;
+    __syncthreads();
+    scan[threadIdx.x] += v;
+    __syncthreads();
+  }
+  if (!FILL) {
+    if (threadIdx.x == BLOCK - 1) counts[u] = scan[BLOCK - 1];
+    return;
+  }
+  long long w = offsets[u] + (scan[threadIdx.x] - cnt);  // exclusive offset
+  if (total_above > 0.f) {
+    for (int c = c0; c < c1; ++c) {
+      const float f = BF16 ? ((c & 1) ? bf_hi(Fb[c >> 1]) : bf_lo(Fb[c >> 1]))
+                           : Ff[c];
+      if (f >= delta) comms[w++] = c;
+    }
+  } else if (fmax > 0.f) {
+    for (int c = c0; c < c1; ++c) {
+      const float f = BF16 ? ((c & 1) ? bf_hi(Fb[c >> 1]) : bf_lo(Fb[c >> 1]))
+                           : Ff[c];
+      if (f == fmax) comms[w++] = c;
+    }
+  }
+}
+
 // ----------------------------------------------------------- host launchers
 
 #include <cstdlib>
@@ -2529,6 +2621,34 @@ extern "C" void launch_kf_mfma_bf16(const void* F, const long long* indptr,
     KFMB_CASE(0);  // fu via LDS only (see kf_phase_a_bf16_lds)
   }
 #undef KFMB_CASE
+  HIP_CHECK(hipGetLastError());
+}
+
+extern "C" void launch_k7_count(const void* F, int bf16, int n, int K,
+                                int ldF, float delta, int* counts,
+                                hipStream_t stream) {
+  if (n == 0) return;
+  if (bf16) {
+    hipLaunchKernelGGL((k7_membership<true, false>), dim3(n), dim3(BLOCK), 0,
+                       stream, F, n, K, ldF, delta, nullptr, counts, nullptr);
+  } else {
+    hipLaunchKernelGGL((k7_membership<false, false>), dim3(n), dim3(BLOCK), 0,
+                       stream, F, n, K, ldF, delta, nullptr, counts, nullptr);
+  }
+  HIP_CHECK(hipGetLastError());
+}
+
+extern "C" void launch_k7_fill(const void* F, int bf16, int n, int K, int ldF,
+                               float delta, const long long* offsets,
+                               int* comms, hipStream_t stream) {
+  if (n == 0) return;
+  if (bf16) {
+    hipLaunchKernelGGL((k7_membership<true, true>), dim3(n), dim3(BLOCK), 0,
+                       stream, F, n, K, ldF, delta, offsets, nullptr, comms);
+  } else {
+    hipLaunchKernelGGL((k7_membership<false, true>), dim3(n), dim3(BLOCK), 0,
+                       stream, F, n, K, ldF, delta, offsets, nullptr, comms);
+  }
   HIP_CHECK(hipGetLastError());
 }
 

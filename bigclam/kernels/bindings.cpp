@@ -51,6 +51,10 @@ extern "C" void launch_kf_mfma_bf16(const void*, const long long*, const int*,
                                     const float*, float*, int, int, int,
                                     float, float, float, float, float,
                                     hipStream_t);
+extern "C" void launch_k7_count(const void*, int, int, int, int, float, int*,
+                                hipStream_t);
+extern "C" void launch_k7_fill(const void*, int, int, int, int, float,
+                               const long long*, int*, hipStream_t);
 extern "C" void launch_mfma_probe_bf16(const void*, const void*, float*,
                                        hipStream_t);
 extern "C" void launch_mfma_probe_f32(const float*, const float*, float*,
@@ -299,6 +303,37 @@ void mfma_probe(torch::Tensor A, torch::Tensor Bc, torch::Tensor D) {
   }
 }
 
+// K7 community extraction: two deterministic passes (count, then fill
+// after a host/torch prefix-sum) — see k7_membership in the .hip file.
+void extract_count(torch::Tensor F_local, int64_t k_true, double delta,
+                   torch::Tensor counts) {
+  CHECK_F(F_local);
+  CHECK_IN(counts, torch::kInt32);
+  const int n = (int)F_local.size(0);
+  const int ldF = (int)F_local.size(1);
+  TORCH_CHECK(counts.size(0) == n);
+  TORCH_CHECK(k_true >= 1 && k_true <= ldF, "bad k_true");
+  launch_k7_count(F_local.data_ptr(), is_bf16(F_local) ? 1 : 0, n,
+                  (int)k_true, ldF, (float)delta, counts.data_ptr<int>(),
+                  current_stream());
+}
+
+void extract_fill(torch::Tensor F_local, int64_t k_true, double delta,
+                  torch::Tensor offsets, torch::Tensor comms) {
+  CHECK_F(F_local);
+  CHECK_IN(offsets, torch::kInt64);
+  CHECK_IN(comms, torch::kInt32);
+  const int n = (int)F_local.size(0);
+  const int ldF = (int)F_local.size(1);
+  TORCH_CHECK(offsets.size(0) == n);
+  TORCH_CHECK(k_true >= 1 && k_true <= ldF, "bad k_true");
+  launch_k7_fill(F_local.data_ptr(), is_bf16(F_local) ? 1 : 0, n,
+                 (int)k_true, ldF, (float)delta,
+                 reinterpret_cast<const long long*>(
+                     offsets.data_ptr<int64_t>()),
+                 comms.data_ptr<int>(), current_stream());
+}
+
 void conductance(torch::Tensor indptr, torch::Tensor indices,
                  torch::Tensor cond, double total_degree) {
   CHECK_IN(indptr, torch::kInt64);
@@ -328,6 +363,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "(MFMA tiles for the high-degree prefix)");
   m.def("apply_step_colsum", &apply_step_colsum,
         "K3+colsum fused (bf16): commit F and emit per-stripe column sums");
+  m.def("extract_count", &extract_count,
+        "K7 pass 1: per-row membership counts (threshold/argmax-fallback)");
+  m.def("extract_fill", &extract_fill,
+        "K7 pass 2: write ascending community ids at per-row offsets");
   m.def("mfma_probe", &mfma_probe,
         "MFMA C/D layout probe: one 16x16 D = A @ Bc^T tile");
 }

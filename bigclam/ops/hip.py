@@ -185,6 +185,28 @@ def conductance_full_graph(graph, device) -> "torch.Tensor":
     return cond
 
 
+def extract_membership(
+    F_local: torch.Tensor, k_true: int, delta: float
+) -> Tuple[torch.Tensor, torch.Tensor]:
+    """K7: per-row community memberships of the local F shard on device.
+
+    Returns ``(counts int32 [n], comms int32 [total])`` — row u's
+    memberships are ``comms[offsets[u] : offsets[u] + counts[u]]`` in
+    ascending community order (offsets = exclusive prefix sum of counts).
+    Replaces the reference's driver-side threshold pass
+    (codes/Bigclamv2.scala:226-229) without materializing N×K anywhere.
+    """
+    ext = ensure_loaded()
+    n = F_local.shape[0]
+    counts = torch.empty(n, device=F_local.device, dtype=torch.int32)
+    ext.extract_count(F_local, k_true, delta, counts)
+    offsets = torch.cumsum(counts, 0, dtype=torch.int64) - counts.to(torch.int64)
+    total = int(counts.sum().item())
+    comms = torch.empty(total, device=F_local.device, dtype=torch.int32)
+    ext.extract_fill(F_local, k_true, delta, offsets, comms)
+    return counts, comms
+
+
 def full_llh(
     F: torch.Tensor,
     indptr: torch.Tensor,

@@ -160,6 +160,45 @@ def test_torchrun_bench_cpu(tmp_path):
     assert rec["value"] > 0 and np.isfinite(rec["ms_per_step"])
 
 
+def _worker_extract(rank, world_size, port, out_dir):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    os.environ["RANK"] = str(rank)
+    os.environ["WORLD_SIZE"] = str(world_size)
+    import torch.distributed as dist
+
+    from bigclam.engine.extract import extract_communities_sharded
+
+    dist.init_process_group("gloo", rank=rank, world_size=world_size)
+    try:
+        g = _graph()
+        cfg = BigClamConfig(k=3, device="cpu", seed=5, max_sweeps=4, tol=0.0)
+        tr = Trainer(g, cfg, device=torch.device("cpu"))
+        tr.fit(init="random")
+        pairs = extract_communities_sharded(tr)
+        if rank == 0:
+            np.save(os.path.join(out_dir, "comms.npy"), pairs[0])
+            np.save(os.path.join(out_dir, "nodes.npy"), pairs[1])
+    finally:
+        dist.destroy_process_group()
+
+
+def test_sharded_extraction_equals_single(tmp_path):
+    """K7 sharded extraction (per-rank threshold + compact gather) == the
+    single-shard extraction on the same fitted model."""
+    from bigclam.engine.extract import extract_communities_sharded
+
+    g = _graph()
+    cfg = BigClamConfig(k=3, device="cpu", seed=5, max_sweeps=4, tol=0.0)
+    tr = Trainer(g, cfg, rank=0, world_size=1, device=torch.device("cpu"))
+    tr.fit(init="random")
+    comms1, nodes1 = extract_communities_sharded(tr)
+    mp.spawn(_worker_extract, args=(2, 29690, str(tmp_path)), nprocs=2,
+             join=True)
+    np.testing.assert_array_equal(np.load(tmp_path / "comms.npy"), comms1)
+    np.testing.assert_array_equal(np.load(tmp_path / "nodes.npy"), nodes1)
+
+
 def test_boundary_split_trailing_empty_rows():
     """Interior/boundary split stays exact when trailing degree-0 local
     rows exist (the reduceat-with-clipped-indptr version

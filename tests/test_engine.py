@@ -192,3 +192,58 @@ def test_k_grid_matches_reference_repl_output():
         50, 54, 59, 64, 70, 76, 83, 91, 99, 108, 118, 129, 141, 154,
         168, 184, 200,
     ]
+
+
+def test_local_memberships_matches_extract(small_graph):
+    """K7 torch path: counts/comms stream == the list-based extractor."""
+    from bigclam.engine.extract import (
+        local_memberships,
+        membership_threshold,
+        write_communities,
+        write_membership_pairs,
+    )
+
+    g = small_graph
+    rng = np.random.default_rng(3)
+    k = 5
+    F = torch.from_numpy(
+        (rng.random((g.num_nodes, k)) * 0.4).astype(np.float32)
+    )
+    # force some fallback rows (max below delta) and an all-zero row
+    delta = membership_threshold(g.num_nodes, g.num_edges)
+    F[1] *= 0.0
+    F[2] = delta * 0.5 * torch.tensor([1.0, 1.0, 0.2, 0.1, 0.0])
+    members = extract_communities(F, g.num_edges)
+    counts, comms = local_memberships(F, k, delta)
+    offs = np.concatenate([[0], np.cumsum(counts)])
+    # rebuild per-community lists from the compact stream
+    nodes = np.repeat(np.arange(g.num_nodes, dtype=np.int64), counts)
+    for c in range(k):
+        got = np.sort(nodes[comms == c])
+        np.testing.assert_array_equal(got, np.sort(members[c]))
+    # file writers agree
+    import io as _io
+    import tempfile, os
+
+    with tempfile.TemporaryDirectory() as d:
+        p1, p2 = os.path.join(d, "a.txt"), os.path.join(d, "b.txt")
+        write_communities(p1, members, g.raw_ids)
+        order = np.lexsort((nodes, comms))
+        write_membership_pairs(p2, comms[order], nodes[order], g.raw_ids)
+        assert open(p1).read() == open(p2).read()
+
+
+def test_extract_communities_sharded_single(small_graph):
+    from bigclam.engine.extract import (
+        extract_communities_sharded,
+        membership_threshold,
+    )
+
+    g = small_graph
+    cfg = BigClamConfig(k=3, device="cpu", max_sweeps=10, seed=1)
+    tr = Trainer(g, cfg, rank=0, world_size=1, device=torch.device("cpu"))
+    tr.fit(init="random")
+    comms, nodes = extract_communities_sharded(tr)
+    members = extract_communities(tr.gather_F(), g.num_edges)
+    for c in range(3):
+        np.testing.assert_array_equal(nodes[comms == c], np.sort(members[c]))

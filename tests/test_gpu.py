@@ -439,3 +439,42 @@ def test_fused_mfma_matches_direct_bf16_nslot8():
     torch.testing.assert_close(llh_m, llh_d, rtol=0, atol=0)
     agree = (best_m == best_d).float().mean().item()
     assert agree > 0.95, f"only {agree:.3f} of best-steps agree"
+
+
+@pytest.mark.parametrize("dtype", ["fp32", "bf16"])
+def test_k7_extract_matches_torch(dtype):
+    """K7 device extraction == the torch reference predicate, including
+    argmax-fallback ties, all-zero rows, and padded bf16 columns."""
+    from bigclam.engine.extract import local_memberships, membership_threshold
+
+    g = rmat_graph(11, 8.0, seed=6)
+    k = 37  # odd: exercises bf16 padding (kp=40 > k)
+    cfg, st = _mkstate(g, k, seed=5, dtype=dtype, scale=0.25)
+    delta = membership_threshold(g.num_nodes, g.num_edges)
+    # force fallback + zero rows
+    with torch.no_grad():
+        st.F[1] *= 0.0
+        st.F[2, : k] = torch.linspace(0.0, delta * 0.9, k, device="cuda").to(
+            st.F.dtype
+        )
+    counts, comms = local_memberships(st.F_local, k, delta, use_hip=True)
+    rcounts, rcomms = local_memberships(st.F_local_k, k, delta, use_hip=False)
+    np.testing.assert_array_equal(counts, rcounts)
+    np.testing.assert_array_equal(comms, rcomms)
+    assert counts[1] == 0
+
+
+def test_k7_extract_sharded_end_to_end():
+    from bigclam.engine.extract import (
+        extract_communities_sharded,
+        extract_communities,
+    )
+
+    g = rmat_graph(11, 8.0, seed=8)
+    cfg = BigClamConfig(k=32, device="cuda", seed=2, max_sweeps=5)
+    tr = Trainer(g, cfg, rank=0, world_size=1, device=torch.device("cuda"))
+    tr.fit(init="random")
+    comms, nodes = extract_communities_sharded(tr)
+    members = extract_communities(tr.gather_F(), g.num_edges)
+    for c in range(32):
+        np.testing.assert_array_equal(nodes[comms == c], np.sort(members[c]))
