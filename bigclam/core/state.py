@@ -110,7 +110,7 @@ class ShardState:
         # the direct kernel measures faster (com-Youtube K=8385: 149 vs
         # 159 ms); above 16384 only the MFMA kernel covers the shape.
         mfma_all = self.storage_dtype == torch.bfloat16 and (
-            self.kp <= 8192 or self.kp > 16384
+            self.kp <= 8192 or 16384 < self.kp <= 26000
         )
         thr = int(os.environ.get("BIGCLAM_MFMA_DEG", "1" if mfma_all else "0"))
         if thr == 1:
@@ -151,20 +151,10 @@ class ShardState:
         )
         if self.use_hip:
             _hip_ops().ensure_loaded()  # fail loudly if the .so is missing
-            # fail at construction, not at the first kernel launch, when
-            # the requested K exceeds what the bf16 kernels cover
-            if self.storage_dtype == torch.bfloat16 and self.kp > 16384:
-                if self.kp > 26000:
-                    raise ValueError(
-                        f"bf16 GPU path covers K <= 26000 (got padded "
-                        f"K={self.kp}); use dtype=fp32 for larger K"
-                    )
-                if self.n_mfma != int(self.order.numel()):
-                    raise ValueError(
-                        "bf16 with 16384 < K <= 26000 requires the MFMA "
-                        "fused kernel on every node; unset BIGCLAM_MFMA_DEG "
-                        "or set it to 1"
-                    )
+            # no K cap: shapes beyond the fused kernels' LDS coverage
+            # (fp32 > 8192, bf16 > 26000, or bf16 16384..26000 with MFMA
+            # disabled) take the separate chunked-K1 + K2 path
+            # (ops/hip.py edge_grad_llh_chunked)
 
     # ------------------------------------------------------------------ util
     @property

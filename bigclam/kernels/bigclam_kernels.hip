@@ -2187,6 +2187,209 @@ extern "C" __global__ void __launch_bounds__(BLOCK) k5_conductance(
   }
 }
 
+// ----------------------------------------------------------- chunked K1
+//
+// Large-K gradient path (fp32 K > 8192, bf16 K > 16384, and the only
+// path above the fused kernels' LDS caps).  The one-pass K1 design needs
+// gacc[K] (+ fu) resident in LDS — 100-150 KB at K=25000 caps occupancy
+// at 1 block/CU and the sweep goes latency-bound (measured 235-247
+// ms/sweep, profiles/r01_kernel_opt_log.md).  Chunking K restores
+// occupancy but the per-edge dot x = Fu.Fv must complete before the
+// weight w = 1/(1-exp(-x)) is known, so the pass splits in two around a
+// tiny per-edge x buffer (nnz fp32 — 7.4 MB at the com-Amazon config):
+//
+//   KD kd_dot_t   per k-chunk: stage unpacked fu chunk in LDS, wave-per-
+//                 edge partial dots accumulated into x[e] (lane-0 RMW,
+//                 block-owned rows, no atomics).  LDS = ch*4 B.
+//   KW kw_grad_t  per k-chunk: zero gacc chunk (owned-k), per 256-edge
+//                 tile precompute w into LDS once, then barrier-free
+//                 owned-k fma accumulate of w*fv chunks; chunk write-out
+//                 folds -sumF + fu and accumulates the -Fu.sumF + Fu.Fu
+//                 node terms across chunks.  llh (x-only) is summed
+//                 edge-strided in the first chunk.  LDS = ch*4 + tile.
+//
+// Traffic = 2 edge passes over F chunks + one grad write + fu re-reads —
+// ~2x the one-pass kernel's bytes but at 4-5 blocks/CU instead of 1.
+
+#define KCHUNK 8192  // fp32 LDS chunk: 32 KB -> 4+ blocks/CU
+
+template <bool BF16>
+__global__ void __launch_bounds__(BLOCK) kd_dot_t(
+    const void* __restrict__ Fvp, const long long* __restrict__ indptr,
+    const int* __restrict__ indices, const int* __restrict__ order,
+    float* __restrict__ xbuf, int K, int ch) {
+  const int u = order[blockIdx.x];
+  const long long e0 = indptr[u];
+  const long long e1 = indptr[u + 1];
+  const int tid = threadIdx.x;
+  const int lane = tid & (WAVE - 1);
+  const int wid = tid >> 6;
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  float* fus = reinterpret_cast<float*>(smem);  // ch floats, unpacked fp32
+  const float* Ff = BF16 ? nullptr : reinterpret_cast<const float*>(Fvp);
+  const u32* Fb = BF16 ? reinterpret_cast<const u32*>(Fvp) : nullptr;
+
+  for (int k0 = 0; k0 < K; k0 += ch) {
+    const int cw = min(ch, K - k0);
+    if (BF16) {
+      const u32* fu = Fb + (size_t)u * (K / 2) + k0 / 2;
+      for (int k = tid * 8; k < cw; k += BLOCK * 8) {
+        const f32x8 v = ld8bf(fu + k / 2);
+        *reinterpret_cast<float4*>(fus + k) = v.a;
+        *reinterpret_cast<float4*>(fus + k + 4) = v.b;
+      }
+    } else {
+      const float* fu = Ff + (size_t)u * K + k0;
+      for (int k = tid * 4; k < cw; k += BLOCK * 4)
+        *reinterpret_cast<float4*>(fus + k) = ld4(fu + k);
+    }
+    __syncthreads();
+    for (long long e = e0 + wid; e < e1; e += NWAVE) {
+      float part = 0.f;
+      if (BF16) {
+        const u32* fv = Fb + (size_t)indices[e] * (K / 2) + k0 / 2;
+        for (int k = lane * 8; k < cw; k += WAVE * 8) {
+          const f32x8 b = ld8bf(fv + k / 2);
+          part = dot4(*reinterpret_cast<const float4*>(fus + k), b.a, part);
+          part =
+              dot4(*reinterpret_cast<const float4*>(fus + k + 4), b.b, part);
+        }
+      } else {
+        const float* fv = Ff + (size_t)indices[e] * K + k0;
+        for (int k = lane * 4; k < cw; k += WAVE * 4)
+          part = dot4(*reinterpret_cast<const float4*>(fus + k), ld4(fv + k),
+                      part);
+      }
+      part = wave_allreduce_sum(part);
+      if (lane == 0) xbuf[e] = k0 ? xbuf[e] + part : part;
+    }
+    __syncthreads();
+  }
+}
+
+template <bool BF16>
+__global__ void __launch_bounds__(BLOCK) kw_grad_t(
+    const void* __restrict__ Fvp, const long long* __restrict__ indptr,
+    const int* __restrict__ indices, const float* __restrict__ sumF,
+    const int* __restrict__ order, const float* __restrict__ xbuf,
+    float* __restrict__ grad, double* __restrict__ llh, int K, int ch,
+    float min_p, float max_p) {
+  const int u = order[blockIdx.x];
+  const long long e0 = indptr[u];
+  const long long e1 = indptr[u + 1];
+  const int tid = threadIdx.x;
+  const int lane = tid & (WAVE - 1);
+  const int wid = tid >> 6;
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  float* gacc = reinterpret_cast<float*>(smem);  // ch floats
+  float* wtile = gacc + ch;                      // BLOCK floats
+  const float* Ff = BF16 ? nullptr : reinterpret_cast<const float*>(Fvp);
+  const u32* Fb = BF16 ? reinterpret_cast<const u32*>(Fvp) : nullptr;
+  __shared__ double dred[NWAVE];
+
+  double lacc = 0.0;  // edge llh terms, x-only; summed once (chunk 0)
+  float p_fs = 0.f, p_ff = 0.f;  // node terms, accumulated across chunks
+
+  for (int k0 = 0; k0 < K; k0 += ch) {
+    const int cw = min(ch, K - k0);
+    for (int k = tid * 4; k < cw; k += BLOCK * 4)
+      *reinterpret_cast<float4*>(gacc + k) = float4{0.f, 0.f, 0.f, 0.f};
+    // owned-k zeroing: no barrier needed before the owned-k accumulate,
+    // but wtile below is cross-thread -> tiles carry their own barriers.
+    for (long long et = e0; et < e1; et += BLOCK) {
+      const int ne = (int)min((long long)BLOCK, e1 - et);
+      __syncthreads();
+      if (tid < ne) {
+        const float x = xbuf[et + tid];
+        const float p = clamp_p(__expf(-x), min_p, max_p);
+        wtile[tid] = 1.f / (1.f - p);
+        if (k0 == 0) lacc += (double)log1pf(-p) + (double)x;
+      }
+      __syncthreads();
+      for (int i = 0; i < ne; ++i) {
+        const float w = wtile[i];
+        if (BF16) {
+          const u32* fv = Fb + (size_t)indices[et + i] * (K / 2) + k0 / 2;
+          for (int k = tid * 8; k < cw; k += BLOCK * 8) {
+            const f32x8 b = ld8bf(fv + k / 2);
+            float4 g = *reinterpret_cast<float4*>(gacc + k);
+            g.x = fmaf(w, b.a.x, g.x);
+            g.y = fmaf(w, b.a.y, g.y);
+            g.z = fmaf(w, b.a.z, g.z);
+            g.w = fmaf(w, b.a.w, g.w);
+            *reinterpret_cast<float4*>(gacc + k) = g;
+            float4 h = *reinterpret_cast<float4*>(gacc + k + 4);
+            h.x = fmaf(w, b.b.x, h.x);
+            h.y = fmaf(w, b.b.y, h.y);
+            h.z = fmaf(w, b.b.z, h.z);
+            h.w = fmaf(w, b.b.w, h.w);
+            *reinterpret_cast<float4*>(gacc + k + 4) = h;
+          }
+        } else {
+          const float* fv = Ff + (size_t)indices[et + i] * K + k0;
+          for (int k = tid * 4; k < cw; k += BLOCK * 4) {
+            const float4 b = ld4(fv + k);
+            float4 g = *reinterpret_cast<float4*>(gacc + k);
+            g.x = fmaf(w, b.x, g.x);
+            g.y = fmaf(w, b.y, g.y);
+            g.z = fmaf(w, b.z, g.z);
+            g.w = fmaf(w, b.w, g.w);
+            *reinterpret_cast<float4*>(gacc + k) = g;
+          }
+        }
+      }
+    }
+    // chunk write-out (owned-k): grad = gacc - sumF + fu; node terms
+    float* __restrict__ gout = grad + (size_t)u * K + k0;
+    if (BF16) {
+      const u32* fu = Fb + (size_t)u * (K / 2) + k0 / 2;
+      for (int k = tid * 8; k < cw; k += BLOCK * 8) {
+        const f32x8 a = ld8bf(fu + k / 2);
+        const float4 sA = ld4(sumF + k0 + k);
+        const float4 sB = ld4(sumF + k0 + k + 4);
+        p_fs = dot4(a.a, sA, dot4(a.b, sB, p_fs));
+        p_ff = dot4(a.a, a.a, dot4(a.b, a.b, p_ff));
+        const float4 g = *reinterpret_cast<const float4*>(gacc + k);
+        const float4 h = *reinterpret_cast<const float4*>(gacc + k + 4);
+        *reinterpret_cast<float4*>(gout + k) =
+            float4{g.x - sA.x + a.a.x, g.y - sA.y + a.a.y,
+                   g.z - sA.z + a.a.z, g.w - sA.w + a.a.w};
+        *reinterpret_cast<float4*>(gout + k + 4) =
+            float4{h.x - sB.x + a.b.x, h.y - sB.y + a.b.y,
+                   h.z - sB.z + a.b.z, h.w - sB.w + a.b.w};
+      }
+    } else {
+      const float* fu = Ff + (size_t)u * K + k0;
+      for (int k = tid * 4; k < cw; k += BLOCK * 4) {
+        const float4 a = ld4(fu + k);
+        const float4 s = ld4(sumF + k0 + k);
+        p_fs = dot4(a, s, p_fs);
+        p_ff = dot4(a, a, p_ff);
+        const float4 g = *reinterpret_cast<const float4*>(gacc + k);
+        *reinterpret_cast<float4*>(gout + k) =
+            float4{g.x - s.x + a.x, g.y - s.y + a.y, g.z - s.z + a.z,
+                   g.w - s.w + a.w};
+      }
+    }
+    __syncthreads();  // gacc reused next chunk after all reads complete
+  }
+
+  // block-reduce the fp64 edge-llh partials and fp32 node terms
+  double dv = lacc;
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) dv += __shfl_xor(dv, off, WAVE);
+  if (lane == 0) dred[wid] = dv;
+  const float fs = block_allreduce_sum(p_fs, wtile);  // wtile as scratch
+  const float ff = block_allreduce_sum(p_ff, wtile);
+  __syncthreads();
+  if (tid == 0) {
+    double t = 0.0;
+#pragma unroll
+    for (int wv = 0; wv < NWAVE; ++wv) t += dred[wv];
+    llh[u] = t + (double)(-fs) + (double)ff;
+  }
+}
+
 // ------------------------------------------------------------------- K7
 //
 // Community extraction (codes/Bigclamv2.scala:223-230): node u belongs to
@@ -2621,6 +2824,32 @@ extern "C" void launch_kf_mfma_bf16(const void* F, const long long* indptr,
     KFMB_CASE(0);  // fu via LDS only (see kf_phase_a_bf16_lds)
   }
 #undef KFMB_CASE
+  HIP_CHECK(hipGetLastError());
+}
+
+extern "C" void launch_k1_chunked(const void* F, int bf16,
+                                  const long long* indptr, const int* indices,
+                                  const float* sumF, const int* order,
+                                  float* xbuf, float* grad, double* llh,
+                                  int n_blocks, int K, float min_p,
+                                  float max_p, hipStream_t stream) {
+  if (n_blocks == 0) return;
+  const int ch = (K < KCHUNK) ? K : KCHUNK;
+  const size_t lds_d = (size_t)ch * 4;
+  const size_t lds_w = (size_t)ch * 4 + BLOCK * 4;
+  if (bf16) {
+    hipLaunchKernelGGL((kd_dot_t<true>), dim3(n_blocks), dim3(BLOCK), lds_d,
+                       stream, F, indptr, indices, order, xbuf, K, ch);
+    hipLaunchKernelGGL((kw_grad_t<true>), dim3(n_blocks), dim3(BLOCK), lds_w,
+                       stream, F, indptr, indices, sumF, order, xbuf, grad,
+                       llh, K, ch, min_p, max_p);
+  } else {
+    hipLaunchKernelGGL((kd_dot_t<false>), dim3(n_blocks), dim3(BLOCK), lds_d,
+                       stream, F, indptr, indices, order, xbuf, K, ch);
+    hipLaunchKernelGGL((kw_grad_t<false>), dim3(n_blocks), dim3(BLOCK),
+                       lds_w, stream, F, indptr, indices, sumF, order, xbuf,
+                       grad, llh, K, ch, min_p, max_p);
+  }
   HIP_CHECK(hipGetLastError());
 }
 

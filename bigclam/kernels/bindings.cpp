@@ -51,6 +51,10 @@ extern "C" void launch_kf_mfma_bf16(const void*, const long long*, const int*,
                                     const float*, float*, int, int, int,
                                     float, float, float, float, float,
                                     hipStream_t);
+extern "C" void launch_k1_chunked(const void*, int, const long long*,
+                                  const int*, const float*, const int*,
+                                  float*, float*, double*, int, int, float,
+                                  float, hipStream_t);
 extern "C" void launch_k7_count(const void*, int, int, int, int, float, int*,
                                 hipStream_t);
 extern "C" void launch_k7_fill(const void*, int, int, int, int, float,
@@ -303,6 +307,36 @@ void mfma_probe(torch::Tensor A, torch::Tensor Bc, torch::Tensor D) {
   }
 }
 
+// Chunked large-K K1 (KD dot pass into the per-edge x buffer, then KW
+// chunked weighted accumulate) — no K cap; see kd_dot_t/kw_grad_t.
+void edge_grad_llh_chunked(torch::Tensor F, torch::Tensor indptr,
+                           torch::Tensor indices, torch::Tensor sumF,
+                           torch::Tensor order, torch::Tensor grad,
+                           torch::Tensor llh, torch::Tensor xbuf,
+                           double min_p, double max_p) {
+  CHECK_F(F);
+  CHECK_IN(indptr, torch::kInt64);
+  CHECK_IN(indices, torch::kInt32);
+  CHECK_IN(sumF, torch::kFloat32);
+  CHECK_IN(order, torch::kInt32);
+  CHECK_IN(grad, torch::kFloat32);
+  CHECK_IN(llh, torch::kFloat64);
+  CHECK_IN(xbuf, torch::kFloat32);
+  const int n_local = (int)indptr.size(0) - 1;
+  const int n_blocks = (int)order.size(0);
+  const int K = (int)F.size(1);
+  TORCH_CHECK(grad.size(0) == n_local && grad.size(1) == K);
+  TORCH_CHECK(xbuf.size(0) >= indices.size(0), "xbuf too small");
+  TORCH_CHECK(K % (is_bf16(F) ? 8 : 4) == 0, "K must be padded");
+  const auto ip =
+      reinterpret_cast<const long long*>(indptr.data_ptr<int64_t>());
+  launch_k1_chunked(F.data_ptr(), is_bf16(F) ? 1 : 0, ip,
+                    indices.data_ptr<int>(), sumF.data_ptr<float>(),
+                    order.data_ptr<int>(), xbuf.data_ptr<float>(),
+                    grad.data_ptr<float>(), llh.data_ptr<double>(), n_blocks,
+                    K, (float)min_p, (float)max_p, current_stream());
+}
+
 // K7 community extraction: two deterministic passes (count, then fill
 // after a host/torch prefix-sum) — see k7_membership in the .hip file.
 void extract_count(torch::Tensor F_local, int64_t k_true, double delta,
@@ -363,6 +397,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "(MFMA tiles for the high-degree prefix)");
   m.def("apply_step_colsum", &apply_step_colsum,
         "K3+colsum fused (bf16): commit F and emit per-stripe column sums");
+  m.def("edge_grad_llh_chunked", &edge_grad_llh_chunked,
+        "K1 large-K: KD per-edge dots + KW chunked weighted accumulate");
   m.def("extract_count", &extract_count,
         "K7 pass 1: per-row membership counts (threshold/argmax-fallback)");
   m.def("extract_fill", &extract_fill,

@@ -32,6 +32,33 @@ def ensure_loaded():
     return _C
 
 
+_xbuf_cache = {}
+
+
+def _xbuf(nnz: int, device) -> torch.Tensor:
+    """Per-edge dot buffer for the chunked K1 (reused across sweeps)."""
+    key = str(device)
+    t = _xbuf_cache.get(key)
+    if t is None or t.shape[0] < nnz:
+        t = torch.empty(nnz, device=device, dtype=torch.float32)
+        _xbuf_cache[key] = t
+    return t
+
+
+def _use_chunked_k1(F: torch.Tensor) -> bool:
+    """Chunked KD+KW replaces the one-pass K1 where gacc[K] LDS residency
+    would cap occupancy at 1 block/CU (measured 235-247 ms/sweep at
+    K=25000, profiles/r01_kernel_opt_log.md).  BIGCLAM_K1_CHUNKED=1/0
+    forces it on/off (tests + dispatch measurements)."""
+    import os
+
+    env = os.environ.get("BIGCLAM_K1_CHUNKED")
+    if env is not None:
+        return env != "0"
+    cap = 16384 if F.dtype == torch.bfloat16 else 8192
+    return F.shape[1] > cap
+
+
 def edge_grad_llh(
     F: torch.Tensor,
     indptr: torch.Tensor,
@@ -53,9 +80,15 @@ def edge_grad_llh(
         llh = torch.empty(n_local, device=F.device, dtype=torch.float64)
     else:
         grad, llh = out
-    ext.edge_grad_llh(
-        F, indptr, indices, sumF, order, grad, llh, cfg.min_p, cfg.max_p
-    )
+    if _use_chunked_k1(F):
+        ext.edge_grad_llh_chunked(
+            F, indptr, indices, sumF, order, grad, llh,
+            _xbuf(indices.shape[0], F.device), cfg.min_p, cfg.max_p,
+        )
+    else:
+        ext.edge_grad_llh(
+            F, indptr, indices, sumF, order, grad, llh, cfg.min_p, cfg.max_p
+        )
     return grad, llh
 
 
