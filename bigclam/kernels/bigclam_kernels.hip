@@ -2556,8 +2556,9 @@ extern "C" void launch_k2_bf16(const void* F, const long long* indptr,
   if (n_local == 0) return;
   if (n_ladder > 16) throw std::runtime_error("ladder length > 16 unsupported");
   const u32* Fb = reinterpret_cast<const u32*>(F);
+  const bool nostage = getenv("BIGCLAM_K2_NOSTAGE") != nullptr;
   const size_t lds = (size_t)K * 6;  // g fp32 + fu raw bf16
-  if (lds + 2048 <= 160 * 1024) {
+  if (!nostage && lds + 2048 <= 160 * 1024) {
     allow_large_lds((const void*)&k2_ls_v3_bf16<true>, lds);
     hipLaunchKernelGGL((k2_ls_v3_bf16<true>), dim3(n_local), dim3(256), lds,
                        stream, Fb, indptr, indices, sumF, grad, llh, order,
