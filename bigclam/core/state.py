@@ -103,15 +103,15 @@ class ShardState:
         # taking the MFMA phase-B kernel (deg >= threshold) are a prefix;
         # precompute the split points.  BIGCLAM_MFMA_DEG overrides
         # (0 = all nodes on the direct kernel).  Default is K-dependent,
-        # measured in profiles/r01_kernel_opt_log.md: at kp <= 8192
-        # (NSLOT<=4, spill-free) MFMA-all wins at EVERY degree (16.4 vs
-        # 22.6 ms on the com-Amazon CSR, mean degree 5.5); at
-        # 8192 < kp <= 16384 the NSLOT=8 template spills 20 B/lane and
-        # the direct kernel measures faster (com-Youtube K=8385: 149 vs
-        # 159 ms); above 16384 only the MFMA kernel covers the shape.
-        mfma_all = self.storage_dtype == torch.bfloat16 and (
-            self.kp <= 8192 or 16384 < self.kp <= 26000
-        )
+        # measured in profiles/r01_kernel_opt_log.md and
+        # r02_largek_dispatch.md: at kp <= 8192 (NSLOT<=4, spill-free)
+        # MFMA-all wins at EVERY degree (16.4 vs 22.6 ms on the com-Amazon
+        # CSR, mean degree 5.5); at 8192 < kp <= 16384 the NSLOT=8
+        # template spills 20 B/lane and the direct kernel measures faster
+        # (com-Youtube K=8385: 149 vs 159 ms); above 16384 the separate
+        # chunked-K1 + unstaged-K2 path beats the 1-block/CU MFMA fused
+        # kernel (145 vs 236 ms at K=25000 bf16), so MFMA is off there.
+        mfma_all = self.storage_dtype == torch.bfloat16 and self.kp <= 8192
         thr = int(os.environ.get("BIGCLAM_MFMA_DEG", "1" if mfma_all else "0"))
         if thr == 1:
             # "all nodes": include any degree-0 rows too (the MFMA kernel

@@ -2558,7 +2558,10 @@ extern "C" void launch_k2_bf16(const void* F, const long long* indptr,
   const u32* Fb = reinterpret_cast<const u32*>(F);
   const bool nostage = getenv("BIGCLAM_K2_NOSTAGE") != nullptr;
   const size_t lds = (size_t)K * 6;  // g fp32 + fu raw bf16
-  if (!nostage && lds + 2048 <= 160 * 1024) {
+  // staged only while >= 2 blocks/CU fit: at K=25000 the 150 KB staged
+  // variant (1 block/CU) measured 205 ms/sweep vs 145 unstaged
+  // (profiles/r02_largek_dispatch.md)
+  if (!nostage && lds + 2048 <= 80 * 1024) {
     allow_large_lds((const void*)&k2_ls_v3_bf16<true>, lds);
     hipLaunchKernelGGL((k2_ls_v3_bf16<true>), dim3(n_local), dim3(256), lds,
                        stream, Fb, indptr, indices, sumF, grad, llh, order,
@@ -2648,7 +2651,9 @@ extern "C" void launch_k2(const float* F, const long long* indptr,
                        indptr, indices, sumF, grad, llh, order, ladder, best,
                        n_local, K, n_ladder, alpha, min_p, max_p, min_f,
                        max_f);
-  } else if (!nostage && lds + 2048 <= 160 * 1024) {
+  } else if (!nostage && lds + 2048 <= 80 * 1024) {
+    // same 2-blocks/CU staging rule as the bf16 launcher (measured:
+    // 1-block/CU staged loses to unstaged at K=25000)
     allow_large_lds((const void*)&k2_ls_v3<true>, lds);
     hipLaunchKernelGGL((k2_ls_v3<true>), dim3(n_local), dim3(256), lds,
                        stream, F, indptr, indices, sumF, grad, llh, order,

@@ -397,12 +397,15 @@ def test_fused_mfma_split_order_launches_match_full_bf16():
     torch.testing.assert_close(best, full[2], rtol=0, atol=0)
 
 
-def test_fused_mfma_large_k_bf16():
+def test_fused_mfma_large_k_bf16(monkeypatch):
     """K=17000 > the direct kernel's 16384 cap exercises the LDS-only
     phase A (kf_phase_a_bf16_lds) + MFMA phase B; compared against the
-    torch reference ops (fp32 math on the same bf16 F)."""
+    torch reference ops (fp32 math on the same bf16 F).  The DEFAULT
+    above K=16384 is now the separate chunked path (r02 dispatch
+    measurement) — MFMA-all is forced here to keep the kernel covered."""
     from bigclam.ops import hip as hip_ops
 
+    monkeypatch.setenv("BIGCLAM_MFMA_DEG", "1")
     g = rmat_graph(8, 6.0, seed=58)  # ~250 nodes: keep K=17k cheap
     cfg, st = _mkstate_dtype(g, 17000, "bf16", seed=22, scale=0.02)
     assert st.fused_ok and st.n_mfma == int(st.order.numel())
