@@ -156,8 +156,10 @@ def test_reference_ops_extreme_values_finite():
 
 def test_mfma_dispatch_defaults_by_k():
     """The measured K-aware dispatch defaults (state.py): bf16 routes ALL
-    nodes to the MFMA kernel at kp <= 8192 and above 16384, none in the
-    spill regime between; fp32 defaults to the direct kernel."""
+    nodes to the MFMA kernel at kp <= 8192 only; above that the direct
+    fused kernel (<= 16384) or the separate chunked path (> 16384) wins
+    (profiles/r02_largek_dispatch.md); fp32 defaults to the direct
+    kernel."""
     import torch
 
     from bigclam.config import BigClamConfig
@@ -169,7 +171,7 @@ def test_mfma_dispatch_defaults_by_k():
     for dtype, k, expect_all in [
         ("bf16", 5000, True),
         ("bf16", 8500, False),
-        ("bf16", 17000, True),
+        ("bf16", 17000, False),
         ("fp32", 5000, False),
     ]:
         st = ShardState(
