@@ -97,9 +97,40 @@ def parse_edge_array(path: str) -> np.ndarray:
         return np.asarray(rows, dtype=np.int64).reshape(-1, 2)
 
 
+def _native_build_threshold() -> int:
+    v = os.environ.get("BIGCLAM_NATIVE_CSR")
+    if v == "0":
+        return 1 << 62  # disabled
+    if v == "1":
+        return 0  # always
+    return 2_000_000  # default: large arrays only
+
+
 def build_graph(edges: np.ndarray, drop_self_loops: bool = True) -> Graph:
-    """Build the canonical undirected CSR from a raw [M, 2] edge array."""
+    """Build the canonical undirected CSR from a raw [M, 2] edge array.
+
+    Large arrays take the parallel native builder (io_native.cpp
+    ``build_csr`` — bucket-sorted dedupe + atomic-cursor scatter), which
+    produces EXACTLY the same Graph as the numpy path below (tested in
+    tests/test_io.py); numpy remains the reference and the fallback for
+    sparse 64-bit id spaces.  ``BIGCLAM_NATIVE_CSR=1/0`` forces it
+    on/off."""
     edges = np.asarray(edges, dtype=np.int64).reshape(-1, 2)
+    if len(edges) >= _native_build_threshold():
+        try:
+            from .. import _io_native
+
+            if len(edges) == 0 or edges.max() < (1 << 31):
+                indptr, indices, raw_ids = _io_native.build_csr(
+                    np.ascontiguousarray(edges), drop_self_loops
+                )
+                return Graph(
+                    indptr=np.asarray(indptr),
+                    indices=np.asarray(indices),
+                    raw_ids=np.asarray(raw_ids),
+                )
+        except ImportError:
+            pass
     if drop_self_loops:
         edges = edges[edges[:, 0] != edges[:, 1]]
     # dense ID remap

@@ -33,6 +33,19 @@ def rmat_edges(
     """
     n = 1 << scale
     m = int(edge_factor * n)
+    if m >= 2_000_000:
+        # counter-based parallel native generator (io_native.cpp):
+        # deterministic per (seed, edge index), thread-count independent.
+        # Different stream than the numpy path below — gated by size so
+        # small test fixtures keep their numpy streams.
+        try:
+            from .. import _io_native
+
+            return np.asarray(
+                _io_native.rmat_edges(scale, edge_factor, a, b, c, seed)
+            )
+        except ImportError:
+            pass
     rng = np.random.default_rng(seed)
     src = np.zeros(m, dtype=np.int64)
     dst = np.zeros(m, dtype=np.int64)
