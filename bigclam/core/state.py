@@ -241,8 +241,11 @@ class ShardState:
     def fused_ok(self) -> bool:
         """The fused K1+K2 kernel covers fp32 rows up to K=8192 and bf16
         rows up to K=16384 (LDS holds grad + fu); other shapes use the
-        separate kernels."""
+        separate kernels.  BIGCLAM_NO_FUSED=1 forces the separate path
+        (dispatch measurements)."""
         if not self.use_hip:
+            return False
+        if os.environ.get("BIGCLAM_NO_FUSED"):
             return False
         if self.storage_dtype == torch.float32:
             return self.kp <= 8192

@@ -347,7 +347,8 @@ static inline uint64_t splitmix64(uint64_t& s) {
 
 py::array_t<int64_t> rmat_edges_native(int scale, double edge_factor,
                                        double a, double b, double c,
-                                       uint64_t seed, int n_threads) {
+                                       uint64_t seed, int n_threads,
+                                       bool scramble_ids) {
   if (scale < 1 || scale > 30) throw std::runtime_error("bad scale");
   const int64_t n = int64_t(1) << scale;
   const int64_t m = (int64_t)(edge_factor * (double)n);
@@ -388,8 +389,8 @@ py::array_t<int64_t> rmat_edges_native(int scale, double edge_factor,
           src = (src << 1) | (uint64_t)sbit;
           dst = (dst << 1) | (uint64_t)dbit;
         }
-        O[2 * i] = (int64_t)scramble(src);
-        O[2 * i + 1] = (int64_t)scramble(dst);
+        O[2 * i] = (int64_t)(scramble_ids ? scramble(src) : src);
+        O[2 * i + 1] = (int64_t)(scramble_ids ? scramble(dst) : dst);
       }
     });
   }
@@ -408,5 +409,6 @@ PYBIND11_MODULE(_io_native, m) {
   m.def("rmat_edges", &rmat_edges_native, py::arg("scale"),
         py::arg("edge_factor"), py::arg("a") = 0.57, py::arg("b") = 0.19,
         py::arg("c") = 0.19, py::arg("seed") = 0, py::arg("n_threads") = 0,
+        py::arg("scramble_ids") = true,
         "counter-based parallel R-MAT edge pairs [m, 2]");
 }

@@ -1,0 +1,111 @@
+#!/usr/bin/env python3
+"""Fit-to-convergence at the headline configs.
+
+Runs the reference's actual job — conductance seed init, Armijo sweeps to
+tol=1e-4, community extraction — on GPU at the named shapes, recording the
+LLH trajectory, sweep count, wall-clock and community stats.
+
+  python tools/convergence_run.py --which amazon-bf16,amazon-fp32,enron-bf16
+"""
+import argparse
+import json
+import os
+import sys
+import time
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+import numpy as np  # noqa: E402
+import torch  # noqa: E402
+
+from bigclam.config import BigClamConfig  # noqa: E402
+from bigclam.engine.extract import (  # noqa: E402
+    extract_communities_sharded,
+    membership_threshold,
+)
+from bigclam.engine.trainer import Trainer  # noqa: E402
+from bigclam.io import shaped_graph  # noqa: E402
+from bigclam.utils.metrics import MetricsLogger  # noqa: E402
+
+SHAPES = {
+    # Email-Enron: 36,692 nodes / 183,831 undirected edges (data/Email-Enron.txt)
+    "enron": (36692, 183831),
+    # com-Amazon: 334,863 nodes / 925,872 undirected edges (SNAP; missing blob upstream)
+    "amazon": (334863, 925872),
+}
+RUNS = {
+    "enron-bf16": ("enron", 500, "bf16"),
+    "amazon-bf16": ("amazon", 5000, "bf16"),
+    "amazon-fp32": ("amazon", 5000, "fp32"),
+}
+
+
+def community_stats(comms, nodes, n, k):
+    sizes = np.bincount(comms, minlength=k)
+    nz = sizes[sizes > 0]
+    covered = len(np.unique(nodes))
+    return {
+        "nonempty_communities": int(len(nz)),
+        "k": k,
+        "size_p50": int(np.percentile(nz, 50)) if len(nz) else 0,
+        "size_p90": int(np.percentile(nz, 90)) if len(nz) else 0,
+        "size_max": int(nz.max()) if len(nz) else 0,
+        "node_coverage": round(covered / n, 4),
+        "total_memberships": int(len(nodes)),
+    }
+
+
+def run(name, init, max_sweeps, out_dir):
+    shape, k, dtype = RUNS[name]
+    n, e = SHAPES[shape]
+    g = shaped_graph(n, e, locality=0.7, seed=42)
+    cfg = BigClamConfig(k=k, dtype=dtype, device="cuda", seed=7, tol=1e-4,
+                        max_sweeps=max_sweeps)
+    metrics = MetricsLogger(
+        os.path.join(out_dir, f"conv_{name}_{init}.jsonl"), rank=0, quiet=True
+    )
+    tr = Trainer(g, cfg, rank=0, world_size=1,
+                 device=torch.device("cuda"), metrics=metrics)
+    t0 = time.perf_counter()
+    res = tr.fit(init=init)
+    torch.cuda.synchronize()
+    fit_s = time.perf_counter() - t0
+    nnz = float((tr.state.F_local_k != 0).float().mean().item())
+    rec = {
+        "run": name, "init": init, "graph": f"{shape}-shaped", "n": n,
+        "undirected_edges": e, "k": k, "dtype": dtype,
+        "sweeps": res.sweeps, "converged": res.converged,
+        "llh_first": res.llh_history[0] if res.llh_history else None,
+        "llh_final": res.llh, "fit_wall_s": round(fit_s, 2),
+        "ms_per_sweep": round(fit_s / max(res.sweeps, 1) * 1000.0, 2),
+        "f_nnz_frac": round(nnz, 4), "collapsed": nnz == 0.0,
+        "tol": 1e-4,
+    }
+    if nnz > 0.0:
+        t0 = time.perf_counter()
+        comms, nodes = extract_communities_sharded(tr)
+        rec["extract_s"] = round(time.perf_counter() - t0, 2)
+        rec["delta"] = round(membership_threshold(n, e), 6)
+        rec["communities"] = community_stats(comms, nodes, n, k)
+    print(json.dumps(rec), flush=True)
+    return rec
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--which", default="amazon-bf16,amazon-fp32,enron-bf16")
+    ap.add_argument("--max-sweeps", type=int, default=400)
+    ap.add_argument("--out-dir", default="gpurun_out")
+    args = ap.parse_args()
+    assert torch.cuda.is_available()
+    os.makedirs(args.out_dir, exist_ok=True)
+    for name in args.which.split(","):
+        rec = run(name, "seed", args.max_sweeps, args.out_dir)
+        if rec["collapsed"]:
+            # the documented absorbing-state fallback (r01 finding: the
+            # reference's indicator init has the same property)
+            run(name, "random", args.max_sweeps, args.out_dir)
+
+
+if __name__ == "__main__":
+    main()

@@ -24,18 +24,25 @@ from bigclam.utils.metrics import MetricsLogger  # noqa: E402
 
 CONFIGS = [
     # label, dtype, env overrides
-    ("bf16-fused-mfma", "bf16", {}),
-    ("bf16-sep-chunked-k2staged", "bf16", {"BIGCLAM_MFMA_DEG": "0"}),
+    ("bf16-default", "bf16", {}),
+    ("bf16-mfma-all", "bf16", {"BIGCLAM_MFMA_DEG": "1"}),
+    ("bf16-sep-chunked-k2staged", "bf16",
+     {"BIGCLAM_MFMA_DEG": "0", "BIGCLAM_NO_FUSED": "1",
+      "BIGCLAM_K1_CHUNKED": "1"}),
     ("bf16-sep-chunked-k2nostage", "bf16",
-     {"BIGCLAM_MFMA_DEG": "0", "BIGCLAM_K2_NOSTAGE": "1"}),
-    ("fp32-chunked-k2unstaged", "fp32", {}),
-    ("fp32-chunked-k2tiled", "fp32", {"BIGCLAM_K2_TILED": "1"}),
-    ("fp32-onepass-k1 (r01 baseline)", "fp32", {"BIGCLAM_K1_CHUNKED": "0"}),
+     {"BIGCLAM_MFMA_DEG": "0", "BIGCLAM_K2_NOSTAGE": "1",
+      "BIGCLAM_NO_FUSED": "1", "BIGCLAM_K1_CHUNKED": "1"}),
+    ("fp32-default", "fp32", {}),
+    ("fp32-sep-chunked-k2tiled", "fp32",
+     {"BIGCLAM_K2_TILED": "1", "BIGCLAM_NO_FUSED": "1",
+      "BIGCLAM_K1_CHUNKED": "1"}),
+    ("fp32-sep-onepass-k1", "fp32",
+     {"BIGCLAM_K1_CHUNKED": "0", "BIGCLAM_NO_FUSED": "1"}),
 ]
 
 ENV_KEYS = [
     "BIGCLAM_MFMA_DEG", "BIGCLAM_K2_NOSTAGE", "BIGCLAM_K2_TILED",
-    "BIGCLAM_K1_CHUNKED",
+    "BIGCLAM_K1_CHUNKED", "BIGCLAM_NO_FUSED",
 ]
 
 
