@@ -298,6 +298,116 @@ class ShardState:
         grad, llh = self.grad_llh_overlap(halo_work)
         return grad, llh, self.linesearch(grad, llh)
 
+    # ------------------------------------------------- sparse-adaptive path
+    @property
+    def sparse_cap(self) -> int:
+        """Active-set bound under which a node routes to K1S/K2S.  kp/4
+        keeps the compact work well under the dense kernels' while
+        letting moderate-degree nodes route (bound ≈ deg·density·K);
+        hubs always stay on the dense fused kernel."""
+        return int(os.environ.get("BIGCLAM_SPARSE_CAP", str(self.kp // 4)))
+
+    @property
+    def sparse_allowed(self) -> bool:
+        """Active-column sweep (docs/sparse_sweep_design.md): exact
+        per-node routing by active-set bound; pays off once F sparsifies
+        (measured 0.4-4% converged density at the headline configs).
+        K1S holds a dense K-float LDS accumulator, and the dense-node
+        remainder runs the fused kernel -> shapes limited to the fused
+        coverage (fp32 kp <= 8192, bf16 kp <= 16384).
+        BIGCLAM_SPARSE=0 disables."""
+        cap = 16384 if self.storage_dtype == torch.bfloat16 else 8192
+        return (
+            self.use_hip
+            and self.kp <= cap
+            and os.environ.get("BIGCLAM_SPARSE", "1") != "0"
+        )
+
+    def grad_ls_auto(self, halo_work):
+        """Per-sweep adaptive dispatch: route nodes whose active-set bound
+        (own support + sum of neighbor supports, exact upper bound on
+        |S_u|) fits under ``sparse_cap`` to the compact K1S/K2S kernels,
+        the rest to the dense fused path.  Returns
+        (grad, llh_nodes, best, sparse_pack|None)."""
+        if not self.sparse_allowed:
+            g, l, b = self.fused_grad_ls_overlap(halo_work)
+            return g, l, b, None
+        ops = _hip_ops()
+        # KAF needs fresh halo rows: wait before scanning F
+        if halo_work is not None:
+            halo_work.wait()
+        dev = self.device
+        n_rows = self.F.shape[0]
+        cap = self.sparse_cap
+        scount = torch.empty(n_rows, device=dev, dtype=torch.int32)
+        soffset = torch.zeros(n_rows, device=dev, dtype=torch.int64)
+        empty_i = torch.empty(0, device=dev, dtype=torch.int32)
+        empty_f = torch.empty(0, device=dev, dtype=torch.float32)
+        ops.ensure_loaded().sparse_support(
+            self.F, soffset, scount, empty_i, empty_f, cap, False
+        )
+        sc = scount.to(torch.int64)
+        # bound_u = s_u + sum_{v in N(u)} s_v  (cumsum segment trick)
+        sv_edges = sc[self.indices.long()]
+        cs = torch.cat(
+            [torch.zeros(1, device=dev, dtype=torch.int64),
+             torch.cumsum(sv_edges, 0)]
+        )
+        bound = sc[: self.n_local] + cs[self.indptr[1:]] - cs[self.indptr[:-1]]
+        is_sparse = bound <= cap
+        n_s = int(is_sparse.sum().item())
+        if n_s < max(64, self.n_local // 20):
+            g, l, b = self.fused_grad_ls_overlap(None)
+            return g, l, b, None
+        # fill the support pools (rows with count <= cap only)
+        keep = torch.where(sc <= cap, sc, torch.zeros_like(sc))
+        soffset = torch.cumsum(keep, 0) - keep
+        pool = int(keep.sum().item())
+        sidx = torch.empty(pool, device=dev, dtype=torch.int32)
+        sval = torch.empty(pool, device=dev, dtype=torch.float32)
+        ops.ensure_loaded().sparse_support(
+            self.F, soffset, scount, sidx, sval, cap, True
+        )
+        # split the degree-descending launch order
+        om = is_sparse[self.order.long()]
+        order_s = self.order[om].contiguous()
+        order_d = self.order[~om].contiguous()
+        bnd_s = bound[order_s.long()]
+        goffset = torch.cumsum(bnd_s, 0) - bnd_s
+        gpool = int(bnd_s.sum().item())
+        n, kp = self.n_local, self.kp
+        grad = torch.empty(n, kp, device=dev, dtype=torch.float32)
+        llh = torch.empty(n, device=dev, dtype=torch.float64)
+        best = torch.empty(n, device=dev, dtype=torch.float32)
+        n_d = int(order_d.numel())
+        if n_d:
+            n_mfma_d = n_d if self.n_mfma == int(self.order.numel()) else 0
+            ops.fused_grad_ls(
+                self.F, self.indptr, self.indices, self.sumF, order_d,
+                self.cfg, out=(grad, llh, best), n_mfma=n_mfma_d,
+            )
+        pack = ops.sparse_sweep_part(
+            self.F, self.indptr, self.indices, self.sumF, order_s,
+            soffset, sidx, sval, scount, goffset, gpool, cap, llh, best,
+            self.cfg,
+        )
+        pack["best"] = best
+        # dense commit must skip sparse rows (their grad rows are unset)
+        steps_dense = best.clone()
+        steps_dense[order_s.long()] = 0.0
+        pack["steps_dense"] = steps_dense
+        return grad, llh, best, pack
+
+    def apply_commit(self, grad, steps, pack):
+        """Commit a sweep's accepted steps: K3S for sparse-routed rows
+        (first), then the dense commit + exact column-sum refresh (which
+        reads the POST-commit F for every row)."""
+        if pack is not None:
+            _hip_ops().sparse_commit(self.F_local, pack, pack["best"],
+                                     self.cfg)
+            steps = pack["steps_dense"]
+        self.apply_step(grad, steps)
+
     def grad_llh_overlap(self, halo_work) -> Tuple[torch.Tensor, torch.Tensor]:
         """K1 overlapped with the in-flight halo exchange: interior nodes
         (no halo neighbors) run while the all_to_all completes, boundary

@@ -139,30 +139,33 @@ class Trainer:
     # trajectory, one edge pass fewer).
 
     def prologue(self):
-        """Halo + fused grad/line-search on the current state; returns
+        """Halo + grad/line-search on the current state; returns
         (carry, llh_total).  llh_total is the objective of the CURRENT
         state (the reference v2's initial LLH, codes/Bigclamv2.scala:204);
-        carry = (grad, steps) feeds the first pipelined_sweep."""
+        carry = (grad, steps, sparse_pack) feeds the first
+        pipelined_sweep."""
         st = self.state
         work = st.halo_exchange(async_op=True)
-        grad, llh_nodes, steps = st.fused_grad_ls_overlap(work)
+        grad, llh_nodes, steps, pack = st.grad_ls_auto(work)
         total = llh_nodes.sum().reshape(1)
         comm.all_reduce_(total)
-        return (grad, steps), float(total.item())
+        return (grad, steps, pack), float(total.item())
 
     def pipelined_sweep(self, carry):
-        """One iteration: K3 commit of the carried steps -> async halo ->
-        fused KF (= K1 grad+LLH and K2 line search; interior nodes overlap
-        the halo all_to_all, boundary after).  Returns
+        """One iteration: commit of the carried steps (K3S for
+        sparse-routed rows + dense K3) -> async halo -> adaptive grad +
+        line search (compact K1S/K2S for nodes whose active-set bound
+        fits, the fused dense kernels for the rest; dense-only sweeps
+        keep the interior/boundary halo overlap).  Returns
         (carry', llh_total_after_commit, committed_steps)."""
         st = self.state
-        grad, steps = carry
-        st.apply_step(grad, steps)
+        grad, steps, pack = carry
+        st.apply_commit(grad, steps, pack)
         work = st.halo_exchange(async_op=True)
-        grad, llh_nodes, steps_next = st.fused_grad_ls_overlap(work)
+        grad, llh_nodes, steps_next, pack = st.grad_ls_auto(work)
         total = llh_nodes.sum().reshape(1)
         comm.all_reduce_(total)
-        return (grad, steps_next), float(total.item()), steps
+        return (grad, steps_next, pack), float(total.item()), steps
 
     def fit(
         self,

@@ -2390,6 +2390,329 @@ __global__ void __launch_bounds__(BLOCK) kw_grad_t(
   }
 }
 
+// ------------------------------------------------- sparse-adaptive sweep
+//
+// Active-column realization of the sweep for the converged phase
+// (docs/sparse_sweep_design.md; the reference's v3 sparse-F insight,
+// codes/bigclamv3-7.scala:86).  All clamps produce EXACT zeros, so the
+// sparsity is exact, not thresholded: measured converged density at the
+// headline configs is 0.4-4% (profiles/r02_convergence.md), far below
+// the 16-candidate dense streaming work.
+//
+//   KAF kaf_support_t  per-row support compaction of F: counts, then
+//                      (idx, fp32 val) lists for rows with <= cap nnz.
+//   K1S k1s_grad_t     per routed node: edge dots walk the NEIGHBOR's
+//                      compact list (gathering fu elements from the
+//                      L1-hot row), scatter w*val into a dense LDS
+//                      gacc, then ONE dense K-scan emits the compact
+//                      gradient (k, g) for the active set
+//                      S_u = supp(gacc) ∪ supp(fu) plus llh_u and
+//                      gg_u = Σ_S (g² − sumF²)  (full ‖g‖² = gg_u + GG,
+//                      GG = Σ_K sumF², since g = −sumF off S).
+//   K2S k2s_ls_t       16-candidate Armijo scoring entirely on the
+//                      compact lists: candidates clamp(fu + s_j·g) are
+//                      supported on S_u exactly, so trial dots walk the
+//                      neighbor lists with a binary search into S_u.
+//   K3S k3s_commit_t   sparse projected commit (writes confined to S_u).
+//
+// Routing is exact and per node: bound_u = s_u + Σ_{v∈N(u)} s_v ≥ |S_u|
+// (host-side from KAF counts); nodes with bound ≤ cap take this path,
+// the rest the dense kernels — the two launches cover disjoint nodes.
+
+template <bool BF16>
+__device__ __forceinline__ float f_elem(const void* __restrict__ Fp, int ldF,
+                                        int row, int k) {
+  if (BF16) {
+    const unsigned short* p = reinterpret_cast<const unsigned short*>(Fp);
+    return __uint_as_float((u32)p[(size_t)row * ldF + k] << 16);
+  }
+  return reinterpret_cast<const float*>(Fp)[(size_t)row * ldF + k];
+}
+
+__device__ __forceinline__ unsigned short pack1_bf16_rne(float v) {
+  u32 l = __float_as_uint(v);
+  l += 0x7fffu + ((l >> 16) & 1u);
+  return (unsigned short)(l >> 16);
+}
+
+template <bool BF16, bool FILL>
+__global__ void __launch_bounds__(BLOCK) kaf_support_t(
+    const void* __restrict__ Fp, int n_rows, int K, int ldF, int cap,
+    const long long* __restrict__ soffset, int* __restrict__ scount,
+    int* __restrict__ sidx, float* __restrict__ sval) {
+  __shared__ int scan[BLOCK];
+  const int r = blockIdx.x;
+  if (r >= n_rows) return;
+  if (FILL && scount[r] > cap) return;  // unrouted row: no list needed
+  const int chunk = (K + BLOCK - 1) / BLOCK;
+  const int c0 = min((int)threadIdx.x * chunk, K);
+  const int c1 = min(c0 + chunk, K);
+  int cnt = 0;
+  for (int c = c0; c < c1; ++c)
+    cnt += f_elem<BF16>(Fp, ldF, r, c) != 0.f;
+  scan[threadIdx.x] = cnt;
+  __syncthreads();
+#pragma unroll
+  for (int off = 1; off < BLOCK; off <<= 1) {
+    const int v = (threadIdx.x >= off) ? scan[threadIdx.x - off] : 0;
+    __syncthreads();
+    scan[threadIdx.x] += v;
+    __syncthreads();
+  }
+  if (!FILL) {
+    if (threadIdx.x == BLOCK - 1) scount[r] = scan[BLOCK - 1];
+    return;
+  }
+  long long w = soffset[r] + (scan[threadIdx.x] - cnt);
+  for (int c = c0; c < c1; ++c) {
+    const float f = f_elem<BF16>(Fp, ldF, r, c);
+    if (f != 0.f) {
+      sidx[w] = c;
+      sval[w] = f;
+      ++w;
+    }
+  }
+}
+
+template <bool BF16>
+__global__ void __launch_bounds__(BLOCK) k1s_grad_t(
+    const void* __restrict__ Fp, int ldF,
+    const long long* __restrict__ indptr, const int* __restrict__ indices,
+    const float* __restrict__ sumF, const int* __restrict__ order,
+    const long long* __restrict__ soffset, const int* __restrict__ sidx,
+    const float* __restrict__ sval, const int* __restrict__ scount,
+    const long long* __restrict__ goffset, int* __restrict__ gidx,
+    float* __restrict__ gval, int* __restrict__ gcount,
+    double* __restrict__ llh, float* __restrict__ gg, int K, float min_p,
+    float max_p) {
+  const int u = order[blockIdx.x];
+  const long long e0 = indptr[u];
+  const long long e1 = indptr[u + 1];
+  const int tid = threadIdx.x;
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  float* gacc = reinterpret_cast<float*>(smem);  // K floats
+  __shared__ float red[NWAVE];
+  __shared__ int scan[BLOCK];
+  for (int k = tid * 4; k < K; k += BLOCK * 4)
+    *reinterpret_cast<float4*>(gacc + k) = float4{0.f, 0.f, 0.f, 0.f};
+  __syncthreads();
+
+  double llh_acc = 0.0;  // thread 0
+  for (long long e = e0; e < e1; ++e) {
+    const int v = indices[e];
+    const long long off = soffset[v];
+    const int sv = scount[v];
+    float part = 0.f;
+    for (int j = tid; j < sv; j += BLOCK)
+      part += f_elem<BF16>(Fp, ldF, u, sidx[off + j]) * sval[off + j];
+    const float x = block_allreduce_sum(part, red);  // syncs the block
+    const float p = clamp_p(__expf(-x), min_p, max_p);
+    const float w = 1.f / (1.f - p);
+    if (tid == 0) llh_acc += (double)log1pf(-p) + (double)x;
+    // threads write DISTINCT gacc slots within one edge; the next
+    // edge's block_allreduce orders cross-edge writes
+    for (int j = tid; j < sv; j += BLOCK)
+      gacc[sidx[off + j]] += w * sval[off + j];
+  }
+  __syncthreads();
+
+  // dense K-scan: emit compact (k, g) for S_u, node terms, gg
+  const int chunk = (K + BLOCK - 1) / BLOCK;
+  const int c0 = min(tid * chunk, K);
+  const int c1 = min(c0 + chunk, K);
+  int cnt = 0;
+  float gg_p = 0.f, fs_p = 0.f, ff_p = 0.f;
+  for (int c = c0; c < c1; ++c) {
+    const float a = gacc[c];
+    const float f = f_elem<BF16>(Fp, ldF, u, c);
+    const float s = sumF[c];
+    fs_p = fmaf(f, s, fs_p);
+    ff_p = fmaf(f, f, ff_p);
+    if (a != 0.f || f != 0.f) {
+      const float g = a - s + f;
+      gg_p += g * g - s * s;
+      ++cnt;
+    }
+  }
+  scan[tid] = cnt;
+  __syncthreads();
+#pragma unroll
+  for (int off = 1; off < BLOCK; off <<= 1) {
+    const int v = (tid >= off) ? scan[tid - off] : 0;
+    __syncthreads();
+    scan[tid] += v;
+    __syncthreads();
+  }
+  long long w = goffset[blockIdx.x] + (scan[tid] - cnt);
+  for (int c = c0; c < c1; ++c) {
+    const float a = gacc[c];
+    const float f = f_elem<BF16>(Fp, ldF, u, c);
+    if (a != 0.f || f != 0.f) {
+      gidx[w] = c;
+      gval[w] = a - sumF[c] + f;
+      ++w;
+    }
+  }
+  const float ggt = block_allreduce_sum(gg_p, red);
+  const float fst = block_allreduce_sum(fs_p, red);
+  const float fft = block_allreduce_sum(ff_p, red);
+  if (tid == 0) {
+    gcount[blockIdx.x] = scan[BLOCK - 1];
+    gg[u] = ggt;
+    llh[u] = llh_acc + (double)(-fst) + (double)fft;
+  }
+}
+
+template <bool BF16>
+__global__ void __launch_bounds__(BLOCK) k2s_ls_t(
+    const void* __restrict__ Fp, int ldF,
+    const long long* __restrict__ indptr, const int* __restrict__ indices,
+    const float* __restrict__ sumF, const int* __restrict__ order,
+    const long long* __restrict__ soffset, const int* __restrict__ sidx,
+    const float* __restrict__ sval, const int* __restrict__ scount,
+    const long long* __restrict__ goffset, const int* __restrict__ gidx,
+    const float* __restrict__ gval, const int* __restrict__ gcount,
+    const double* __restrict__ llh, const float* __restrict__ gg, float GG,
+    const float* __restrict__ ladder, float* __restrict__ best, int n_ladder,
+    float alpha, float min_p, float max_p, float min_f, float max_f) {
+  const int u = order[blockIdx.x];
+  const long long e0 = indptr[u];
+  const long long e1 = indptr[u + 1];
+  const int tid = threadIdx.x;
+  const int lane = tid & (WAVE - 1);
+  const int wid = tid >> 6;
+  const int ns = gcount[blockIdx.x];
+  const long long go = goffset[blockIdx.x];
+
+  __shared__ float s_lad[MAX_LS];
+  __shared__ double acc_llh[NWAVE][MAX_LS];
+  __shared__ float acc_nt[NWAVE][MAX_LS];
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  // layout: k_S[stride] | g_S[stride] | fu_S[stride], stride = ns rounded
+  // to 4; the host sizes the dynamic LDS for the launch's max gcount
+  int* k_S = reinterpret_cast<int*>(smem);
+
+  if (tid < MAX_LS) s_lad[tid] = (tid < n_ladder) ? ladder[tid] : 0.f;
+  const int stride = (ns + 3) & ~3;
+  float* gS = reinterpret_cast<float*>(k_S + stride);
+  float* fuS = gS + stride;
+  for (int i = tid; i < ns; i += BLOCK) {
+    const int k = gidx[go + i];
+    k_S[i] = k;
+    gS[i] = gval[go + i];
+    fuS[i] = f_elem<BF16>(Fp, ldF, u, k);
+  }
+  __syncthreads();
+
+  float s[MAX_LS];
+#pragma unroll
+  for (int j = 0; j < MAX_LS; ++j) s[j] = s_lad[j];
+  const int jmine = (((lane >> 5) & 1) << 3) | (((lane >> 4) & 1) << 2) |
+                    (((lane >> 3) & 1) << 1) | ((lane >> 2) & 1);
+  double llh_mine = 0.0;
+
+  for (long long e = e0 + wid; e < e1; e += NWAVE) {
+    const int v = indices[e];
+    const long long off = soffset[v];
+    const int sv = scount[v];
+    float acc[MAX_LS];
+#pragma unroll
+    for (int j = 0; j < MAX_LS; ++j) acc[j] = 0.f;
+    for (int j0 = lane; j0 < sv; j0 += WAVE) {
+      const int k = sidx[off + j0];
+      // binary search k in k_S (sorted ascending)
+      int lo = 0, hi = ns;
+      while (lo < hi) {
+        const int mid = (lo + hi) >> 1;
+        if (k_S[mid] < k)
+          lo = mid + 1;
+        else
+          hi = mid;
+      }
+      if (lo < ns && k_S[lo] == k) {
+        const float fv = sval[off + j0];
+        const float fu = fuS[lo];
+        const float gk = gS[lo];
+#pragma unroll
+        for (int j = 0; j < MAX_LS; ++j) {
+          const float c = __builtin_amdgcn_fmed3f(
+              fmaf(s[j], gk, fu), min_f, max_f);
+          acc[j] = fmaf(c, fv, acc[j]);
+        }
+      }
+    }
+    wave_reduce16(acc, lane);
+    if ((lane & 3) == 0) {
+      const float x = acc[0];
+      const float p = clamp_p(__expf(-x), min_p, max_p);
+      llh_mine += (double)log1pf(-p) + (double)x;
+    }
+  }
+
+  // node terms over S_u: c_j · (fu − sumF)
+  float accn[MAX_LS];
+#pragma unroll
+  for (int j = 0; j < MAX_LS; ++j) accn[j] = 0.f;
+  for (int i = tid; i < ns; i += BLOCK) {
+    const float fu = fuS[i];
+    const float gk = gS[i];
+    const float d = fu - sumF[k_S[i]];
+#pragma unroll
+    for (int j = 0; j < MAX_LS; ++j) {
+      const float c =
+          __builtin_amdgcn_fmed3f(fmaf(s[j], gk, fu), min_f, max_f);
+      accn[j] = fmaf(c, d, accn[j]);
+    }
+  }
+  wave_reduce16(accn, lane);
+  if ((lane & 3) == 0) {
+    acc_nt[wid][jmine] = accn[0];
+    acc_llh[wid][jmine] = llh_mine;
+  }
+  __syncthreads();
+
+  if (wid == 0) {
+    bool ok = false;
+    if (lane < MAX_LS) {
+      double trial = 0.0;
+#pragma unroll
+      for (int wv = 0; wv < NWAVE; ++wv)
+        trial += acc_llh[wv][lane] + (double)acc_nt[wv][lane];
+      const float ggfull = gg[u] + GG;
+      ok = (lane < n_ladder) &&
+           (trial >= llh[u] + (double)(alpha * s_lad[lane] * ggfull));
+    }
+    const unsigned long long bal = __ballot(ok);
+    if (lane == 0)
+      best[u] = bal ? s_lad[__ffsll((unsigned long long)bal) - 1] : 0.f;
+  }
+}
+
+template <bool BF16>
+__global__ void __launch_bounds__(BLOCK) k3s_commit_t(
+    void* __restrict__ Fp, int ldF, const int* __restrict__ order,
+    const long long* __restrict__ goffset, const int* __restrict__ gidx,
+    const float* __restrict__ gval, const int* __restrict__ gcount,
+    const float* __restrict__ best, float min_f, float max_f) {
+  const int u = order[blockIdx.x];
+  const float s = best[u];
+  if (s <= 0.f) return;
+  const int ns = gcount[blockIdx.x];
+  const long long go = goffset[blockIdx.x];
+  for (int i = threadIdx.x; i < ns; i += BLOCK) {
+    const int k = gidx[go + i];
+    const float f = f_elem<BF16>(Fp, ldF, u, k);
+    const float nf =
+        __builtin_amdgcn_fmed3f(fmaf(s, gval[go + i], f), min_f, max_f);
+    if (BF16) {
+      reinterpret_cast<unsigned short*>(Fp)[(size_t)u * ldF + k] =
+          pack1_bf16_rne(nf);
+    } else {
+      reinterpret_cast<float*>(Fp)[(size_t)u * ldF + k] = nf;
+    }
+  }
+}
+
 // ------------------------------------------------------------------- K7
 //
 // Community extraction (codes/Bigclamv2.scala:223-230): node u belongs to
@@ -2855,6 +3178,107 @@ extern "C" void launch_k1_chunked(const void* F, int bf16,
     hipLaunchKernelGGL((kw_grad_t<false>), dim3(n_blocks), dim3(BLOCK),
                        lds_w, stream, F, indptr, indices, sumF, order, xbuf,
                        grad, llh, K, ch, min_p, max_p);
+  }
+  HIP_CHECK(hipGetLastError());
+}
+
+extern "C" void launch_kaf(const void* F, int bf16, int n_rows, int K,
+                           int cap, const long long* soffset, int* scount,
+                           int* sidx, float* sval, int fill,
+                           hipStream_t stream) {
+  if (n_rows == 0) return;
+#define KAF_CASE(B, FI)                                                     \
+  hipLaunchKernelGGL((kaf_support_t<B, FI>), dim3(n_rows), dim3(BLOCK), 0,  \
+                     stream, F, n_rows, K, K, cap, soffset, scount, sidx,   \
+                     sval)
+  if (bf16) {
+    if (fill)
+      KAF_CASE(true, true);
+    else
+      KAF_CASE(true, false);
+  } else {
+    if (fill)
+      KAF_CASE(false, true);
+    else
+      KAF_CASE(false, false);
+  }
+#undef KAF_CASE
+  HIP_CHECK(hipGetLastError());
+}
+
+extern "C" void launch_k1s(const void* F, int bf16,
+                           const long long* indptr, const int* indices,
+                           const float* sumF, const int* order, int n_blocks,
+                           const long long* soffset, const int* sidx,
+                           const float* sval, const int* scount,
+                           const long long* goffset, int* gidx, float* gval,
+                           int* gcount, double* llh, float* gg, int K,
+                           float min_p, float max_p, hipStream_t stream) {
+  if (n_blocks == 0) return;
+  const size_t lds = (size_t)K * 4;
+  if (bf16) {
+    allow_large_lds((const void*)&k1s_grad_t<true>, lds);
+    hipLaunchKernelGGL((k1s_grad_t<true>), dim3(n_blocks), dim3(BLOCK), lds,
+                       stream, F, K, indptr, indices, sumF, order, soffset,
+                       sidx, sval, scount, goffset, gidx, gval, gcount, llh,
+                       gg, K, min_p, max_p);
+  } else {
+    allow_large_lds((const void*)&k1s_grad_t<false>, lds);
+    hipLaunchKernelGGL((k1s_grad_t<false>), dim3(n_blocks), dim3(BLOCK), lds,
+                       stream, F, K, indptr, indices, sumF, order, soffset,
+                       sidx, sval, scount, goffset, gidx, gval, gcount, llh,
+                       gg, K, min_p, max_p);
+  }
+  HIP_CHECK(hipGetLastError());
+}
+
+extern "C" void launch_k2s(const void* F, int bf16,
+                           const long long* indptr, const int* indices,
+                           const float* sumF, const int* order, int n_blocks,
+                           const long long* soffset, const int* sidx,
+                           const float* sval, const int* scount,
+                           const long long* goffset, const int* gidx,
+                           const float* gval, const int* gcount,
+                           const double* llh, const float* gg, float GG,
+                           const float* ladder, float* best, int n_ladder,
+                           int max_count, int K, float alpha, float min_p,
+                           float max_p, float min_f, float max_f,
+                           hipStream_t stream) {
+  if (n_blocks == 0) return;
+  if (n_ladder > 16) throw std::runtime_error("ladder length > 16 unsupported");
+  const size_t lds = 12 * ((size_t)max_count + 4);
+  if (bf16) {
+    allow_large_lds((const void*)&k2s_ls_t<true>, lds);
+    hipLaunchKernelGGL((k2s_ls_t<true>), dim3(n_blocks), dim3(BLOCK), lds,
+                       stream, F, K, indptr, indices, sumF, order, soffset,
+                       sidx, sval, scount, goffset, gidx, gval, gcount, llh,
+                       gg, GG, ladder, best, n_ladder, alpha, min_p, max_p,
+                       min_f, max_f);
+  } else {
+    allow_large_lds((const void*)&k2s_ls_t<false>, lds);
+    hipLaunchKernelGGL((k2s_ls_t<false>), dim3(n_blocks), dim3(BLOCK), lds,
+                       stream, F, K, indptr, indices, sumF, order, soffset,
+                       sidx, sval, scount, goffset, gidx, gval, gcount, llh,
+                       gg, GG, ladder, best, n_ladder, alpha, min_p, max_p,
+                       min_f, max_f);
+  }
+  HIP_CHECK(hipGetLastError());
+}
+
+extern "C" void launch_k3s(void* F, int bf16, const int* order, int n_blocks,
+                           const long long* goffset, const int* gidx,
+                           const float* gval, const int* gcount,
+                           const float* best, int K, float min_f,
+                           float max_f, hipStream_t stream) {
+  if (n_blocks == 0) return;
+  if (bf16) {
+    hipLaunchKernelGGL((k3s_commit_t<true>), dim3(n_blocks), dim3(BLOCK), 0,
+                       stream, F, K, order, goffset, gidx, gval, gcount,
+                       best, min_f, max_f);
+  } else {
+    hipLaunchKernelGGL((k3s_commit_t<false>), dim3(n_blocks), dim3(BLOCK), 0,
+                       stream, F, K, order, goffset, gidx, gval, gcount,
+                       best, min_f, max_f);
   }
   HIP_CHECK(hipGetLastError());
 }

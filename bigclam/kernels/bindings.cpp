@@ -337,6 +337,114 @@ void edge_grad_llh_chunked(torch::Tensor F, torch::Tensor indptr,
                     K, (float)min_p, (float)max_p, current_stream());
 }
 
+// Sparse-adaptive sweep (KAF/K1S/K2S/K3S — docs/sparse_sweep_design.md).
+extern "C" void launch_kaf(const void*, int, int, int, int,
+                           const long long*, int*, int*, float*, int,
+                           hipStream_t);
+extern "C" void launch_k1s(const void*, int, const long long*, const int*,
+                           const float*, const int*, int, const long long*,
+                           const int*, const float*, const int*,
+                           const long long*, int*, float*, int*, double*,
+                           float*, int, float, float, hipStream_t);
+extern "C" void launch_k2s(const void*, int, const long long*, const int*,
+                           const float*, const int*, int, const long long*,
+                           const int*, const float*, const int*,
+                           const long long*, const int*, const float*,
+                           const int*, const double*, const float*, float,
+                           const float*, float*, int, int, int, float, float,
+                           float, float, float, hipStream_t);
+extern "C" void launch_k3s(void*, int, const int*, int, const long long*,
+                           const int*, const float*, const int*,
+                           const float*, int, float, float, hipStream_t);
+
+static const long long* i64p(const torch::Tensor& t) {
+  return reinterpret_cast<const long long*>(t.data_ptr<int64_t>());
+}
+
+void sparse_support(torch::Tensor F, torch::Tensor soffset,
+                    torch::Tensor scount, torch::Tensor sidx,
+                    torch::Tensor sval, int64_t cap, bool fill) {
+  CHECK_F(F);
+  CHECK_IN(soffset, torch::kInt64);
+  CHECK_IN(scount, torch::kInt32);
+  CHECK_IN(sidx, torch::kInt32);
+  CHECK_IN(sval, torch::kFloat32);
+  const int n_rows = (int)F.size(0);
+  const int K = (int)F.size(1);
+  TORCH_CHECK(scount.size(0) == n_rows && soffset.size(0) == n_rows);
+  launch_kaf(F.data_ptr(), is_bf16(F) ? 1 : 0, n_rows, K, (int)cap,
+             i64p(soffset), scount.data_ptr<int>(), sidx.data_ptr<int>(),
+             sval.data_ptr<float>(), fill ? 1 : 0, current_stream());
+}
+
+void sparse_grad(torch::Tensor F, torch::Tensor indptr, torch::Tensor indices,
+                 torch::Tensor sumF, torch::Tensor order,
+                 torch::Tensor soffset, torch::Tensor sidx,
+                 torch::Tensor sval, torch::Tensor scount,
+                 torch::Tensor goffset, torch::Tensor gidx,
+                 torch::Tensor gval, torch::Tensor gcount, torch::Tensor llh,
+                 torch::Tensor gg, double min_p, double max_p) {
+  CHECK_F(F);
+  CHECK_IN(indptr, torch::kInt64);
+  CHECK_IN(indices, torch::kInt32);
+  CHECK_IN(sumF, torch::kFloat32);
+  CHECK_IN(order, torch::kInt32);
+  CHECK_IN(goffset, torch::kInt64);
+  CHECK_IN(gidx, torch::kInt32);
+  CHECK_IN(gval, torch::kFloat32);
+  CHECK_IN(gcount, torch::kInt32);
+  CHECK_IN(llh, torch::kFloat64);
+  CHECK_IN(gg, torch::kFloat32);
+  const int n_blocks = (int)order.size(0);
+  TORCH_CHECK(goffset.size(0) >= n_blocks && gcount.size(0) >= n_blocks);
+  launch_k1s(F.data_ptr(), is_bf16(F) ? 1 : 0, i64p(indptr),
+             indices.data_ptr<int>(), sumF.data_ptr<float>(),
+             order.data_ptr<int>(), n_blocks, i64p(soffset),
+             sidx.data_ptr<int>(), sval.data_ptr<float>(),
+             scount.data_ptr<int>(), i64p(goffset), gidx.data_ptr<int>(),
+             gval.data_ptr<float>(), gcount.data_ptr<int>(),
+             llh.data_ptr<double>(), gg.data_ptr<float>(), (int)F.size(1),
+             (float)min_p, (float)max_p, current_stream());
+}
+
+void sparse_ls(torch::Tensor F, torch::Tensor indptr, torch::Tensor indices,
+               torch::Tensor sumF, torch::Tensor order, torch::Tensor soffset,
+               torch::Tensor sidx, torch::Tensor sval, torch::Tensor scount,
+               torch::Tensor goffset, torch::Tensor gidx, torch::Tensor gval,
+               torch::Tensor gcount, torch::Tensor llh, torch::Tensor gg,
+               double GG, torch::Tensor ladder, torch::Tensor best,
+               int64_t max_count, double alpha, double min_p, double max_p,
+               double min_f, double max_f) {
+  CHECK_F(F);
+  CHECK_IN(ladder, torch::kFloat32);
+  CHECK_IN(best, torch::kFloat32);
+  const int n_blocks = (int)order.size(0);
+  launch_k2s(F.data_ptr(), is_bf16(F) ? 1 : 0, i64p(indptr),
+             indices.data_ptr<int>(), sumF.data_ptr<float>(),
+             order.data_ptr<int>(), n_blocks, i64p(soffset),
+             sidx.data_ptr<int>(), sval.data_ptr<float>(),
+             scount.data_ptr<int>(), i64p(goffset), gidx.data_ptr<int>(),
+             gval.data_ptr<float>(), gcount.data_ptr<int>(),
+             llh.data_ptr<double>(), gg.data_ptr<float>(), (float)GG,
+             ladder.data_ptr<float>(), best.data_ptr<float>(),
+             (int)ladder.size(0), (int)max_count, (int)F.size(1),
+             (float)alpha, (float)min_p, (float)max_p, (float)min_f,
+             (float)max_f, current_stream());
+}
+
+void sparse_commit(torch::Tensor F, torch::Tensor order,
+                   torch::Tensor goffset, torch::Tensor gidx,
+                   torch::Tensor gval, torch::Tensor gcount,
+                   torch::Tensor best, double min_f, double max_f) {
+  CHECK_F(F);
+  CHECK_IN(best, torch::kFloat32);
+  launch_k3s(F.data_ptr(), is_bf16(F) ? 1 : 0, order.data_ptr<int>(),
+             (int)order.size(0), i64p(goffset), gidx.data_ptr<int>(),
+             gval.data_ptr<float>(), gcount.data_ptr<int>(),
+             best.data_ptr<float>(), (int)F.size(1), (float)min_f,
+             (float)max_f, current_stream());
+}
+
 // K7 community extraction: two deterministic passes (count, then fill
 // after a host/torch prefix-sum) — see k7_membership in the .hip file.
 void extract_count(torch::Tensor F_local, int64_t k_true, double delta,
@@ -397,6 +505,14 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "(MFMA tiles for the high-degree prefix)");
   m.def("apply_step_colsum", &apply_step_colsum,
         "K3+colsum fused (bf16): commit F and emit per-stripe column sums");
+  m.def("sparse_support", &sparse_support,
+        "KAF: per-row support compaction of F (count or fill pass)");
+  m.def("sparse_grad", &sparse_grad,
+        "K1S: compact gradient + llh + gg for routed nodes");
+  m.def("sparse_ls", &sparse_ls,
+        "K2S: 16-candidate Armijo on compact active sets");
+  m.def("sparse_commit", &sparse_commit,
+        "K3S: sparse projected commit confined to the active set");
   m.def("edge_grad_llh_chunked", &edge_grad_llh_chunked,
         "K1 large-K: KD per-edge dots + KW chunked weighted accumulate");
   m.def("extract_count", &extract_count,

@@ -218,6 +218,61 @@ def conductance_full_graph(graph, device) -> "torch.Tensor":
     return cond
 
 
+def sparse_sweep_part(
+    F: torch.Tensor,
+    indptr: torch.Tensor,
+    indices: torch.Tensor,
+    sumF: torch.Tensor,
+    order_sparse: torch.Tensor,
+    soffset: torch.Tensor,
+    sidx: torch.Tensor,
+    sval: torch.Tensor,
+    scount: torch.Tensor,
+    goffset: torch.Tensor,
+    gpool_size: int,
+    cap: int,
+    llh_out: torch.Tensor,
+    best_out: torch.Tensor,
+    cfg: BigClamConfig,
+):
+    """K1S + K2S for the routed (sparse) nodes: compact gradient pools,
+    llh/gg per node, Armijo best step.  Writes llh_out/best_out at the
+    routed node positions; returns the commit pack for K3S."""
+    ext = ensure_loaded()
+    dev = F.device
+    n_s = int(order_sparse.numel())
+    gidx = torch.empty(gpool_size, device=dev, dtype=torch.int32)
+    gval = torch.empty(gpool_size, device=dev, dtype=torch.float32)
+    gcount = torch.empty(n_s, device=dev, dtype=torch.int32)
+    gg = torch.empty(int(indptr.numel()) - 1, device=dev, dtype=torch.float32)
+    ext.sparse_grad(
+        F, indptr, indices, sumF, order_sparse, soffset, sidx, sval, scount,
+        goffset, gidx, gval, gcount, llh_out, gg, cfg.min_p, cfg.max_p,
+    )
+    GG = float((sumF.float() ** 2).sum().item())
+    ext.sparse_ls(
+        F, indptr, indices, sumF, order_sparse, soffset, sidx, sval, scount,
+        goffset, gidx, gval, gcount, llh_out, gg, GG, _ladder(cfg, dev),
+        best_out, cap, cfg.alpha, cfg.min_p, cfg.max_p, cfg.min_f, cfg.max_f,
+    )
+    return {
+        "order": order_sparse,
+        "goffset": goffset,
+        "gidx": gidx,
+        "gval": gval,
+        "gcount": gcount,
+    }
+
+
+def sparse_commit(F_local: torch.Tensor, pack: dict, best: torch.Tensor,
+                  cfg: BigClamConfig):
+    """K3S: projected commit confined to each routed node's active set."""
+    ensure_loaded().sparse_commit(
+        F_local, pack["order"], pack["goffset"], pack["gidx"], pack["gval"],
+        pack["gcount"], best, cfg.min_f, cfg.max_f,
+    )
+
+
 def extract_membership(
     F_local: torch.Tensor, k_true: int, delta: float
 ) -> Tuple[torch.Tensor, torch.Tensor]:

@@ -528,3 +528,112 @@ def test_bf16_k_above_26000_runs():
     steps = st.linesearch(grad, llh)
     st.apply_step(grad, steps)
     assert torch.isfinite(st.sumF).all()
+
+
+def _converged_state(dtype="fp32", k=512, seed=31):
+    """A partially-converged fit whose rows carry exact zeros (sparse
+    routing engages)."""
+    g = rmat_graph(10, 6.0, seed=seed)
+    cfg = BigClamConfig(k=k, device="cuda", dtype=dtype, seed=9,
+                        max_sweeps=15, tol=0.0)
+    tr = Trainer(g, cfg, rank=0, world_size=1, device=torch.device("cuda"))
+    import os as _os
+
+    _os.environ["BIGCLAM_SPARSE"] = "0"  # converge on the dense path
+    try:
+        tr.fit(init="seed")
+    finally:
+        _os.environ.pop("BIGCLAM_SPARSE", None)
+    return g, cfg, tr
+
+
+@pytest.mark.parametrize("dtype", ["fp32", "bf16"])
+def test_sparse_sweep_matches_dense_gpu(dtype, monkeypatch):
+    """Hybrid K1S/K2S/K3S sweep == the dense fused sweep on the same
+    partially-converged state: llh equal, step picks agree above the
+    noise rung (ties checked by the fp64 margin oracle), committed F
+    rows close."""
+    g, cfg, tr = _converged_state(dtype=dtype)
+    st = tr.state
+    nnz = float((st.F_local_k != 0).float().mean().item())
+    assert nnz < 0.25, f"fixture not sparse enough: {nnz}"
+    F0 = st.F.clone()
+    sumF0 = st.sumF.clone()
+
+    # dense reference sweep
+    monkeypatch.setenv("BIGCLAM_SPARSE", "0")
+    grad_d, llh_d, best_d, pack = st.grad_ls_auto(None)
+    assert pack is None
+    st.apply_commit(grad_d, best_d, None)
+    F_dense = st.F_local_k.float().clone()
+
+    # restore and run the hybrid sweep
+    st.F.copy_(F0)
+    st.sumF.copy_(sumF0)
+    monkeypatch.setenv("BIGCLAM_SPARSE", "1")
+    grad_s, llh_s, best_s, pack = st.grad_ls_auto(None)
+    assert pack is not None and int(pack["order"].numel()) > 0, "no routing"
+    # llh per node equal (fp32 accumulation-order tolerance)
+    torch.testing.assert_close(llh_s.sum(), llh_d.sum(), rtol=2e-5, atol=1.0)
+    # compact grad matches the dense grad at its active positions for a
+    # few routed nodes
+    go = pack["goffset"].cpu().numpy()
+    gc = pack["gcount"].cpu().numpy()
+    gi = pack["gidx"].cpu().numpy()
+    gv = pack["gval"].cpu().numpy()
+    order_s = pack["order"].cpu().numpy()
+    gd = grad_d.cpu().numpy()
+    atol = 2e-3 if dtype == "fp32" else 2e-2
+    for b in range(0, min(len(order_s), 40), 7):
+        u = order_s[b]
+        ks = gi[go[b] : go[b] + gc[b]]
+        np.testing.assert_allclose(
+            gv[go[b] : go[b] + gc[b]], gd[u, ks], rtol=2e-3, atol=atol
+        )
+    # step picks: agreement or fp64-margin tie above the noise rung
+    bs = best_s.cpu().numpy()
+    bd = best_d.cpu().numpy()
+    disagree = np.flatnonzero(bs != bd)
+    noise = cfg.beta ** 5
+    import oracle as _oracle
+
+    Fh = F0[: st.n_local, : cfg.k].float().cpu().numpy()
+    checked = 0
+    for u in disagree:
+        if max(bs[u], bd[u]) <= noise:
+            continue
+        m = _oracle.armijo_margin_f64(
+            g, Fh, gd[u, : cfg.k], int(u), max(bs[u], bd[u]), cfg
+        )
+        assert abs(m) < 1e-4 * max(1.0, abs(float(llh_d[u]))), (u, bs[u],
+                                                                bd[u], m)
+        checked += 1
+        if checked > 50:
+            break
+    # committed rows close where picks agree
+    st.apply_commit(grad_s, best_s, pack)
+    F_sparse = st.F_local_k.float()
+    rows = torch.from_numpy((bs == bd).nonzero()[0]).cuda()
+    torch.testing.assert_close(
+        F_sparse[rows], F_dense.cuda()[rows], rtol=2e-3, atol=atol
+    )
+
+
+def test_sparse_fit_trajectory_matches_dense(monkeypatch):
+    """Full fits (adaptive sparse vs forced dense) track each other's
+    LLH trajectory on a seed-init fit."""
+    g = rmat_graph(10, 6.0, seed=33)
+
+    def fit(env):
+        monkeypatch.setenv("BIGCLAM_SPARSE", env)
+        cfg = BigClamConfig(k=256, device="cuda", seed=4, max_sweeps=20,
+                            tol=0.0)
+        tr = Trainer(g, cfg, rank=0, world_size=1,
+                     device=torch.device("cuda"))
+        res = tr.fit(init="seed")
+        return res.llh_history
+
+    h_dense = fit("0")
+    h_sparse = fit("1")
+    for a, b in zip(h_dense, h_sparse):
+        assert abs(a - b) < 2e-4 * max(1.0, abs(a)), (h_dense, h_sparse)
