@@ -77,6 +77,25 @@ def _load(args):
         # rmat:<scale>:<edge_factor> synthetic graph
         _, scale, ef = args.edgelist.split(":")
         return rmat_graph(int(scale), float(ef))
+    if args.edgelist.startswith("shaped:"):
+        # shaped:<nodes>:<edges> — power-law graph with exactly that
+        # shape (the bench/convergence configs, io/synthetic.py); named
+        # presets for the headline shapes:
+        #   shaped:amazon = 334863:925872, shaped:enron = 36692:183831,
+        #   shaped:youtube = 1134890:2987624
+        from .io import shaped_graph
+
+        presets = {
+            "amazon": (334863, 925872),
+            "enron": (36692, 183831),
+            "youtube": (1134890, 2987624),
+        }
+        parts = args.edgelist.split(":")[1:]
+        if len(parts) == 1 and parts[0] in presets:
+            n, e = presets[parts[0]]
+        else:
+            n, e = int(parts[0]), int(parts[1])
+        return shaped_graph(n, e, locality=0.7, seed=42)
     return load_graph(args.edgelist)
 
 

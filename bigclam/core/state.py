@@ -127,6 +127,7 @@ class ShardState:
             self.n_mfma = self.n_mfma_interior = self.n_mfma_boundary = 0
         self._halo_send: Optional[torch.Tensor] = None
         self._colsum_partials: Optional[torch.Tensor] = None
+        self._sp_soffset: Optional[torch.Tensor] = None
         # per-edge source row (torch reference path); built lazily on CPU
         self._edge_src: Optional[torch.Tensor] = None
 
@@ -339,12 +340,19 @@ class ShardState:
         dev = self.device
         n_rows = self.F.shape[0]
         cap = self.sparse_cap
+        # Single host sync per sweep (the boolean order split below);
+        # pools use a fixed per-row stride of `cap` so their sizes are
+        # shape-derived, not data-dependent.
         scount = torch.empty(n_rows, device=dev, dtype=torch.int32)
-        soffset = torch.zeros(n_rows, device=dev, dtype=torch.int64)
-        empty_i = torch.empty(0, device=dev, dtype=torch.int32)
-        empty_f = torch.empty(0, device=dev, dtype=torch.float32)
+        if self._sp_soffset is None or self._sp_soffset.numel() != n_rows:
+            self._sp_soffset = (
+                torch.arange(n_rows, device=dev, dtype=torch.int64) * cap
+            )
+        soffset = self._sp_soffset
+        sidx = ops._pool(dev, "sidx", n_rows * cap, torch.int32)
+        sval = ops._pool(dev, "sval", n_rows * cap, torch.float32)
         ops.ensure_loaded().sparse_support(
-            self.F, soffset, scount, empty_i, empty_f, cap, False
+            self.F, soffset, scount, sidx, sval, cap, False
         )
         sc = scount.to(torch.int64)
         # bound_u = s_u + sum_{v in N(u)} s_v  (cumsum segment trick)
@@ -355,26 +363,17 @@ class ShardState:
         )
         bound = sc[: self.n_local] + cs[self.indptr[1:]] - cs[self.indptr[:-1]]
         is_sparse = bound <= cap
-        n_s = int(is_sparse.sum().item())
+        om = is_sparse[self.order.long()]
+        order_s = self.order[om].contiguous()  # <- the one host sync
+        n_s = int(order_s.numel())
         if n_s < max(64, self.n_local // 20):
             g, l, b = self.fused_grad_ls_overlap(None)
             return g, l, b, None
-        # fill the support pools (rows with count <= cap only)
-        keep = torch.where(sc <= cap, sc, torch.zeros_like(sc))
-        soffset = torch.cumsum(keep, 0) - keep
-        pool = int(keep.sum().item())
-        sidx = torch.empty(pool, device=dev, dtype=torch.int32)
-        sval = torch.empty(pool, device=dev, dtype=torch.float32)
+        order_d = self.order[~om].contiguous()
         ops.ensure_loaded().sparse_support(
             self.F, soffset, scount, sidx, sval, cap, True
         )
-        # split the degree-descending launch order
-        om = is_sparse[self.order.long()]
-        order_s = self.order[om].contiguous()
-        order_d = self.order[~om].contiguous()
-        bnd_s = bound[order_s.long()]
-        goffset = torch.cumsum(bnd_s, 0) - bnd_s
-        gpool = int(bnd_s.sum().item())
+        goffset = torch.arange(n_s, device=dev, dtype=torch.int64) * cap
         n, kp = self.n_local, self.kp
         grad = torch.empty(n, kp, device=dev, dtype=torch.float32)
         llh = torch.empty(n, device=dev, dtype=torch.float64)
@@ -388,7 +387,7 @@ class ShardState:
             )
         pack = ops.sparse_sweep_part(
             self.F, self.indptr, self.indices, self.sumF, order_s,
-            soffset, sidx, sval, scount, goffset, gpool, cap, llh, best,
+            soffset, sidx, sval, scount, goffset, n_s * cap, cap, llh, best,
             self.cfg,
         )
         pack["best"] = best

@@ -2572,9 +2572,10 @@ __global__ void __launch_bounds__(BLOCK) k2s_ls_t(
     const float* __restrict__ sval, const int* __restrict__ scount,
     const long long* __restrict__ goffset, const int* __restrict__ gidx,
     const float* __restrict__ gval, const int* __restrict__ gcount,
-    const double* __restrict__ llh, const float* __restrict__ gg, float GG,
-    const float* __restrict__ ladder, float* __restrict__ best, int n_ladder,
-    float alpha, float min_p, float max_p, float min_f, float max_f) {
+    const double* __restrict__ llh, const float* __restrict__ gg,
+    const float* __restrict__ GGp, const float* __restrict__ ladder,
+    float* __restrict__ best, int n_ladder, float alpha, float min_p,
+    float max_p, float min_f, float max_f) {
   const int u = order[blockIdx.x];
   const long long e0 = indptr[u];
   const long long e1 = indptr[u + 1];
@@ -2678,7 +2679,7 @@ __global__ void __launch_bounds__(BLOCK) k2s_ls_t(
 #pragma unroll
       for (int wv = 0; wv < NWAVE; ++wv)
         trial += acc_llh[wv][lane] + (double)acc_nt[wv][lane];
-      const float ggfull = gg[u] + GG;
+      const float ggfull = gg[u] + GGp[0];
       ok = (lane < n_ladder) &&
            (trial >= llh[u] + (double)(alpha * s_lad[lane] * ggfull));
     }
@@ -3239,11 +3240,11 @@ extern "C" void launch_k2s(const void* F, int bf16,
                            const float* sval, const int* scount,
                            const long long* goffset, const int* gidx,
                            const float* gval, const int* gcount,
-                           const double* llh, const float* gg, float GG,
-                           const float* ladder, float* best, int n_ladder,
-                           int max_count, int K, float alpha, float min_p,
-                           float max_p, float min_f, float max_f,
-                           hipStream_t stream) {
+                           const double* llh, const float* gg,
+                           const float* GGp, const float* ladder,
+                           float* best, int n_ladder, int max_count, int K,
+                           float alpha, float min_p, float max_p,
+                           float min_f, float max_f, hipStream_t stream) {
   if (n_blocks == 0) return;
   if (n_ladder > 16) throw std::runtime_error("ladder length > 16 unsupported");
   const size_t lds = 12 * ((size_t)max_count + 4);
@@ -3252,14 +3253,14 @@ extern "C" void launch_k2s(const void* F, int bf16,
     hipLaunchKernelGGL((k2s_ls_t<true>), dim3(n_blocks), dim3(BLOCK), lds,
                        stream, F, K, indptr, indices, sumF, order, soffset,
                        sidx, sval, scount, goffset, gidx, gval, gcount, llh,
-                       gg, GG, ladder, best, n_ladder, alpha, min_p, max_p,
+                       gg, GGp, ladder, best, n_ladder, alpha, min_p, max_p,
                        min_f, max_f);
   } else {
     allow_large_lds((const void*)&k2s_ls_t<false>, lds);
     hipLaunchKernelGGL((k2s_ls_t<false>), dim3(n_blocks), dim3(BLOCK), lds,
                        stream, F, K, indptr, indices, sumF, order, soffset,
                        sidx, sval, scount, goffset, gidx, gval, gcount, llh,
-                       gg, GG, ladder, best, n_ladder, alpha, min_p, max_p,
+                       gg, GGp, ladder, best, n_ladder, alpha, min_p, max_p,
                        min_f, max_f);
   }
   HIP_CHECK(hipGetLastError());

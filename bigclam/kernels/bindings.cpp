@@ -350,9 +350,9 @@ extern "C" void launch_k2s(const void*, int, const long long*, const int*,
                            const float*, const int*, int, const long long*,
                            const int*, const float*, const int*,
                            const long long*, const int*, const float*,
-                           const int*, const double*, const float*, float,
-                           const float*, float*, int, int, int, float, float,
-                           float, float, float, hipStream_t);
+                           const int*, const double*, const float*,
+                           const float*, const float*, float*, int, int, int,
+                           float, float, float, float, float, hipStream_t);
 extern "C" void launch_k3s(void*, int, const int*, int, const long long*,
                            const int*, const float*, const int*,
                            const float*, int, float, float, hipStream_t);
@@ -412,7 +412,7 @@ void sparse_ls(torch::Tensor F, torch::Tensor indptr, torch::Tensor indices,
                torch::Tensor sidx, torch::Tensor sval, torch::Tensor scount,
                torch::Tensor goffset, torch::Tensor gidx, torch::Tensor gval,
                torch::Tensor gcount, torch::Tensor llh, torch::Tensor gg,
-               double GG, torch::Tensor ladder, torch::Tensor best,
+               torch::Tensor GG, torch::Tensor ladder, torch::Tensor best,
                int64_t max_count, double alpha, double min_p, double max_p,
                double min_f, double max_f) {
   CHECK_F(F);
@@ -425,7 +425,8 @@ void sparse_ls(torch::Tensor F, torch::Tensor indptr, torch::Tensor indices,
              sidx.data_ptr<int>(), sval.data_ptr<float>(),
              scount.data_ptr<int>(), i64p(goffset), gidx.data_ptr<int>(),
              gval.data_ptr<float>(), gcount.data_ptr<int>(),
-             llh.data_ptr<double>(), gg.data_ptr<float>(), (float)GG,
+             llh.data_ptr<double>(), gg.data_ptr<float>(),
+             GG.data_ptr<float>(),
              ladder.data_ptr<float>(), best.data_ptr<float>(),
              (int)ladder.size(0), (int)max_count, (int)F.size(1),
              (float)alpha, (float)min_p, (float)max_p, (float)min_f,

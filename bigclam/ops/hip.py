@@ -241,15 +241,15 @@ def sparse_sweep_part(
     ext = ensure_loaded()
     dev = F.device
     n_s = int(order_sparse.numel())
-    gidx = torch.empty(gpool_size, device=dev, dtype=torch.int32)
-    gval = torch.empty(gpool_size, device=dev, dtype=torch.float32)
+    gidx = _pool(dev, "gidx", gpool_size, torch.int32)
+    gval = _pool(dev, "gval", gpool_size, torch.float32)
     gcount = torch.empty(n_s, device=dev, dtype=torch.int32)
     gg = torch.empty(int(indptr.numel()) - 1, device=dev, dtype=torch.float32)
     ext.sparse_grad(
         F, indptr, indices, sumF, order_sparse, soffset, sidx, sval, scount,
         goffset, gidx, gval, gcount, llh_out, gg, cfg.min_p, cfg.max_p,
     )
-    GG = float((sumF.float() ** 2).sum().item())
+    GG = (sumF * sumF).sum().reshape(1)  # device scalar: no host sync
     ext.sparse_ls(
         F, indptr, indices, sumF, order_sparse, soffset, sidx, sval, scount,
         goffset, gidx, gval, gcount, llh_out, gg, GG, _ladder(cfg, dev),
@@ -262,6 +262,20 @@ def sparse_sweep_part(
         "gval": gval,
         "gcount": gcount,
     }
+
+
+_pools = {}
+
+
+def _pool(dev, name, size, dtype) -> torch.Tensor:
+    """Grow-only cached device buffer (avoids per-sweep allocator churn
+    and the size-sync it would force)."""
+    key = (str(dev), name)
+    t = _pools.get(key)
+    if t is None or t.numel() < size:
+        t = torch.empty(size, device=dev, dtype=dtype)
+        _pools[key] = t
+    return t
 
 
 def sparse_commit(F_local: torch.Tensor, pack: dict, best: torch.Tensor,
