@@ -387,8 +387,8 @@ class ShardState:
             )
         pack = ops.sparse_sweep_part(
             self.F, self.indptr, self.indices, self.sumF, order_s,
-            soffset, sidx, sval, scount, goffset, n_s * cap, cap, llh, best,
-            self.cfg,
+            soffset, sidx, sval, scount, cs, goffset, n_s * cap, cap, llh,
+            best, self.cfg,
         )
         pack["best"] = best
         # dense commit must skip sparse rows (their grad rows are unset)

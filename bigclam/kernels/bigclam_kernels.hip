@@ -2444,12 +2444,28 @@ __global__ void __launch_bounds__(BLOCK) kaf_support_t(
   const int r = blockIdx.x;
   if (r >= n_rows) return;
   if (FILL && scount[r] > cap) return;  // unrouted row: no list needed
-  const int chunk = (K + BLOCK - 1) / BLOCK;
+  // chunk rounded to 8 so the row reads vectorize (uint4 of 8 bf16 /
+  // two float4); K is padded (bf16: %8, fp32: %4 -> guard the last 4)
+  const int chunk = (((K + BLOCK - 1) / BLOCK) + 7) & ~7;
   const int c0 = min((int)threadIdx.x * chunk, K);
   const int c1 = min(c0 + chunk, K);
   int cnt = 0;
-  for (int c = c0; c < c1; ++c)
-    cnt += f_elem<BF16>(Fp, ldF, r, c) != 0.f;
+  if (BF16) {
+    const u32* row =
+        reinterpret_cast<const u32*>(Fp) + (size_t)r * (ldF / 2);
+    for (int c = c0; c < c1; c += 8) {
+      const f32x8 v = ld8bf(row + c / 2);
+      cnt += (v.a.x != 0.f) + (v.a.y != 0.f) + (v.a.z != 0.f) +
+             (v.a.w != 0.f) + (v.b.x != 0.f) + (v.b.y != 0.f) +
+             (v.b.z != 0.f) + (v.b.w != 0.f);
+    }
+  } else {
+    const float* row = reinterpret_cast<const float*>(Fp) + (size_t)r * ldF;
+    for (int c = c0; c < c1; c += 4) {
+      const float4 v = ld4(row + c);
+      cnt += (v.x != 0.f) + (v.y != 0.f) + (v.z != 0.f) + (v.w != 0.f);
+    }
+  }
   scan[threadIdx.x] = cnt;
   __syncthreads();
 #pragma unroll
@@ -2464,76 +2480,143 @@ __global__ void __launch_bounds__(BLOCK) kaf_support_t(
     return;
   }
   long long w = soffset[r] + (scan[threadIdx.x] - cnt);
-  for (int c = c0; c < c1; ++c) {
-    const float f = f_elem<BF16>(Fp, ldF, r, c);
-    if (f != 0.f) {
-      sidx[w] = c;
-      sval[w] = f;
-      ++w;
+  if (BF16) {
+    const u32* row =
+        reinterpret_cast<const u32*>(Fp) + (size_t)r * (ldF / 2);
+    for (int c = c0; c < c1; c += 8) {
+      const f32x8 v = ld8bf(row + c / 2);
+      const float el[8] = {v.a.x, v.a.y, v.a.z, v.a.w,
+                           v.b.x, v.b.y, v.b.z, v.b.w};
+#pragma unroll
+      for (int t = 0; t < 8; ++t)
+        if (el[t] != 0.f) {
+          sidx[w] = c + t;
+          sval[w] = el[t];
+          ++w;
+        }
+    }
+  } else {
+    const float* row = reinterpret_cast<const float*>(Fp) + (size_t)r * ldF;
+    for (int c = c0; c < c1; c += 4) {
+      const float4 v = ld4(row + c);
+      const float el[4] = {v.x, v.y, v.z, v.w};
+#pragma unroll
+      for (int t = 0; t < 4; ++t)
+        if (el[t] != 0.f) {
+          sidx[w] = c + t;
+          sval[w] = el[t];
+          ++w;
+        }
     }
   }
 }
 
+// KFS: fused K1S+K2S — ONE launch per sweep for the routed nodes.
+// The first cut ran K1S and K2S separately with a dense K-column scan
+// and global compact-pool round-trips; measured 8.0 + 5.6 ms/sweep at
+// the converged headline config (rocprofv3, r03).  This fusion:
+//   * stages ALL neighbor-list entries in LDS once (<= cap by routing;
+//     per-edge bases come FREE from the host's epos prefix sum — the
+//     same cumsum grad_ls_auto builds for the routing bounds);
+//   * tracks the active set in an LDS BITMAP (K/32 words), so the
+//     compact emission scans ~K/32 words instead of K columns;
+//   * keeps the compact (k, g, fu) set in LDS for the trial phase (and
+//     mirrors it to the global pools for K3S).
+// LDS: gacc[K] | nidx[cap] | nval[cap] | kS[cap] | gS[cap] | fuS[cap]
+//      | bmap[ceil(K/32)]  (~46 KB at K=5000, cap=K/4 -> 3 blocks/CU).
+
 template <bool BF16>
-__global__ void __launch_bounds__(BLOCK) k1s_grad_t(
-    const void* __restrict__ Fp, int ldF,
+__global__ void __launch_bounds__(BLOCK) kfs_sparse_t(
+    const void* __restrict__ Fp, int K,
     const long long* __restrict__ indptr, const int* __restrict__ indices,
     const float* __restrict__ sumF, const int* __restrict__ order,
     const long long* __restrict__ soffset, const int* __restrict__ sidx,
     const float* __restrict__ sval, const int* __restrict__ scount,
-    const long long* __restrict__ goffset, int* __restrict__ gidx,
-    float* __restrict__ gval, int* __restrict__ gcount,
-    double* __restrict__ llh, float* __restrict__ gg, int K, float min_p,
-    float max_p) {
+    const long long* __restrict__ epos, const long long* __restrict__ goffset,
+    int* __restrict__ gidx, float* __restrict__ gval,
+    int* __restrict__ gcount, double* __restrict__ llh,
+    const float* __restrict__ GGp, const float* __restrict__ ladder,
+    float* __restrict__ best, int n_ladder, int cap, float alpha,
+    float min_p, float max_p, float min_f, float max_f) {
   const int u = order[blockIdx.x];
   const long long e0 = indptr[u];
   const long long e1 = indptr[u + 1];
   const int tid = threadIdx.x;
+  const int lane = tid & (WAVE - 1);
+  const int wid = tid >> 6;
+  const int nw = (K + 31) >> 5;
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  float* gacc = reinterpret_cast<float*>(smem);  // K floats
+  float* gacc = reinterpret_cast<float*>(smem);        // K
+  int* nidx = reinterpret_cast<int*>(gacc + K);        // cap
+  float* nval = reinterpret_cast<float*>(nidx + cap);  // cap
+  int* kS = reinterpret_cast<int*>(nval + cap);        // cap
+  float* gS = reinterpret_cast<float*>(kS + cap);      // cap
+  float* fuS = gS + cap;                               // cap
+  u32* bmap = reinterpret_cast<u32*>(fuS + cap);       // nw
   __shared__ float red[NWAVE];
   __shared__ int scan[BLOCK];
+  __shared__ float s_lad[MAX_LS];
+  __shared__ double acc_llh[NWAVE][MAX_LS];
+  __shared__ float acc_nt[NWAVE][MAX_LS];
+  __shared__ double sh_llh_base;
+  __shared__ float sh_gg;
+
+  if (tid < MAX_LS) s_lad[tid] = (tid < n_ladder) ? ladder[tid] : 0.f;
   for (int k = tid * 4; k < K; k += BLOCK * 4)
     *reinterpret_cast<float4*>(gacc + k) = float4{0.f, 0.f, 0.f, 0.f};
+  for (int i = tid; i < nw; i += BLOCK) bmap[i] = 0u;
   __syncthreads();
 
-  double llh_acc = 0.0;  // thread 0
-  for (long long e = e0; e < e1; ++e) {
+  // phase 1: stage neighbor lists (wave-per-edge) + OR support bits;
+  // per-edge LDS base = epos[e] - epos[e0] (host prefix of scount)
+  const long long p0 = epos[e0];
+  for (long long e = e0 + wid; e < e1; e += NWAVE) {
     const int v = indices[e];
     const long long off = soffset[v];
     const int sv = scount[v];
+    const int base = (int)(epos[e] - p0);
+    for (int j = lane; j < sv; j += WAVE) {
+      const int k = sidx[off + j];
+      nidx[base + j] = k;
+      nval[base + j] = sval[off + j];
+      atomicOr(&bmap[k >> 5], 1u << (k & 31));
+    }
+  }
+  {  // own support bits
+    const long long offu = soffset[u];
+    const int su = scount[u];
+    for (int j = tid; j < su; j += BLOCK) {
+      const int k = sidx[offu + j];
+      atomicOr(&bmap[k >> 5], 1u << (k & 31));
+    }
+  }
+  __syncthreads();
+
+  // phase 2: per-edge dot (entries from LDS, fu gathered L1-hot) ->
+  // w-weighted scatter into gacc; edge llh on thread 0
+  double llh_acc = 0.0;
+  for (long long e = e0; e < e1; ++e) {
+    const int base = (int)(epos[e] - p0);
+    const int sv = (int)(epos[e + 1] - epos[e]);
     float part = 0.f;
     for (int j = tid; j < sv; j += BLOCK)
-      part += f_elem<BF16>(Fp, ldF, u, sidx[off + j]) * sval[off + j];
+      part += f_elem<BF16>(Fp, K, u, nidx[base + j]) * nval[base + j];
     const float x = block_allreduce_sum(part, red);  // syncs the block
     const float p = clamp_p(__expf(-x), min_p, max_p);
     const float w = 1.f / (1.f - p);
     if (tid == 0) llh_acc += (double)log1pf(-p) + (double)x;
-    // threads write DISTINCT gacc slots within one edge; the next
-    // edge's block_allreduce orders cross-edge writes
     for (int j = tid; j < sv; j += BLOCK)
-      gacc[sidx[off + j]] += w * sval[off + j];
+      gacc[nidx[base + j]] += w * nval[base + j];
   }
   __syncthreads();
 
-  // dense K-scan: emit compact (k, g) for S_u, node terms, gg
-  const int chunk = (K + BLOCK - 1) / BLOCK;
-  const int c0 = min(tid * chunk, K);
-  const int c1 = min(c0 + chunk, K);
+  // phase 3: bitmap scan -> compact (k, g, fu) into LDS + global pools;
+  // node terms fs/ff and gg accumulate at emission
+  const int wchunk = (nw + BLOCK - 1) / BLOCK;
+  const int w0 = min(tid * wchunk, nw);
+  const int w1 = min(w0 + wchunk, nw);
   int cnt = 0;
-  float gg_p = 0.f, fs_p = 0.f, ff_p = 0.f;
-  for (int c = c0; c < c1; ++c) {
-    const float a = gacc[c];
-    const float f = f_elem<BF16>(Fp, ldF, u, c);
-    const float s = sumF[c];
-    fs_p = fmaf(f, s, fs_p);
-    ff_p = fmaf(f, f, ff_p);
-    if (a != 0.f || f != 0.f) {
-      const float g = a - s + f;
-      gg_p += g * g - s * s;
-      ++cnt;
-    }
-  }
+  for (int wv = w0; wv < w1; ++wv) cnt += __popc(bmap[wv]);
   scan[tid] = cnt;
   __syncthreads();
 #pragma unroll
@@ -2543,103 +2626,75 @@ __global__ void __launch_bounds__(BLOCK) k1s_grad_t(
     scan[tid] += v;
     __syncthreads();
   }
-  long long w = goffset[blockIdx.x] + (scan[tid] - cnt);
-  for (int c = c0; c < c1; ++c) {
-    const float a = gacc[c];
-    const float f = f_elem<BF16>(Fp, ldF, u, c);
-    if (a != 0.f || f != 0.f) {
-      gidx[w] = c;
-      gval[w] = a - sumF[c] + f;
-      ++w;
+  const int ns = scan[BLOCK - 1];
+  int pos = scan[tid] - cnt;
+  const long long go = goffset[blockIdx.x];
+  float gg_p = 0.f, fs_p = 0.f, ff_p = 0.f;
+  for (int wv = w0; wv < w1; ++wv) {
+    u32 bits = bmap[wv];
+    while (bits) {
+      const int b = __ffs(bits) - 1;
+      bits &= bits - 1;
+      const int k = (wv << 5) + b;
+      const float a = gacc[k];
+      const float f = f_elem<BF16>(Fp, K, u, k);
+      const float sfk = sumF[k];
+      const float g = a - sfk + f;
+      kS[pos] = k;
+      gS[pos] = g;
+      fuS[pos] = f;
+      gidx[go + pos] = k;
+      gval[go + pos] = g;
+      gg_p += g * g - sfk * sfk;
+      fs_p = fmaf(f, sfk, fs_p);
+      ff_p = fmaf(f, f, ff_p);
+      ++pos;
     }
   }
   const float ggt = block_allreduce_sum(gg_p, red);
   const float fst = block_allreduce_sum(fs_p, red);
   const float fft = block_allreduce_sum(ff_p, red);
   if (tid == 0) {
-    gcount[blockIdx.x] = scan[BLOCK - 1];
-    gg[u] = ggt;
-    llh[u] = llh_acc + (double)(-fst) + (double)fft;
-  }
-}
-
-template <bool BF16>
-__global__ void __launch_bounds__(BLOCK) k2s_ls_t(
-    const void* __restrict__ Fp, int ldF,
-    const long long* __restrict__ indptr, const int* __restrict__ indices,
-    const float* __restrict__ sumF, const int* __restrict__ order,
-    const long long* __restrict__ soffset, const int* __restrict__ sidx,
-    const float* __restrict__ sval, const int* __restrict__ scount,
-    const long long* __restrict__ goffset, const int* __restrict__ gidx,
-    const float* __restrict__ gval, const int* __restrict__ gcount,
-    const double* __restrict__ llh, const float* __restrict__ gg,
-    const float* __restrict__ GGp, const float* __restrict__ ladder,
-    float* __restrict__ best, int n_ladder, float alpha, float min_p,
-    float max_p, float min_f, float max_f) {
-  const int u = order[blockIdx.x];
-  const long long e0 = indptr[u];
-  const long long e1 = indptr[u + 1];
-  const int tid = threadIdx.x;
-  const int lane = tid & (WAVE - 1);
-  const int wid = tid >> 6;
-  const int ns = gcount[blockIdx.x];
-  const long long go = goffset[blockIdx.x];
-
-  __shared__ float s_lad[MAX_LS];
-  __shared__ double acc_llh[NWAVE][MAX_LS];
-  __shared__ float acc_nt[NWAVE][MAX_LS];
-  extern __shared__ __attribute__((aligned(16))) char smem[];
-  // layout: k_S[stride] | g_S[stride] | fu_S[stride], stride = ns rounded
-  // to 4; the host sizes the dynamic LDS for the launch's max gcount
-  int* k_S = reinterpret_cast<int*>(smem);
-
-  if (tid < MAX_LS) s_lad[tid] = (tid < n_ladder) ? ladder[tid] : 0.f;
-  const int stride = (ns + 3) & ~3;
-  float* gS = reinterpret_cast<float*>(k_S + stride);
-  float* fuS = gS + stride;
-  for (int i = tid; i < ns; i += BLOCK) {
-    const int k = gidx[go + i];
-    k_S[i] = k;
-    gS[i] = gval[go + i];
-    fuS[i] = f_elem<BF16>(Fp, ldF, u, k);
+    gcount[blockIdx.x] = ns;
+    sh_llh_base = llh_acc + (double)(-fst) + (double)fft;
+    sh_gg = ggt + GGp[0];
+    llh[u] = sh_llh_base;
   }
   __syncthreads();
 
+  // phase 4: 16-candidate trial scoring from LDS (wave-per-edge; binary
+  // search of each staged entry into the compact kS)
   float s[MAX_LS];
 #pragma unroll
   for (int j = 0; j < MAX_LS; ++j) s[j] = s_lad[j];
   const int jmine = (((lane >> 5) & 1) << 3) | (((lane >> 4) & 1) << 2) |
                     (((lane >> 3) & 1) << 1) | ((lane >> 2) & 1);
   double llh_mine = 0.0;
-
   for (long long e = e0 + wid; e < e1; e += NWAVE) {
-    const int v = indices[e];
-    const long long off = soffset[v];
-    const int sv = scount[v];
+    const int base = (int)(epos[e] - p0);
+    const int sv = (int)(epos[e + 1] - epos[e]);
     float acc[MAX_LS];
 #pragma unroll
     for (int j = 0; j < MAX_LS; ++j) acc[j] = 0.f;
     for (int j0 = lane; j0 < sv; j0 += WAVE) {
-      const int k = sidx[off + j0];
-      // binary search k in k_S (sorted ascending)
+      const int k = nidx[base + j0];
       int lo = 0, hi = ns;
       while (lo < hi) {
         const int mid = (lo + hi) >> 1;
-        if (k_S[mid] < k)
+        if (kS[mid] < k)
           lo = mid + 1;
         else
           hi = mid;
       }
-      if (lo < ns && k_S[lo] == k) {
-        const float fv = sval[off + j0];
-        const float fu = fuS[lo];
-        const float gk = gS[lo];
+      // every staged k is in the bitmap, so kS[lo] == k always
+      const float fv = nval[base + j0];
+      const float fu = fuS[lo];
+      const float gk = gS[lo];
 #pragma unroll
-        for (int j = 0; j < MAX_LS; ++j) {
-          const float c = __builtin_amdgcn_fmed3f(
-              fmaf(s[j], gk, fu), min_f, max_f);
-          acc[j] = fmaf(c, fv, acc[j]);
-        }
+      for (int j = 0; j < MAX_LS; ++j) {
+        const float c =
+            __builtin_amdgcn_fmed3f(fmaf(s[j], gk, fu), min_f, max_f);
+        acc[j] = fmaf(c, fv, acc[j]);
       }
     }
     wave_reduce16(acc, lane);
@@ -2649,15 +2704,13 @@ __global__ void __launch_bounds__(BLOCK) k2s_ls_t(
       llh_mine += (double)log1pf(-p) + (double)x;
     }
   }
-
-  // node terms over S_u: c_j · (fu − sumF)
   float accn[MAX_LS];
 #pragma unroll
   for (int j = 0; j < MAX_LS; ++j) accn[j] = 0.f;
   for (int i = tid; i < ns; i += BLOCK) {
     const float fu = fuS[i];
     const float gk = gS[i];
-    const float d = fu - sumF[k_S[i]];
+    const float d = fu - sumF[kS[i]];
 #pragma unroll
     for (int j = 0; j < MAX_LS; ++j) {
       const float c =
@@ -2679,9 +2732,8 @@ __global__ void __launch_bounds__(BLOCK) k2s_ls_t(
 #pragma unroll
       for (int wv = 0; wv < NWAVE; ++wv)
         trial += acc_llh[wv][lane] + (double)acc_nt[wv][lane];
-      const float ggfull = gg[u] + GGp[0];
       ok = (lane < n_ladder) &&
-           (trial >= llh[u] + (double)(alpha * s_lad[lane] * ggfull));
+           (trial >= sh_llh_base + (double)(alpha * s_lad[lane] * sh_gg));
     }
     const unsigned long long bal = __ballot(ok);
     if (lane == 0)
@@ -3207,61 +3259,34 @@ extern "C" void launch_kaf(const void* F, int bf16, int n_rows, int K,
   HIP_CHECK(hipGetLastError());
 }
 
-extern "C" void launch_k1s(const void* F, int bf16,
+extern "C" void launch_kfs(const void* F, int bf16,
                            const long long* indptr, const int* indices,
                            const float* sumF, const int* order, int n_blocks,
                            const long long* soffset, const int* sidx,
                            const float* sval, const int* scount,
-                           const long long* goffset, int* gidx, float* gval,
-                           int* gcount, double* llh, float* gg, int K,
-                           float min_p, float max_p, hipStream_t stream) {
-  if (n_blocks == 0) return;
-  const size_t lds = (size_t)K * 4;
-  if (bf16) {
-    allow_large_lds((const void*)&k1s_grad_t<true>, lds);
-    hipLaunchKernelGGL((k1s_grad_t<true>), dim3(n_blocks), dim3(BLOCK), lds,
-                       stream, F, K, indptr, indices, sumF, order, soffset,
-                       sidx, sval, scount, goffset, gidx, gval, gcount, llh,
-                       gg, K, min_p, max_p);
-  } else {
-    allow_large_lds((const void*)&k1s_grad_t<false>, lds);
-    hipLaunchKernelGGL((k1s_grad_t<false>), dim3(n_blocks), dim3(BLOCK), lds,
-                       stream, F, K, indptr, indices, sumF, order, soffset,
-                       sidx, sval, scount, goffset, gidx, gval, gcount, llh,
-                       gg, K, min_p, max_p);
-  }
-  HIP_CHECK(hipGetLastError());
-}
-
-extern "C" void launch_k2s(const void* F, int bf16,
-                           const long long* indptr, const int* indices,
-                           const float* sumF, const int* order, int n_blocks,
-                           const long long* soffset, const int* sidx,
-                           const float* sval, const int* scount,
-                           const long long* goffset, const int* gidx,
-                           const float* gval, const int* gcount,
-                           const double* llh, const float* gg,
+                           const long long* epos, const long long* goffset,
+                           int* gidx, float* gval, int* gcount, double* llh,
                            const float* GGp, const float* ladder,
-                           float* best, int n_ladder, int max_count, int K,
+                           float* best, int n_ladder, int cap, int K,
                            float alpha, float min_p, float max_p,
                            float min_f, float max_f, hipStream_t stream) {
   if (n_blocks == 0) return;
   if (n_ladder > 16) throw std::runtime_error("ladder length > 16 unsupported");
-  const size_t lds = 12 * ((size_t)max_count + 4);
+  const size_t lds = (size_t)K * 4 + (size_t)cap * 20 + ((K + 31) / 32) * 4;
   if (bf16) {
-    allow_large_lds((const void*)&k2s_ls_t<true>, lds);
-    hipLaunchKernelGGL((k2s_ls_t<true>), dim3(n_blocks), dim3(BLOCK), lds,
-                       stream, F, K, indptr, indices, sumF, order, soffset,
-                       sidx, sval, scount, goffset, gidx, gval, gcount, llh,
-                       gg, GGp, ladder, best, n_ladder, alpha, min_p, max_p,
-                       min_f, max_f);
+    allow_large_lds((const void*)&kfs_sparse_t<true>, lds);
+    hipLaunchKernelGGL((kfs_sparse_t<true>), dim3(n_blocks), dim3(BLOCK),
+                       lds, stream, F, K, indptr, indices, sumF, order,
+                       soffset, sidx, sval, scount, epos, goffset, gidx,
+                       gval, gcount, llh, GGp, ladder, best, n_ladder, cap,
+                       alpha, min_p, max_p, min_f, max_f);
   } else {
-    allow_large_lds((const void*)&k2s_ls_t<false>, lds);
-    hipLaunchKernelGGL((k2s_ls_t<false>), dim3(n_blocks), dim3(BLOCK), lds,
-                       stream, F, K, indptr, indices, sumF, order, soffset,
-                       sidx, sval, scount, goffset, gidx, gval, gcount, llh,
-                       gg, GGp, ladder, best, n_ladder, alpha, min_p, max_p,
-                       min_f, max_f);
+    allow_large_lds((const void*)&kfs_sparse_t<false>, lds);
+    hipLaunchKernelGGL((kfs_sparse_t<false>), dim3(n_blocks), dim3(BLOCK),
+                       lds, stream, F, K, indptr, indices, sumF, order,
+                       soffset, sidx, sval, scount, epos, goffset, gidx,
+                       gval, gcount, llh, GGp, ladder, best, n_ladder, cap,
+                       alpha, min_p, max_p, min_f, max_f);
   }
   HIP_CHECK(hipGetLastError());
 }

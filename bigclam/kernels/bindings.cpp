@@ -341,18 +341,13 @@ void edge_grad_llh_chunked(torch::Tensor F, torch::Tensor indptr,
 extern "C" void launch_kaf(const void*, int, int, int, int,
                            const long long*, int*, int*, float*, int,
                            hipStream_t);
-extern "C" void launch_k1s(const void*, int, const long long*, const int*,
+extern "C" void launch_kfs(const void*, int, const long long*, const int*,
                            const float*, const int*, int, const long long*,
                            const int*, const float*, const int*,
-                           const long long*, int*, float*, int*, double*,
-                           float*, int, float, float, hipStream_t);
-extern "C" void launch_k2s(const void*, int, const long long*, const int*,
-                           const float*, const int*, int, const long long*,
-                           const int*, const float*, const int*,
-                           const long long*, const int*, const float*,
-                           const int*, const double*, const float*,
-                           const float*, const float*, float*, int, int, int,
-                           float, float, float, float, float, hipStream_t);
+                           const long long*, const long long*, int*, float*,
+                           int*, double*, const float*, const float*,
+                           float*, int, int, int, float, float, float,
+                           float, float, hipStream_t);
 extern "C" void launch_k3s(void*, int, const int*, int, const long long*,
                            const int*, const float*, const int*,
                            const float*, int, float, float, hipStream_t);
@@ -377,60 +372,52 @@ void sparse_support(torch::Tensor F, torch::Tensor soffset,
              sval.data_ptr<float>(), fill ? 1 : 0, current_stream());
 }
 
-void sparse_grad(torch::Tensor F, torch::Tensor indptr, torch::Tensor indices,
-                 torch::Tensor sumF, torch::Tensor order,
-                 torch::Tensor soffset, torch::Tensor sidx,
-                 torch::Tensor sval, torch::Tensor scount,
-                 torch::Tensor goffset, torch::Tensor gidx,
-                 torch::Tensor gval, torch::Tensor gcount, torch::Tensor llh,
-                 torch::Tensor gg, double min_p, double max_p) {
+void sparse_fused(torch::Tensor F, torch::Tensor indptr,
+                  torch::Tensor indices, torch::Tensor sumF,
+                  torch::Tensor order, torch::Tensor soffset,
+                  torch::Tensor sidx, torch::Tensor sval,
+                  torch::Tensor scount, torch::Tensor epos,
+                  torch::Tensor goffset, torch::Tensor gidx,
+                  torch::Tensor gval, torch::Tensor gcount,
+                  torch::Tensor llh, torch::Tensor GG, torch::Tensor ladder,
+                  torch::Tensor best, int64_t cap, double alpha,
+                  double min_p, double max_p, double min_f, double max_f) {
   CHECK_F(F);
   CHECK_IN(indptr, torch::kInt64);
   CHECK_IN(indices, torch::kInt32);
   CHECK_IN(sumF, torch::kFloat32);
   CHECK_IN(order, torch::kInt32);
+  CHECK_IN(soffset, torch::kInt64);
+  CHECK_IN(sidx, torch::kInt32);
+  CHECK_IN(sval, torch::kFloat32);
+  CHECK_IN(scount, torch::kInt32);
+  CHECK_IN(epos, torch::kInt64);
   CHECK_IN(goffset, torch::kInt64);
   CHECK_IN(gidx, torch::kInt32);
   CHECK_IN(gval, torch::kFloat32);
   CHECK_IN(gcount, torch::kInt32);
   CHECK_IN(llh, torch::kFloat64);
-  CHECK_IN(gg, torch::kFloat32);
-  const int n_blocks = (int)order.size(0);
-  TORCH_CHECK(goffset.size(0) >= n_blocks && gcount.size(0) >= n_blocks);
-  launch_k1s(F.data_ptr(), is_bf16(F) ? 1 : 0, i64p(indptr),
-             indices.data_ptr<int>(), sumF.data_ptr<float>(),
-             order.data_ptr<int>(), n_blocks, i64p(soffset),
-             sidx.data_ptr<int>(), sval.data_ptr<float>(),
-             scount.data_ptr<int>(), i64p(goffset), gidx.data_ptr<int>(),
-             gval.data_ptr<float>(), gcount.data_ptr<int>(),
-             llh.data_ptr<double>(), gg.data_ptr<float>(), (int)F.size(1),
-             (float)min_p, (float)max_p, current_stream());
-}
-
-void sparse_ls(torch::Tensor F, torch::Tensor indptr, torch::Tensor indices,
-               torch::Tensor sumF, torch::Tensor order, torch::Tensor soffset,
-               torch::Tensor sidx, torch::Tensor sval, torch::Tensor scount,
-               torch::Tensor goffset, torch::Tensor gidx, torch::Tensor gval,
-               torch::Tensor gcount, torch::Tensor llh, torch::Tensor gg,
-               torch::Tensor GG, torch::Tensor ladder, torch::Tensor best,
-               int64_t max_count, double alpha, double min_p, double max_p,
-               double min_f, double max_f) {
-  CHECK_F(F);
+  CHECK_IN(GG, torch::kFloat32);
   CHECK_IN(ladder, torch::kFloat32);
   CHECK_IN(best, torch::kFloat32);
   const int n_blocks = (int)order.size(0);
-  launch_k2s(F.data_ptr(), is_bf16(F) ? 1 : 0, i64p(indptr),
+  TORCH_CHECK(goffset.size(0) >= n_blocks && gcount.size(0) >= n_blocks);
+  TORCH_CHECK(epos.size(0) == indices.size(0) + 1, "epos must be nnz+1");
+  launch_kfs(F.data_ptr(), is_bf16(F) ? 1 : 0,
+             reinterpret_cast<const long long*>(indptr.data_ptr<int64_t>()),
              indices.data_ptr<int>(), sumF.data_ptr<float>(),
-             order.data_ptr<int>(), n_blocks, i64p(soffset),
+             order.data_ptr<int>(), n_blocks,
+             reinterpret_cast<const long long*>(soffset.data_ptr<int64_t>()),
              sidx.data_ptr<int>(), sval.data_ptr<float>(),
-             scount.data_ptr<int>(), i64p(goffset), gidx.data_ptr<int>(),
-             gval.data_ptr<float>(), gcount.data_ptr<int>(),
-             llh.data_ptr<double>(), gg.data_ptr<float>(),
-             GG.data_ptr<float>(),
-             ladder.data_ptr<float>(), best.data_ptr<float>(),
-             (int)ladder.size(0), (int)max_count, (int)F.size(1),
-             (float)alpha, (float)min_p, (float)max_p, (float)min_f,
-             (float)max_f, current_stream());
+             scount.data_ptr<int>(),
+             reinterpret_cast<const long long*>(epos.data_ptr<int64_t>()),
+             reinterpret_cast<const long long*>(goffset.data_ptr<int64_t>()),
+             gidx.data_ptr<int>(), gval.data_ptr<float>(),
+             gcount.data_ptr<int>(), llh.data_ptr<double>(),
+             GG.data_ptr<float>(), ladder.data_ptr<float>(),
+             best.data_ptr<float>(), (int)ladder.size(0), (int)cap,
+             (int)F.size(1), (float)alpha, (float)min_p, (float)max_p,
+             (float)min_f, (float)max_f, current_stream());
 }
 
 void sparse_commit(torch::Tensor F, torch::Tensor order,
@@ -508,10 +495,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "K3+colsum fused (bf16): commit F and emit per-stripe column sums");
   m.def("sparse_support", &sparse_support,
         "KAF: per-row support compaction of F (count or fill pass)");
-  m.def("sparse_grad", &sparse_grad,
-        "K1S: compact gradient + llh + gg for routed nodes");
-  m.def("sparse_ls", &sparse_ls,
-        "K2S: 16-candidate Armijo on compact active sets");
+  m.def("sparse_fused", &sparse_fused,
+        "KFS: fused compact gradient + 16-candidate Armijo for routed "
+        "nodes (LDS bitmap active sets)");
   m.def("sparse_commit", &sparse_commit,
         "K3S: sparse projected commit confined to the active set");
   m.def("edge_grad_llh_chunked", &edge_grad_llh_chunked,

@@ -228,6 +228,7 @@ def sparse_sweep_part(
     sidx: torch.Tensor,
     sval: torch.Tensor,
     scount: torch.Tensor,
+    epos: torch.Tensor,
     goffset: torch.Tensor,
     gpool_size: int,
     cap: int,
@@ -235,24 +236,20 @@ def sparse_sweep_part(
     best_out: torch.Tensor,
     cfg: BigClamConfig,
 ):
-    """K1S + K2S for the routed (sparse) nodes: compact gradient pools,
-    llh/gg per node, Armijo best step.  Writes llh_out/best_out at the
-    routed node positions; returns the commit pack for K3S."""
+    """KFS for the routed (sparse) nodes: one fused launch computing the
+    compact gradient pools, llh per node, and the Armijo best step.
+    Writes llh_out/best_out at the routed node positions; returns the
+    commit pack for K3S."""
     ext = ensure_loaded()
     dev = F.device
     n_s = int(order_sparse.numel())
     gidx = _pool(dev, "gidx", gpool_size, torch.int32)
     gval = _pool(dev, "gval", gpool_size, torch.float32)
     gcount = torch.empty(n_s, device=dev, dtype=torch.int32)
-    gg = torch.empty(int(indptr.numel()) - 1, device=dev, dtype=torch.float32)
-    ext.sparse_grad(
-        F, indptr, indices, sumF, order_sparse, soffset, sidx, sval, scount,
-        goffset, gidx, gval, gcount, llh_out, gg, cfg.min_p, cfg.max_p,
-    )
     GG = (sumF * sumF).sum().reshape(1)  # device scalar: no host sync
-    ext.sparse_ls(
+    ext.sparse_fused(
         F, indptr, indices, sumF, order_sparse, soffset, sidx, sval, scount,
-        goffset, gidx, gval, gcount, llh_out, gg, GG, _ladder(cfg, dev),
+        epos, goffset, gidx, gval, gcount, llh_out, GG, _ladder(cfg, dev),
         best_out, cap, cfg.alpha, cfg.min_p, cfg.max_p, cfg.min_f, cfg.max_f,
     )
     return {
