@@ -128,6 +128,9 @@ class ShardState:
         self._halo_send: Optional[torch.Tensor] = None
         self._colsum_partials: Optional[torch.Tensor] = None
         self._sp_soffset: Optional[torch.Tensor] = None
+        self._indices64: Optional[torch.Tensor] = None
+        self._order64: Optional[torch.Tensor] = None
+        self._last_nnz: Optional[torch.Tensor] = None  # from KAF scount
         # per-edge source row (torch reference path); built lazily on CPU
         self._edge_src: Optional[torch.Tensor] = None
 
@@ -331,6 +334,7 @@ class ShardState:
         the rest to the dense fused path.  Returns
         (grad, llh_nodes, best, sparse_pack|None)."""
         if not self.sparse_allowed:
+            self._last_nnz = None
             g, l, b = self.fused_grad_ls_overlap(halo_work)
             return g, l, b, None
         ops = _hip_ops()
@@ -348,31 +352,32 @@ class ShardState:
             self._sp_soffset = (
                 torch.arange(n_rows, device=dev, dtype=torch.int64) * cap
             )
+            self._indices64 = self.indices.long()
+            self._order64 = self.order.long()
         soffset = self._sp_soffset
         sidx = ops._pool(dev, "sidx", n_rows * cap, torch.int32)
         sval = ops._pool(dev, "sval", n_rows * cap, torch.float32)
+        # single KAF pass: writes counts AND fills lists for rows <= cap
         ops.ensure_loaded().sparse_support(
-            self.F, soffset, scount, sidx, sval, cap, False
+            self.F, soffset, scount, sidx, sval, cap, True
         )
+        self._last_nnz = scount[: self.n_local].sum()
         sc = scount.to(torch.int64)
         # bound_u = s_u + sum_{v in N(u)} s_v  (cumsum segment trick)
-        sv_edges = sc[self.indices.long()]
+        sv_edges = sc[self._indices64]
         cs = torch.cat(
             [torch.zeros(1, device=dev, dtype=torch.int64),
              torch.cumsum(sv_edges, 0)]
         )
         bound = sc[: self.n_local] + cs[self.indptr[1:]] - cs[self.indptr[:-1]]
         is_sparse = bound <= cap
-        om = is_sparse[self.order.long()]
+        om = is_sparse[self._order64]
         order_s = self.order[om].contiguous()  # <- the one host sync
         n_s = int(order_s.numel())
         if n_s < max(64, self.n_local // 20):
             g, l, b = self.fused_grad_ls_overlap(None)
             return g, l, b, None
         order_d = self.order[~om].contiguous()
-        ops.ensure_loaded().sparse_support(
-            self.F, soffset, scount, sidx, sval, cap, True
-        )
         goffset = torch.arange(n_s, device=dev, dtype=torch.int64) * cap
         n, kp = self.n_local, self.kp
         grad = torch.empty(n, kp, device=dev, dtype=torch.float32)

@@ -225,8 +225,15 @@ class Trainer:
                     # F-row density: rows densify early and sparsify as
                     # the fit converges (input to a sparsity-adaptive
                     # candidate path; reference v3 keeps rows sparse)
-                    "f_nnz_frac": (nnz := float(
-                        (self.state.F_local != 0).float().mean().item()
+                    # density: reuse the sparse path's KAF counts when
+                    # available (a dedicated N*K pass otherwise)
+                    "f_nnz_frac": (nnz := (
+                        float(self.state._last_nnz.item())
+                        / (self.state.n_local * self.state.kp)
+                        if getattr(self.state, "_last_nnz", None) is not None
+                        else float(
+                            (self.state.F_local != 0).float().mean().item()
+                        )
                     )),
                     # all-zero F is an ABSORBING state (grad == -sumF == 0,
                     # llh stuck at the x=0 floor): surface it.  Observed

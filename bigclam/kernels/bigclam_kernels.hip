@@ -2443,7 +2443,6 @@ __global__ void __launch_bounds__(BLOCK) kaf_support_t(
   __shared__ int scan[BLOCK];
   const int r = blockIdx.x;
   if (r >= n_rows) return;
-  if (FILL && scount[r] > cap) return;  // unrouted row: no list needed
   // chunk rounded to 8 so the row reads vectorize (uint4 of 8 bf16 /
   // two float4); K is padded (bf16: %8, fp32: %4 -> guard the last 4)
   const int chunk = (((K + BLOCK - 1) / BLOCK) + 7) & ~7;
@@ -2475,10 +2474,8 @@ __global__ void __launch_bounds__(BLOCK) kaf_support_t(
     scan[threadIdx.x] += v;
     __syncthreads();
   }
-  if (!FILL) {
-    if (threadIdx.x == BLOCK - 1) scount[r] = scan[BLOCK - 1];
-    return;
-  }
+  if (threadIdx.x == BLOCK - 1) scount[r] = scan[BLOCK - 1];
+  if (!FILL || scan[BLOCK - 1] > cap) return;  // over-cap rows: count only
   long long w = soffset[r] + (scan[threadIdx.x] - cnt);
   if (BF16) {
     const u32* row =
@@ -2560,6 +2557,7 @@ __global__ void __launch_bounds__(BLOCK) kfs_sparse_t(
   __shared__ float acc_nt[NWAVE][MAX_LS];
   __shared__ double sh_llh_base;
   __shared__ float sh_gg;
+  __shared__ double dred[NWAVE];
 
   if (tid < MAX_LS) s_lad[tid] = (tid < n_ladder) ? ladder[tid] : 0.f;
   for (int k = tid * 4; k < K; k += BLOCK * 4)
@@ -2592,21 +2590,24 @@ __global__ void __launch_bounds__(BLOCK) kfs_sparse_t(
   }
   __syncthreads();
 
-  // phase 2: per-edge dot (entries from LDS, fu gathered L1-hot) ->
-  // w-weighted scatter into gacc; edge llh on thread 0
-  double llh_acc = 0.0;
-  for (long long e = e0; e < e1; ++e) {
+  // phase 2: WAVE-per-edge dot (entries from LDS, fu gathered L1-hot)
+  // -> w-weighted scatter via LDS atomicAdd.  The atomics make gacc's
+  // fp32 summation order run-dependent (unlike every dense kernel) —
+  // the documented determinism trade of the sparse path; barrier-free
+  // edge parallelism across the 4 waves is worth ~2x here.
+  double llh_acc = 0.0;  // per-wave lane-0 partials, combined below
+  for (long long e = e0 + wid; e < e1; e += NWAVE) {
     const int base = (int)(epos[e] - p0);
     const int sv = (int)(epos[e + 1] - epos[e]);
     float part = 0.f;
-    for (int j = tid; j < sv; j += BLOCK)
+    for (int j = lane; j < sv; j += WAVE)
       part += f_elem<BF16>(Fp, K, u, nidx[base + j]) * nval[base + j];
-    const float x = block_allreduce_sum(part, red);  // syncs the block
+    const float x = wave_allreduce_sum(part);
     const float p = clamp_p(__expf(-x), min_p, max_p);
     const float w = 1.f / (1.f - p);
-    if (tid == 0) llh_acc += (double)log1pf(-p) + (double)x;
-    for (int j = tid; j < sv; j += BLOCK)
-      gacc[nidx[base + j]] += w * nval[base + j];
+    if (lane == 0) llh_acc += (double)log1pf(-p) + (double)x;
+    for (int j = lane; j < sv; j += WAVE)
+      atomicAdd(&gacc[nidx[base + j]], w * nval[base + j]);
   }
   __syncthreads();
 
@@ -2617,17 +2618,23 @@ __global__ void __launch_bounds__(BLOCK) kfs_sparse_t(
   const int w1 = min(w0 + wchunk, nw);
   int cnt = 0;
   for (int wv = w0; wv < w1; ++wv) cnt += __popc(bmap[wv]);
-  scan[tid] = cnt;
-  __syncthreads();
+  // 2-barrier block scan: wave-local inclusive scan (shuffles), then
+  // cross-wave offsets via 4 wave totals in LDS
+  int incl = cnt;
 #pragma unroll
-  for (int off = 1; off < BLOCK; off <<= 1) {
-    const int v = (tid >= off) ? scan[tid - off] : 0;
-    __syncthreads();
-    scan[tid] += v;
-    __syncthreads();
+  for (int off = 1; off < WAVE; off <<= 1) {
+    const int v = __shfl_up(incl, off, WAVE);
+    if (lane >= off) incl += v;
   }
-  const int ns = scan[BLOCK - 1];
-  int pos = scan[tid] - cnt;
+  if (lane == WAVE - 1) scan[wid] = incl;
+  __syncthreads();
+  int wbase = 0;
+#pragma unroll
+  for (int wv = 0; wv < NWAVE; ++wv)
+    if (wv < wid) wbase += scan[wv];
+  const int ns = scan[0] + scan[1] + scan[2] + scan[3];
+  int pos = wbase + incl - cnt;
+  __syncthreads();
   const long long go = goffset[blockIdx.x];
   float gg_p = 0.f, fs_p = 0.f, ff_p = 0.f;
   for (int wv = w0; wv < w1; ++wv) {
@@ -2654,9 +2661,14 @@ __global__ void __launch_bounds__(BLOCK) kfs_sparse_t(
   const float ggt = block_allreduce_sum(gg_p, red);
   const float fst = block_allreduce_sum(fs_p, red);
   const float fft = block_allreduce_sum(ff_p, red);
+  if (lane == 0) dred[wid] = llh_acc;
+  __syncthreads();
   if (tid == 0) {
+    double lt = 0.0;
+#pragma unroll
+    for (int wv = 0; wv < NWAVE; ++wv) lt += dred[wv];
     gcount[blockIdx.x] = ns;
-    sh_llh_base = llh_acc + (double)(-fst) + (double)fft;
+    sh_llh_base = lt + (double)(-fst) + (double)fft;
     sh_gg = ggt + GGp[0];
     llh[u] = sh_llh_base;
   }
