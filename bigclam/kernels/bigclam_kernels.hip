@@ -2510,17 +2510,29 @@ __global__ void __launch_bounds__(BLOCK) kaf_support_t(
 
 // KFS: fused K1S+K2S — ONE launch per sweep for the routed nodes.
 // The first cut ran K1S and K2S separately with a dense K-column scan
-// and global compact-pool round-trips; measured 8.0 + 5.6 ms/sweep at
-// the converged headline config (rocprofv3, r03).  This fusion:
-//   * stages ALL neighbor-list entries in LDS once (<= cap by routing;
-//     per-edge bases come FREE from the host's epos prefix sum — the
-//     same cumsum grad_ls_auto builds for the routing bounds);
-//   * tracks the active set in an LDS BITMAP (K/32 words), so the
-//     compact emission scans ~K/32 words instead of K columns;
-//   * keeps the compact (k, g, fu) set in LDS for the trial phase (and
-//     mirrors it to the global pools for K3S).
-// LDS: gacc[K] | nidx[cap] | nval[cap] | kS[cap] | gS[cap] | fuS[cap]
-//      | bmap[ceil(K/32)]  (~46 KB at K=5000, cap=K/4 -> 3 blocks/CU).
+// and global compact-pool round-trips (8.0 + 5.6 ms/sweep measured at
+// the converged headline config); the fusion + the optimizations below
+// take the whole routed sweep under the dense kernel's time:
+//   * ALL neighbor-list entries staged in LDS once (<= cap by routing;
+//     per-edge bases come FREE from the host's epos prefix — the same
+//     cumsum grad_ls_auto builds for the routing bounds);
+//   * WAVE-per-edge dot + scatter via LDS atomicAdd — barrier-free edge
+//     parallelism.  The atomics make gacc's fp32 summation order
+//     run-dependent (unlike every dense kernel): the sparse path's
+//     documented determinism trade (BIGCLAM_SPARSE=0 restores bitwise
+//     reproducibility);
+//   * the active set tracked in an LDS BITMAP (K/32 words): compact
+//     emission scans words, not K columns;
+//   * LDS diet: u16 column ids, storage-dtype values -> 31 KB (bf16) /
+//     38 KB (fp32) at K=5000, cap=K/4 -> 4-5 blocks/CU;
+//   * one packed 3-value block reduction (gg, fs, ff) instead of three.
+
+template <bool BF16>
+__device__ __forceinline__ float vget(unsigned short v);
+template <>
+__device__ __forceinline__ float vget<true>(unsigned short v) {
+  return __uint_as_float((u32)v << 16);
+}
 
 template <bool BF16>
 __global__ void __launch_bounds__(BLOCK) kfs_sparse_t(
@@ -2543,21 +2555,32 @@ __global__ void __launch_bounds__(BLOCK) kfs_sparse_t(
   const int wid = tid >> 6;
   const int nw = (K + 31) >> 5;
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  float* gacc = reinterpret_cast<float*>(smem);        // K
-  int* nidx = reinterpret_cast<int*>(gacc + K);        // cap
-  float* nval = reinterpret_cast<float*>(nidx + cap);  // cap
-  int* kS = reinterpret_cast<int*>(nval + cap);        // cap
-  float* gS = reinterpret_cast<float*>(kS + cap);      // cap
-  float* fuS = gS + cap;                               // cap
-  u32* bmap = reinterpret_cast<u32*>(fuS + cap);       // nw
+  // layout: gacc[K] f32 | gS[cap] f32 | fuS[cap] f32|u16 | nval[cap]
+  // f32|u16 | nidx[cap] u16 | kS[cap] u16 | bmap[nw] u32
+  float* gacc = reinterpret_cast<float*>(smem);
+  float* gS = gacc + K;
+  char* pv = reinterpret_cast<char*>(gS + cap);
+  float* fuS_f = reinterpret_cast<float*>(pv);
+  unsigned short* fuS_h = reinterpret_cast<unsigned short*>(pv);
+  pv += (size_t)cap * (BF16 ? 2 : 4);
+  float* nval_f = reinterpret_cast<float*>(pv);
+  unsigned short* nval_h = reinterpret_cast<unsigned short*>(pv);
+  pv += (size_t)cap * (BF16 ? 2 : 4);
+  unsigned short* nidx = reinterpret_cast<unsigned short*>(pv);
+  pv += (size_t)cap * 2;
+  unsigned short* kS = reinterpret_cast<unsigned short*>(pv);
+  pv += (size_t)cap * 2;
+  pv = (char*)(((size_t)pv + 3) & ~(size_t)3);
+  u32* bmap = reinterpret_cast<u32*>(pv);
   __shared__ float red[NWAVE];
-  __shared__ int scan[BLOCK];
+  __shared__ int scan[NWAVE];
   __shared__ float s_lad[MAX_LS];
   __shared__ double acc_llh[NWAVE][MAX_LS];
   __shared__ float acc_nt[NWAVE][MAX_LS];
+  __shared__ float red3[NWAVE][3];
+  __shared__ double dred[NWAVE];
   __shared__ double sh_llh_base;
   __shared__ float sh_gg;
-  __shared__ double dred[NWAVE];
 
   if (tid < MAX_LS) s_lad[tid] = (tid < n_ladder) ? ladder[tid] : 0.f;
   for (int k = tid * 4; k < K; k += BLOCK * 4)
@@ -2565,8 +2588,7 @@ __global__ void __launch_bounds__(BLOCK) kfs_sparse_t(
   for (int i = tid; i < nw; i += BLOCK) bmap[i] = 0u;
   __syncthreads();
 
-  // phase 1: stage neighbor lists (wave-per-edge) + OR support bits;
-  // per-edge LDS base = epos[e] - epos[e0] (host prefix of scount)
+  // phase 1: stage neighbor lists (wave-per-edge) + OR support bits
   const long long p0 = epos[e0];
   for (long long e = e0 + wid; e < e1; e += NWAVE) {
     const int v = indices[e];
@@ -2575,51 +2597,51 @@ __global__ void __launch_bounds__(BLOCK) kfs_sparse_t(
     const int base = (int)(epos[e] - p0);
     for (int j = lane; j < sv; j += WAVE) {
       const int k = sidx[off + j];
-      nidx[base + j] = k;
-      nval[base + j] = sval[off + j];
+      nidx[base + j] = (unsigned short)k;
+      if (BF16)
+        nval_h[base + j] = pack1_bf16_rne(sval[off + j]);
+      else
+        nval_f[base + j] = sval[off + j];
       atomicOr(&bmap[k >> 5], 1u << (k & 31));
     }
   }
   {  // own support bits
     const long long offu = soffset[u];
     const int su = scount[u];
-    for (int j = tid; j < su; j += BLOCK) {
-      const int k = sidx[offu + j];
-      atomicOr(&bmap[k >> 5], 1u << (k & 31));
-    }
+    for (int j = tid; j < su; j += BLOCK)
+      atomicOr(&bmap[sidx[offu + j] >> 5], 1u << (sidx[offu + j] & 31));
   }
   __syncthreads();
 
-  // phase 2: WAVE-per-edge dot (entries from LDS, fu gathered L1-hot)
-  // -> w-weighted scatter via LDS atomicAdd.  The atomics make gacc's
-  // fp32 summation order run-dependent (unlike every dense kernel) —
-  // the documented determinism trade of the sparse path; barrier-free
-  // edge parallelism across the 4 waves is worth ~2x here.
-  double llh_acc = 0.0;  // per-wave lane-0 partials, combined below
+  // phase 2: WAVE-per-edge dot (fu gathered L1-hot) -> w-weighted LDS
+  // atomicAdd scatter; per-wave edge llh partials
+  double llh_acc = 0.0;
   for (long long e = e0 + wid; e < e1; e += NWAVE) {
     const int base = (int)(epos[e] - p0);
     const int sv = (int)(epos[e + 1] - epos[e]);
     float part = 0.f;
-    for (int j = lane; j < sv; j += WAVE)
-      part += f_elem<BF16>(Fp, K, u, nidx[base + j]) * nval[base + j];
+    for (int j = lane; j < sv; j += WAVE) {
+      const float fv = BF16 ? vget<true>(nval_h[base + j]) : nval_f[base + j];
+      part += f_elem<BF16>(Fp, K, u, nidx[base + j]) * fv;
+    }
     const float x = wave_allreduce_sum(part);
     const float p = clamp_p(__expf(-x), min_p, max_p);
     const float w = 1.f / (1.f - p);
     if (lane == 0) llh_acc += (double)log1pf(-p) + (double)x;
-    for (int j = lane; j < sv; j += WAVE)
-      atomicAdd(&gacc[nidx[base + j]], w * nval[base + j]);
+    for (int j = lane; j < sv; j += WAVE) {
+      const float fv = BF16 ? vget<true>(nval_h[base + j]) : nval_f[base + j];
+      atomicAdd(&gacc[nidx[base + j]], w * fv);
+    }
   }
   __syncthreads();
 
   // phase 3: bitmap scan -> compact (k, g, fu) into LDS + global pools;
-  // node terms fs/ff and gg accumulate at emission
+  // 2-barrier block scan (wave shuffles + cross-wave offsets)
   const int wchunk = (nw + BLOCK - 1) / BLOCK;
   const int w0 = min(tid * wchunk, nw);
   const int w1 = min(w0 + wchunk, nw);
   int cnt = 0;
   for (int wv = w0; wv < w1; ++wv) cnt += __popc(bmap[wv]);
-  // 2-barrier block scan: wave-local inclusive scan (shuffles), then
-  // cross-wave offsets via 4 wave totals in LDS
   int incl = cnt;
 #pragma unroll
   for (int off = 1; off < WAVE; off <<= 1) {
@@ -2634,7 +2656,6 @@ __global__ void __launch_bounds__(BLOCK) kfs_sparse_t(
     if (wv < wid) wbase += scan[wv];
   const int ns = scan[0] + scan[1] + scan[2] + scan[3];
   int pos = wbase + incl - cnt;
-  __syncthreads();
   const long long go = goffset[blockIdx.x];
   float gg_p = 0.f, fs_p = 0.f, ff_p = 0.f;
   for (int wv = w0; wv < w1; ++wv) {
@@ -2647,9 +2668,12 @@ __global__ void __launch_bounds__(BLOCK) kfs_sparse_t(
       const float f = f_elem<BF16>(Fp, K, u, k);
       const float sfk = sumF[k];
       const float g = a - sfk + f;
-      kS[pos] = k;
+      kS[pos] = (unsigned short)k;
       gS[pos] = g;
-      fuS[pos] = f;
+      if (BF16)
+        fuS_h[pos] = pack1_bf16_rne(f);
+      else
+        fuS_f[pos] = f;
       gidx[go + pos] = k;
       gval[go + pos] = g;
       gg_p += g * g - sfk * sfk;
@@ -2658,15 +2682,31 @@ __global__ void __launch_bounds__(BLOCK) kfs_sparse_t(
       ++pos;
     }
   }
-  const float ggt = block_allreduce_sum(gg_p, red);
-  const float fst = block_allreduce_sum(fs_p, red);
-  const float fft = block_allreduce_sum(ff_p, red);
-  if (lane == 0) dred[wid] = llh_acc;
+  // one packed reduction round: (gg, fs, ff) + the llh doubles
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) {
+    gg_p += __shfl_xor(gg_p, off, WAVE);
+    fs_p += __shfl_xor(fs_p, off, WAVE);
+    ff_p += __shfl_xor(ff_p, off, WAVE);
+    llh_acc += __shfl_xor(llh_acc, off, WAVE);
+  }
+  if (lane == 0) {
+    red3[wid][0] = gg_p;
+    red3[wid][1] = fs_p;
+    red3[wid][2] = ff_p;
+    dred[wid] = llh_acc;
+  }
   __syncthreads();
   if (tid == 0) {
+    float ggt = 0.f, fst = 0.f, fft = 0.f;
     double lt = 0.0;
 #pragma unroll
-    for (int wv = 0; wv < NWAVE; ++wv) lt += dred[wv];
+    for (int wv = 0; wv < NWAVE; ++wv) {
+      ggt += red3[wv][0];
+      fst += red3[wv][1];
+      fft += red3[wv][2];
+      lt += dred[wv];
+    }
     gcount[blockIdx.x] = ns;
     sh_llh_base = lt + (double)(-fst) + (double)fft;
     sh_gg = ggt + GGp[0];
@@ -2689,7 +2729,7 @@ __global__ void __launch_bounds__(BLOCK) kfs_sparse_t(
 #pragma unroll
     for (int j = 0; j < MAX_LS; ++j) acc[j] = 0.f;
     for (int j0 = lane; j0 < sv; j0 += WAVE) {
-      const int k = nidx[base + j0];
+      const unsigned short k = nidx[base + j0];
       int lo = 0, hi = ns;
       while (lo < hi) {
         const int mid = (lo + hi) >> 1;
@@ -2699,8 +2739,8 @@ __global__ void __launch_bounds__(BLOCK) kfs_sparse_t(
           hi = mid;
       }
       // every staged k is in the bitmap, so kS[lo] == k always
-      const float fv = nval[base + j0];
-      const float fu = fuS[lo];
+      const float fv = BF16 ? vget<true>(nval_h[base + j0]) : nval_f[base + j0];
+      const float fu = BF16 ? vget<true>(fuS_h[lo]) : fuS_f[lo];
       const float gk = gS[lo];
 #pragma unroll
       for (int j = 0; j < MAX_LS; ++j) {
@@ -2720,7 +2760,7 @@ __global__ void __launch_bounds__(BLOCK) kfs_sparse_t(
 #pragma unroll
   for (int j = 0; j < MAX_LS; ++j) accn[j] = 0.f;
   for (int i = tid; i < ns; i += BLOCK) {
-    const float fu = fuS[i];
+    const float fu = BF16 ? vget<true>(fuS_h[i]) : fuS_f[i];
     const float gk = gS[i];
     const float d = fu - sumF[kS[i]];
 #pragma unroll
@@ -3284,7 +3324,8 @@ extern "C" void launch_kfs(const void* F, int bf16,
                            float min_f, float max_f, hipStream_t stream) {
   if (n_blocks == 0) return;
   if (n_ladder > 16) throw std::runtime_error("ladder length > 16 unsupported");
-  const size_t lds = (size_t)K * 4 + (size_t)cap * 20 + ((K + 31) / 32) * 4;
+  const size_t lds = (size_t)K * 4 + (size_t)cap * (bf16 ? 12 : 16) +
+                     ((K + 31) / 32) * 4 + 16;
   if (bf16) {
     allow_large_lds((const void*)&kfs_sparse_t<true>, lds);
     hipLaunchKernelGGL((kfs_sparse_t<true>), dim3(n_blocks), dim3(BLOCK),
