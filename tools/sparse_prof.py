@@ -32,6 +32,8 @@ def main():
     ap.add_argument("--edges", type=int, default=925872)
     ap.add_argument("--converge", type=int, default=10)
     ap.add_argument("--steps", type=int, default=50)
+    ap.add_argument("--torchprof", action="store_true",
+                    help="print a torch.profiler op table for 3 sweeps")
     args = ap.parse_args()
     assert torch.cuda.is_available()
 
@@ -49,6 +51,17 @@ def main():
     for _ in range(3):
         carry, _, _ = tr.pipelined_sweep(carry)
     torch.cuda.synchronize()
+    if args.torchprof:
+        from torch.profiler import ProfilerActivity, profile
+
+        with profile(
+            activities=[ProfilerActivity.CPU, ProfilerActivity.CUDA]
+        ) as prof:
+            for _ in range(3):
+                carry, _, _ = tr.pipelined_sweep(carry)
+            torch.cuda.synchronize()
+        print(prof.key_averages().table(
+            sort_by="cuda_time_total", row_limit=25))
     t0 = time.perf_counter()
     for _ in range(args.steps):
         carry, llh, _ = tr.pipelined_sweep(carry)
