@@ -2546,7 +2546,7 @@ __global__ void __launch_bounds__(BLOCK) kfs_sparse_t(
     int* __restrict__ gcount, double* __restrict__ llh,
     const float* __restrict__ GGp, const float* __restrict__ ladder,
     float* __restrict__ best, int n_ladder, int cap, float alpha,
-    float min_p, float max_p, float min_f, float max_f) {
+    float min_p, float max_p, float min_f, float max_f, int phases) {
   const int u = order[blockIdx.x];
   const long long e0 = indptr[u];
   const long long e1 = indptr[u + 1];
@@ -2590,6 +2590,7 @@ __global__ void __launch_bounds__(BLOCK) kfs_sparse_t(
 
   // phase 1: stage neighbor lists (wave-per-edge) + OR support bits
   const long long p0 = epos[e0];
+  if (phases & 8)
   for (long long e = e0 + wid; e < e1; e += NWAVE) {
     const int v = indices[e];
     const long long off = soffset[v];
@@ -2616,6 +2617,7 @@ __global__ void __launch_bounds__(BLOCK) kfs_sparse_t(
   // phase 2: WAVE-per-edge dot (fu gathered L1-hot) -> w-weighted LDS
   // atomicAdd scatter; per-wave edge llh partials
   double llh_acc = 0.0;
+  if (phases & 1)
   for (long long e = e0 + wid; e < e1; e += NWAVE) {
     const int base = (int)(epos[e] - p0);
     const int sv = (int)(epos[e + 1] - epos[e]);
@@ -2641,7 +2643,8 @@ __global__ void __launch_bounds__(BLOCK) kfs_sparse_t(
   const int w0 = min(tid * wchunk, nw);
   const int w1 = min(w0 + wchunk, nw);
   int cnt = 0;
-  for (int wv = w0; wv < w1; ++wv) cnt += __popc(bmap[wv]);
+  if (phases & 2)
+    for (int wv = w0; wv < w1; ++wv) cnt += __popc(bmap[wv]);
   int incl = cnt;
 #pragma unroll
   for (int off = 1; off < WAVE; off <<= 1) {
@@ -2728,6 +2731,7 @@ __global__ void __launch_bounds__(BLOCK) kfs_sparse_t(
   const int jmine = (((lane >> 5) & 1) << 3) | (((lane >> 4) & 1) << 2) |
                     (((lane >> 3) & 1) << 1) | ((lane >> 2) & 1);
   double llh_mine = 0.0;
+  if (phases & 4)
   for (long long e = e0 + wid; e < e1; e += NWAVE) {
     const int base = (int)(epos[e] - p0);
     const int sv = (int)(epos[e + 1] - epos[e]);
@@ -3325,20 +3329,22 @@ extern "C" void launch_kfs(const void* F, int bf16,
   if (n_ladder > 16) throw std::runtime_error("ladder length > 16 unsupported");
   const size_t lds = (size_t)K * 4 + (size_t)cap * (bf16 ? 12 : 16) +
                      ((K + 31) / 32) * 4 + 16;
+  const char* ph = getenv("BIGCLAM_KFS_PHASES");  // measurement bisect only
+  const int phases = ph ? atoi(ph) : 0xF;
   if (bf16) {
     allow_large_lds((const void*)&kfs_sparse_t<true>, lds);
     hipLaunchKernelGGL((kfs_sparse_t<true>), dim3(n_blocks), dim3(BLOCK),
                        lds, stream, F, K, indptr, indices, sumF, order,
                        soffset, sidx, sval, scount, epos, goffset, gidx,
                        gval, gcount, llh, GGp, ladder, best, n_ladder, cap,
-                       alpha, min_p, max_p, min_f, max_f);
+                       alpha, min_p, max_p, min_f, max_f, phases);
   } else {
     allow_large_lds((const void*)&kfs_sparse_t<false>, lds);
     hipLaunchKernelGGL((kfs_sparse_t<false>), dim3(n_blocks), dim3(BLOCK),
                        lds, stream, F, K, indptr, indices, sumF, order,
                        soffset, sidx, sval, scount, epos, goffset, gidx,
                        gval, gcount, llh, GGp, ladder, best, n_ladder, cap,
-                       alpha, min_p, max_p, min_f, max_f);
+                       alpha, min_p, max_p, min_f, max_f, phases);
   }
   HIP_CHECK(hipGetLastError());
 }
