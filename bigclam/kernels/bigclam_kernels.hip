@@ -2575,8 +2575,8 @@ __global__ void __launch_bounds__(BLOCK) kfs_sparse_t(
   __shared__ float red[NWAVE];
   __shared__ int scan[NWAVE];
   __shared__ float s_lad[MAX_LS];
-  __shared__ double acc_llh[NWAVE][MAX_LS];
-  __shared__ float acc_nt[NWAVE][MAX_LS];
+  __shared__ double cllh[16][MAX_LS];
+  __shared__ float cnt16[16][MAX_LS];
   __shared__ float red3[NWAVE][3];
   __shared__ double dred[NWAVE];
   __shared__ double sh_llh_base;
@@ -2723,76 +2723,66 @@ __global__ void __launch_bounds__(BLOCK) kfs_sparse_t(
   }
   __syncthreads();
 
-  // phase 4: 16-candidate trial scoring from LDS (wave-per-edge; binary
-  // search of each staged entry into the compact kS)
-  float s[MAX_LS];
-#pragma unroll
-  for (int j = 0; j < MAX_LS; ++j) s[j] = s_lad[j];
-  const int jmine = (((lane >> 5) & 1) << 3) | (((lane >> 4) & 1) << 2) |
-                    (((lane >> 3) & 1) << 1) | ((lane >> 2) & 1);
-  double llh_mine = 0.0;
-  if (phases & 4)
-  for (long long e = e0 + wid; e < e1; e += NWAVE) {
-    const int base = (int)(epos[e] - p0);
-    const int sv = (int)(epos[e + 1] - epos[e]);
-    float acc[MAX_LS];
-#pragma unroll
-    for (int j = 0; j < MAX_LS; ++j) acc[j] = 0.f;
-    for (int j0 = lane; j0 < sv; j0 += WAVE) {
-      const unsigned short k = nidx[base + j0];
-      // every staged k is in the bitmap, so pmap[k] is always valid
-      const int lo = (int)pmap[k];
-      const float fv = BF16 ? vget<true>(nval_h[base + j0]) : nval_f[base + j0];
-      const float fu = BF16 ? vget<true>(fuS_h[lo]) : fuS_f[lo];
-      const float gk = gS[lo];
-#pragma unroll
-      for (int j = 0; j < MAX_LS; ++j) {
-        const float c =
-            __builtin_amdgcn_fmed3f(fmaf(s[j], gk, fu), min_f, max_f);
-        acc[j] = fmaf(c, fv, acc[j]);
+  // phase 4: 16-candidate trial scoring with an (edge-slot x candidate)
+  // THREAD mapping: thread (e, j) serially accumulates one edge's
+  // candidate-j trial dot over the staged entries — no wave-wide
+  // reductions at all (the 17-shuffle butterflies were the phase's cost
+  // at these tiny per-node sizes; measured in the r04 phase bisect).
+  if (phases & 4) {
+    const int ej = tid >> 4;  // edge slot 0..15
+    const int jc = tid & 15;  // candidate 0..15
+    const float sj = s_lad[jc];
+    double myllh = 0.0;
+    for (long long et = e0; et < e1; et += 16) {
+      const int ne = (int)min((long long)16, e1 - et);
+      if (ej < ne && jc < n_ladder) {
+        const long long e = et + ej;
+        const int base = (int)(epos[e] - p0);
+        const int sv = (int)(epos[e + 1] - epos[e]);
+        float x = 0.f;
+        for (int t = 0; t < sv; ++t) {
+          const int pos = (int)pmap[nidx[base + t]];
+          const float fv =
+              BF16 ? vget<true>(nval_h[base + t]) : nval_f[base + t];
+          const float fu = BF16 ? vget<true>(fuS_h[pos]) : fuS_f[pos];
+          const float c = __builtin_amdgcn_fmed3f(fmaf(sj, gS[pos], fu),
+                                                  min_f, max_f);
+          x = fmaf(c, fv, x);
+        }
+        const float p = clamp_p(__expf(-x), min_p, max_p);
+        myllh += (double)log1pf(-p) + (double)x;
       }
     }
-    wave_reduce16(acc, lane);
-    if ((lane & 3) == 0) {
-      const float x = acc[0];
-      const float p = clamp_p(__expf(-x), min_p, max_p);
-      llh_mine += (double)log1pf(-p) + (double)x;
+    // node terms: thread (stripe=ej, j=jc) over strided elements
+    float mynt = 0.f;
+    if (jc < n_ladder) {
+      for (int i = ej; i < ns; i += 16) {
+        const float fu = BF16 ? vget<true>(fuS_h[i]) : fuS_f[i];
+        const float c = __builtin_amdgcn_fmed3f(fmaf(sj, gS[i], fu), min_f,
+                                                max_f);
+        mynt = fmaf(c, fu - sumF[kS[i]], mynt);
+      }
     }
-  }
-  float accn[MAX_LS];
-#pragma unroll
-  for (int j = 0; j < MAX_LS; ++j) accn[j] = 0.f;
-  for (int i = tid; i < ns; i += BLOCK) {
-    const float fu = BF16 ? vget<true>(fuS_h[i]) : fuS_f[i];
-    const float gk = gS[i];
-    const float d = fu - sumF[kS[i]];
-#pragma unroll
-    for (int j = 0; j < MAX_LS; ++j) {
-      const float c =
-          __builtin_amdgcn_fmed3f(fmaf(s[j], gk, fu), min_f, max_f);
-      accn[j] = fmaf(c, d, accn[j]);
-    }
-  }
-  wave_reduce16(accn, lane);
-  if ((lane & 3) == 0) {
-    acc_nt[wid][jmine] = accn[0];
-    acc_llh[wid][jmine] = llh_mine;
-  }
-  __syncthreads();
-
-  if (wid == 0) {
-    bool ok = false;
-    if (lane < MAX_LS) {
+    cllh[ej][jc] = myllh;
+    cnt16[ej][jc] = mynt;
+    __syncthreads();
+    if (tid < MAX_LS) {
       double trial = 0.0;
+      float nt = 0.f;
 #pragma unroll
-      for (int wv = 0; wv < NWAVE; ++wv)
-        trial += acc_llh[wv][lane] + (double)acc_nt[wv][lane];
-      ok = (lane < n_ladder) &&
-           (trial >= sh_llh_base + (double)(alpha * s_lad[lane] * sh_gg));
+      for (int e = 0; e < 16; ++e) {
+        trial += cllh[e][tid];
+        nt += cnt16[e][tid];
+      }
+      const bool ok = (tid < n_ladder) &&
+                      (trial + (double)nt >=
+                       sh_llh_base + (double)(alpha * s_lad[tid] * sh_gg));
+      const unsigned long long bal = __ballot(ok);
+      if (tid == 0)
+        best[u] = bal ? s_lad[__ffsll((unsigned long long)bal) - 1] : 0.f;
     }
-    const unsigned long long bal = __ballot(ok);
-    if (lane == 0)
-      best[u] = bal ? s_lad[__ffsll((unsigned long long)bal) - 1] : 0.f;
+  } else if (tid == 0) {
+    best[u] = 0.f;
   }
 }
 
