@@ -327,6 +327,22 @@ class ShardState:
             and os.environ.get("BIGCLAM_SPARSE", "1") != "0"
         )
 
+    @staticmethod
+    def sparse_bounds(scount: torch.Tensor, indptr: torch.Tensor,
+                      indices64: torch.Tensor, n_local: int):
+        """Exact per-node active-set bound ``s_u + Σ_{v∈N(u)} s_v`` and the
+        per-edge support prefix (the kernel's staging offsets).  Pure
+        tensor math — works on CPU for the ws>1 routing tests even though
+        the kernels are GPU-only."""
+        sc = scount.to(torch.int64)
+        sv_edges = sc[indices64]
+        cs = torch.cat(
+            [torch.zeros(1, device=scount.device, dtype=torch.int64),
+             torch.cumsum(sv_edges, 0)]
+        )
+        bound = sc[:n_local] + cs[indptr[1:]] - cs[indptr[:-1]]
+        return bound, cs
+
     def grad_ls_auto(self, halo_work):
         """Per-sweep adaptive dispatch: route nodes whose active-set bound
         (own support + sum of neighbor supports, exact upper bound on
@@ -362,14 +378,9 @@ class ShardState:
             self.F, soffset, scount, sidx, sval, cap, True
         )
         self._last_nnz = scount[: self.n_local].sum()
-        sc = scount.to(torch.int64)
-        # bound_u = s_u + sum_{v in N(u)} s_v  (cumsum segment trick)
-        sv_edges = sc[self._indices64]
-        cs = torch.cat(
-            [torch.zeros(1, device=dev, dtype=torch.int64),
-             torch.cumsum(sv_edges, 0)]
+        bound, cs = self.sparse_bounds(
+            scount, self.indptr, self._indices64, self.n_local
         )
-        bound = sc[: self.n_local] + cs[self.indptr[1:]] - cs[self.indptr[:-1]]
         is_sparse = bound <= cap
         om = is_sparse[self._order64]
         order_s = self.order[om].contiguous()  # <- the one host sync

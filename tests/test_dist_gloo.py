@@ -262,3 +262,32 @@ def test_halo_plan_consistency():
                 np.testing.assert_array_equal(
                     np.sort(row), g.neighbors(u_local + s.start)
                 )
+
+
+def test_sparse_bounds_with_halo_rows():
+    """Routing bounds for the sparse-adaptive sweep must count HALO
+    neighbors' supports (a sharded node's active set spans remote rows).
+    Pure tensor math checked against a numpy brute force at ws=2 —
+    the GPU kernels consume exactly these bounds (state.sparse_bounds)."""
+    from bigclam.core.state import ShardState
+
+    g = _graph()
+    for ws, rank in [(2, 0), (2, 1), (3, 1)]:
+        bounds = partition_bounds(g, ws)
+        s = make_shard(g, rank, ws, bounds)
+        rng = np.random.default_rng(rank + 7)
+        scount = torch.from_numpy(
+            rng.integers(0, 9, size=s.n_rows).astype(np.int32)
+        )
+        indptr = torch.from_numpy(s.indptr)
+        idx64 = torch.from_numpy(s.indices.astype(np.int64))
+        bound, cs = ShardState.sparse_bounds(scount, indptr, idx64, s.n_local)
+        sc = scount.numpy()
+        for u in range(s.n_local):
+            nbrs = s.indices[s.indptr[u] : s.indptr[u + 1]]
+            want = sc[u] + int(sc[nbrs].sum())  # includes halo rows
+            assert int(bound[u]) == want, (ws, rank, u)
+        # cs is the per-edge staging prefix the KFS kernel uses
+        np.testing.assert_array_equal(
+            cs.numpy(), np.concatenate([[0], np.cumsum(sc[s.indices])])
+        )
