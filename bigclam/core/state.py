@@ -305,25 +305,33 @@ class ShardState:
     # ------------------------------------------------- sparse-adaptive path
     @property
     def sparse_cap(self) -> int:
-        """Active-set bound under which a node routes to K1S/K2S.  kp/4
-        keeps the compact work well under the dense kernels' while
-        letting moderate-degree nodes route (bound ≈ deg·density·K);
-        hubs always stay on the dense fused kernel."""
-        return int(os.environ.get("BIGCLAM_SPARSE_CAP", str(self.kp // 4)))
+        """Active-set bound under which a node routes to the KFS sparse
+        kernel.  kp/4 keeps the compact work well under the dense
+        kernels' while letting moderate-degree nodes route
+        (bound ≈ deg·density·K); hubs always stay on the dense path.
+        At large K the cap additionally shrinks to what fits next to the
+        dense-K LDS accumulator (gacc K·4 + pools cap·12/16 + bitmap
+        within the 160 KB workgroup LDS, ~4 KB margin for statics)."""
+        env = os.environ.get("BIGCLAM_SPARSE_CAP")
+        if env is not None:
+            return int(env)
+        per = 12 if self.storage_dtype == torch.bfloat16 else 16
+        nw4 = ((self.kp + 31) // 32) * 4
+        lds_cap = (160 * 1024 - 4096 - self.kp * 4 - nw4 - 64) // per
+        return max(0, min(self.kp // 4, lds_cap) & ~7)
 
     @property
     def sparse_allowed(self) -> bool:
         """Active-column sweep (docs/sparse_sweep_design.md): exact
         per-node routing by active-set bound; pays off once F sparsifies
         (measured 0.4-4% converged density at the headline configs).
-        K1S holds a dense K-float LDS accumulator, and the dense-node
-        remainder runs the fused kernel -> shapes limited to the fused
-        coverage (fp32 kp <= 8192, bf16 kp <= 16384).
-        BIGCLAM_SPARSE=0 disables."""
-        cap = 16384 if self.storage_dtype == torch.bfloat16 else 8192
+        KFS holds a dense K-float LDS accumulator, so kp is bounded by
+        the 160 KB workgroup LDS (the dense-node remainder runs the
+        fused kernel where it covers, else the separate chunked path —
+        both subset-capable).  BIGCLAM_SPARSE=0 disables."""
         return (
             self.use_hip
-            and self.kp <= cap
+            and self.sparse_cap >= 256
             and os.environ.get("BIGCLAM_SPARSE", "1") != "0"
         )
 
@@ -396,11 +404,22 @@ class ShardState:
         best = torch.empty(n, device=dev, dtype=torch.float32)
         n_d = int(order_d.numel())
         if n_d:
-            n_mfma_d = n_d if self.n_mfma == int(self.order.numel()) else 0
-            ops.fused_grad_ls(
-                self.F, self.indptr, self.indices, self.sumF, order_d,
-                self.cfg, out=(grad, llh, best), n_mfma=n_mfma_d,
-            )
+            if self.fused_ok:
+                n_mfma_d = n_d if self.n_mfma == int(self.order.numel()) else 0
+                ops.fused_grad_ls(
+                    self.F, self.indptr, self.indices, self.sumF, order_d,
+                    self.cfg, out=(grad, llh, best), n_mfma=n_mfma_d,
+                )
+            else:
+                # large-K hub remainder: chunked K1 + K2, subset launches
+                ops.edge_grad_llh(
+                    self.F, self.indptr, self.indices, self.sumF, order_d,
+                    self.cfg, out=(grad, llh),
+                )
+                ops.linesearch(
+                    self.F, self.indptr, self.indices, self.sumF, grad,
+                    llh, order_d, self.cfg, out=best,
+                )
         pack = ops.sparse_sweep_part(
             self.F, self.indptr, self.indices, self.sumF, order_s,
             soffset, sidx, sval, scount, cs, goffset, n_s * cap, cap, llh,

@@ -159,7 +159,9 @@ void linesearch(torch::Tensor F, torch::Tensor indptr, torch::Tensor indices,
   CHECK_IN(order, torch::kInt32);
   CHECK_IN(ladder, torch::kFloat32);
   CHECK_IN(best, torch::kFloat32);
-  const int n_local = (int)indptr.size(0) - 1;
+  // grid = #nodes listed in `order` (may be a subset: the sparse path's
+  // dense hub remainder); unlisted best[] entries stay untouched
+  const int n_blocks = (int)order.size(0);
   const int K = (int)F.size(1);
   const auto ip = reinterpret_cast<const long long*>(indptr.data_ptr<int64_t>());
   if (is_bf16(F)) {
@@ -167,16 +169,16 @@ void linesearch(torch::Tensor F, torch::Tensor indptr, torch::Tensor indices,
     launch_k2_bf16(F.data_ptr(), ip, indices.data_ptr<int>(),
                    sumF.data_ptr<float>(), grad.data_ptr<float>(),
                    llh.data_ptr<double>(), order.data_ptr<int>(),
-                   ladder.data_ptr<float>(), best.data_ptr<float>(), n_local,
-                   K, (int)ladder.size(0), (float)alpha, (float)min_p,
-                   (float)max_p, (float)min_f, (float)max_f,
+                   ladder.data_ptr<float>(), best.data_ptr<float>(),
+                   n_blocks, K, (int)ladder.size(0), (float)alpha,
+                   (float)min_p, (float)max_p, (float)min_f, (float)max_f,
                    current_stream());
   } else {
     TORCH_CHECK(K % 4 == 0, "K must be padded to a multiple of 4");
     launch_k2(F.data_ptr<float>(), ip, indices.data_ptr<int>(),
               sumF.data_ptr<float>(), grad.data_ptr<float>(),
               llh.data_ptr<double>(), order.data_ptr<int>(),
-              ladder.data_ptr<float>(), best.data_ptr<float>(), n_local, K,
+              ladder.data_ptr<float>(), best.data_ptr<float>(), n_blocks, K,
               (int)ladder.size(0), (float)alpha, (float)min_p, (float)max_p,
               (float)min_f, (float)max_f, current_stream());
   }

@@ -149,10 +149,16 @@ def linesearch(
     llh: torch.Tensor,
     order: torch.Tensor,
     cfg: BigClamConfig,
+    out: torch.Tensor = None,
 ) -> torch.Tensor:
+    """K2 over the nodes listed in ``order`` (subset-capable like K1);
+    with ``out`` the results land in the caller's buffer at the listed
+    positions."""
     ext = ensure_loaded()
     n_local = len(indptr) - 1
-    best = torch.empty(n_local, device=F.device, dtype=torch.float32)
+    best = out if out is not None else torch.empty(
+        n_local, device=F.device, dtype=torch.float32
+    )
     ladder = _ladder(cfg, F.device)
     ext.linesearch(
         F,
