@@ -331,7 +331,7 @@ class ShardState:
         both subset-capable).  BIGCLAM_SPARSE=0 disables."""
         return (
             self.use_hip
-            and self.sparse_cap >= 256
+            and self.sparse_cap >= 64
             and os.environ.get("BIGCLAM_SPARSE", "1") != "0"
         )
 

@@ -645,11 +645,15 @@ def test_sparse_large_k_matches_dense(monkeypatch):
     on a partially-converged K=17000 bf16 state."""
     g = rmat_graph(9, 5.0, seed=35)
     cfg = BigClamConfig(k=17000, device="cuda", dtype="bf16", seed=3,
-                        max_sweeps=8, tol=0.0)
+                        max_sweeps=12, tol=0.0)
     monkeypatch.setenv("BIGCLAM_SPARSE", "0")
     tr = Trainer(g, cfg, rank=0, world_size=1, device=torch.device("cuda"))
-    tr.fit(init="seed")
+    # random init (seed init at K >> #seeds pads DENSE Bernoulli columns);
+    # the early sweeps clamp most entries to exact zero
+    tr.fit(init="random")
     st = tr.state
+    nnz = float((st.F_local_k != 0).float().mean().item())
+    assert nnz < 0.2, f"fixture not sparse enough: {nnz}"
     assert st.sparse_cap >= 256  # large-K sparse coverage engaged
     F0 = st.F.clone()
     sumF0 = st.sumF.clone()
