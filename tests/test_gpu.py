@@ -648,12 +648,20 @@ def test_sparse_large_k_matches_dense(monkeypatch):
                         max_sweeps=12, tol=0.0)
     monkeypatch.setenv("BIGCLAM_SPARSE", "0")
     tr = Trainer(g, cfg, rank=0, world_size=1, device=torch.device("cuda"))
-    # random init (seed init at K >> #seeds pads DENSE Bernoulli columns);
-    # the early sweeps clamp most entries to exact zero
+    # random init, then threshold to ~1.5% density (any valid F state is
+    # fair game for the equality check; small graphs don't converge to
+    # the <1% density the real configs reach)
     tr.fit(init="random")
     st = tr.state
+    with torch.no_grad():
+        Fl = st.F_local_k.float()
+        cut = torch.quantile(Fl.flatten()[:: 97], 0.985)
+        st.F[: st.n_local, : cfg.k] = torch.where(
+            Fl > cut, Fl, torch.zeros_like(Fl)
+        ).to(st.storage_dtype)
+        st.sumF = st.F_local.float().sum(0)
     nnz = float((st.F_local_k != 0).float().mean().item())
-    assert nnz < 0.2, f"fixture not sparse enough: {nnz}"
+    assert nnz < 0.05, f"fixture not sparse enough: {nnz}"
     assert st.sparse_cap >= 256  # large-K sparse coverage engaged
     F0 = st.F.clone()
     sumF0 = st.sumF.clone()
