@@ -88,6 +88,23 @@ class Trainer:
 
     def init_F(self, kind: str = "seed"):
         s = self.shard
+        if kind == "seed" and self.state.use_hip:
+            seeds = self.seeds()[: self.cfg.k]
+            if len(seeds) == self.cfg.k:
+                # K6 device path: no pad columns needed -> the whole init
+                # scatters on-GPU (the host path materialized the full
+                # [n_local, k] fp32 slice + H2D — the dominant fit-wall
+                # cost at the headline configs)
+                from ..ops import hip as hip_ops
+
+                hip_ops.seed_init_device(
+                    self.state, self.graph, seeds,
+                    self.cfg.init_include_seed,
+                )
+                st = self.state
+                st.sumF = st.F_local.float().sum(dim=0)
+                comm.all_reduce_(st.sumF)
+                return
         if kind == "seed":
             F_local = seed_init_local_F(
                 self.graph,

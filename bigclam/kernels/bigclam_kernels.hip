@@ -2811,6 +2811,46 @@ __global__ void __launch_bounds__(BLOCK) k3s_commit_t(
   }
 }
 
+// ------------------------------------------------------------------- K6
+//
+// Seed-init F scatter (codes/bigclamv3-7.scala:60-87): community c's
+// initial members are the neighbors of the c-th ranked seed —
+// F[v, c] = 1 for v in N(seed_c) (plus the seed itself in the v2
+// variant).  The caller uploads the seeds' COMPACT adjacency (k rows,
+// global ids) — a few hundred KB — and the kernel scatters into the
+// rank's F slice; rows outside [start, stop) are skipped, so the same
+// launch works at any world size.  One block per seed/community.
+
+template <bool BF16>
+__global__ void __launch_bounds__(BLOCK) k6_seed_init_t(
+    void* __restrict__ Fp, int K, const long long* __restrict__ sindptr,
+    const long long* __restrict__ snbrs, const long long* __restrict__ seeds,
+    int n_seeds, long long start, long long stop, int include_seed) {
+  const int c = blockIdx.x;
+  if (c >= n_seeds) return;
+  const long long b0 = sindptr[c];
+  const long long b1 = sindptr[c + 1];
+  for (long long j = b0 + threadIdx.x; j < b1; j += BLOCK) {
+    const long long v = snbrs[j];
+    if (v < start || v >= stop) continue;
+    const size_t at = (size_t)(v - start) * K + c;
+    if (BF16)
+      reinterpret_cast<unsigned short*>(Fp)[at] = pack1_bf16_rne(1.0f);
+    else
+      reinterpret_cast<float*>(Fp)[at] = 1.0f;
+  }
+  if (include_seed && threadIdx.x == 0) {
+    const long long s = seeds[c];
+    if (s >= start && s < stop) {
+      const size_t at = (size_t)(s - start) * K + c;
+      if (BF16)
+        reinterpret_cast<unsigned short*>(Fp)[at] = pack1_bf16_rne(1.0f);
+      else
+        reinterpret_cast<float*>(Fp)[at] = 1.0f;
+    }
+  }
+}
+
 // ------------------------------------------------------------------- K7
 //
 // Community extraction (codes/Bigclamv2.scala:223-230): node u belongs to
@@ -3353,6 +3393,23 @@ extern "C" void launch_k3s(void* F, int bf16, const int* order, int n_blocks,
     hipLaunchKernelGGL((k3s_commit_t<false>), dim3(n_blocks), dim3(BLOCK), 0,
                        stream, F, K, order, goffset, gidx, gval, gcount,
                        best, min_f, max_f);
+  }
+  HIP_CHECK(hipGetLastError());
+}
+
+extern "C" void launch_k6(void* F, int bf16, int K, const long long* sindptr,
+                          const long long* snbrs, const long long* seeds,
+                          int n_seeds, long long start, long long stop,
+                          int include_seed, hipStream_t stream) {
+  if (n_seeds == 0) return;
+  if (bf16) {
+    hipLaunchKernelGGL((k6_seed_init_t<true>), dim3(n_seeds), dim3(BLOCK), 0,
+                       stream, F, K, sindptr, snbrs, seeds, n_seeds, start,
+                       stop, include_seed);
+  } else {
+    hipLaunchKernelGGL((k6_seed_init_t<false>), dim3(n_seeds), dim3(BLOCK),
+                       0, stream, F, K, sindptr, snbrs, seeds, n_seeds,
+                       start, stop, include_seed);
   }
   HIP_CHECK(hipGetLastError());
 }

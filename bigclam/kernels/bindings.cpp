@@ -55,6 +55,9 @@ extern "C" void launch_k1_chunked(const void*, int, const long long*,
                                   const int*, const float*, const int*,
                                   float*, float*, double*, int, int, float,
                                   float, hipStream_t);
+extern "C" void launch_k6(void*, int, int, const long long*,
+                          const long long*, const long long*, int, long long,
+                          long long, int, hipStream_t);
 extern "C" void launch_k7_count(const void*, int, int, int, int, float, int*,
                                 hipStream_t);
 extern "C" void launch_k7_fill(const void*, int, int, int, int, float,
@@ -435,6 +438,27 @@ void sparse_commit(torch::Tensor F, torch::Tensor order,
              (float)max_f, current_stream());
 }
 
+// K6: seed-init F scatter from the seeds' compact adjacency (see
+// k6_seed_init_t); F_local is the rank's [n_local, kp] slice.
+void seed_init(torch::Tensor F_local, torch::Tensor sindptr,
+               torch::Tensor snbrs, torch::Tensor seeds, int64_t start,
+               int64_t stop, bool include_seed) {
+  CHECK_F(F_local);
+  CHECK_IN(sindptr, torch::kInt64);
+  CHECK_IN(snbrs, torch::kInt64);
+  CHECK_IN(seeds, torch::kInt64);
+  const int n_seeds = (int)seeds.size(0);
+  TORCH_CHECK(sindptr.size(0) == n_seeds + 1);
+  TORCH_CHECK(stop - start == F_local.size(0));
+  launch_k6(F_local.data_ptr(), is_bf16(F_local) ? 1 : 0,
+            (int)F_local.size(1),
+            reinterpret_cast<const long long*>(sindptr.data_ptr<int64_t>()),
+            reinterpret_cast<const long long*>(snbrs.data_ptr<int64_t>()),
+            reinterpret_cast<const long long*>(seeds.data_ptr<int64_t>()),
+            n_seeds, (long long)start, (long long)stop,
+            include_seed ? 1 : 0, current_stream());
+}
+
 // K7 community extraction: two deterministic passes (count, then fill
 // after a host/torch prefix-sum) — see k7_membership in the .hip file.
 void extract_count(torch::Tensor F_local, int64_t k_true, double delta,
@@ -504,6 +528,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "K3S: sparse projected commit confined to the active set");
   m.def("edge_grad_llh_chunked", &edge_grad_llh_chunked,
         "K1 large-K: KD per-edge dots + KW chunked weighted accumulate");
+  m.def("seed_init", &seed_init,
+        "K6: seed-init F scatter (community c = neighbors of seed c)");
   m.def("extract_count", &extract_count,
         "K7 pass 1: per-row membership counts (threshold/argmax-fallback)");
   m.def("extract_fill", &extract_fill,

@@ -207,6 +207,37 @@ def apply_step_colsum(
     return partials.sum(dim=0)
 
 
+def seed_init_device(state, graph, seeds, include_seed: bool) -> None:
+    """K6: device-side seed-init scatter — F[v, c] = 1 for v in
+    N(seed_c) (codes/bigclamv3-7.scala:60-87).  The seeds' compact
+    adjacency (k rows) uploads in a few hundred KB; works at any world
+    size (off-shard rows are skipped in-kernel).  The caller handles
+    pad columns (only exist when #seeds < k) and the sumF refresh."""
+    import numpy as np
+
+    ext = ensure_loaded()
+    seeds = np.asarray(seeds, dtype=np.int64)
+    deg = (graph.indptr[seeds + 1] - graph.indptr[seeds]).astype(np.int64)
+    sindptr = np.concatenate([[0], np.cumsum(deg)])
+    offs = np.arange(int(deg.sum()), dtype=np.int64) - np.repeat(
+        np.cumsum(deg) - deg, deg
+    )
+    snbrs = graph.indices[
+        np.repeat(graph.indptr[seeds], deg) + offs
+    ].astype(np.int64)
+    dev = state.device
+    state.F.zero_()
+    ext.seed_init(
+        state.F_local,
+        torch.from_numpy(sindptr).to(dev),
+        torch.from_numpy(snbrs).to(dev),
+        torch.from_numpy(seeds).to(dev),
+        state.shard.start,
+        state.shard.stop,
+        include_seed,
+    )
+
+
 def conductance_full_graph(graph, device) -> "torch.Tensor":
     """K5: ego-net conductance of every node of the FULL graph on GPU.
 

@@ -675,3 +675,34 @@ def test_sparse_large_k_matches_dense(monkeypatch):
     torch.testing.assert_close(llh_s.sum(), llh_d.sum(), rtol=2e-5, atol=5.0)
     agree = (best_s == best_d).float().mean().item()
     assert agree > 0.9, agree
+
+
+@pytest.mark.parametrize("dtype", ["fp32", "bf16"])
+def test_k6_device_seed_init_matches_host(dtype):
+    """K6 device scatter == the host seed_init_local_F (no pad columns),
+    including the sumF refresh and the include_seed variant."""
+    from bigclam.core.init import seed_init_local_F
+
+    g = rmat_graph(10, 6.0, seed=44)
+    cfg = BigClamConfig(k=48, device="cuda", dtype=dtype, seed=2)
+    tr = Trainer(g, cfg, rank=0, world_size=1, device=torch.device("cuda"))
+    seeds = tr.seeds()
+    assert len(seeds) >= 48
+    tr.init_F("seed")  # takes the K6 device path (no pads needed)
+    F_dev = tr.state.F_local_k.float().cpu().numpy()
+    F_host = seed_init_local_F(g, 48, 0, g.num_nodes, seeds=seeds)
+    np.testing.assert_array_equal(F_dev, F_host)
+    np.testing.assert_allclose(
+        tr.state.sumF[:48].cpu().numpy(), F_host.sum(0), rtol=1e-6
+    )
+    # v2 variant: seed itself included
+    cfg2 = BigClamConfig(k=48, device="cuda", dtype=dtype, seed=2,
+                         init_include_seed=True)
+    tr2 = Trainer(g, cfg2, rank=0, world_size=1, device=torch.device("cuda"))
+    tr2._seeds = seeds
+    tr2.init_F("seed")
+    F_host2 = seed_init_local_F(g, 48, 0, g.num_nodes, seeds=seeds,
+                                include_seed=True)
+    np.testing.assert_array_equal(
+        tr2.state.F_local_k.float().cpu().numpy(), F_host2
+    )
