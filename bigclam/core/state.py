@@ -361,6 +361,30 @@ class ShardState:
             self._last_nnz = None
             g, l, b = self.fused_grad_ls_overlap(halo_work)
             return g, l, b, None
+        if halo_work is not None:  # waited HERE so the fallback can't
+            halo_work.wait()       # ever run on stale halo rows
+            halo_work = None
+        try:
+            return self._grad_ls_sparse(halo_work)
+        except RuntimeError as e:
+            # The sparse path is an OPTIMIZATION of the dense sweep (same
+            # math); if it ever fails at runtime (e.g. an untested shape
+            # on a fresh topology), degrade loudly to the dense path for
+            # the rest of the session instead of killing an unattended
+            # multi-GPU run.  Kernel-load failures still raise (they hit
+            # the dense path identically).
+            import warnings
+
+            warnings.warn(
+                f"sparse sweep failed ({e}); falling back to the dense "
+                "path for this session (BIGCLAM_SPARSE=0)"
+            )
+            os.environ["BIGCLAM_SPARSE"] = "0"
+            self._last_nnz = None
+            g, l, b = self.fused_grad_ls_overlap(None)
+            return g, l, b, None
+
+    def _grad_ls_sparse(self, halo_work):
         ops = _hip_ops()
         # KAF needs fresh halo rows: wait before scanning F
         if halo_work is not None:

@@ -706,3 +706,24 @@ def test_k6_device_seed_init_matches_host(dtype):
     np.testing.assert_array_equal(
         tr2.state.F_local_k.float().cpu().numpy(), F_host2
     )
+
+
+def test_cli_fit_end_to_end_gpu(tmp_path):
+    """The full CLI fit on GPU: shaped graph, seed init (K6 device path),
+    sparse-adaptive sweeps, sharded K7 extraction, checkpoint write."""
+    import json as _json
+
+    from bigclam.cli import main as cli_main
+
+    out = tmp_path / "comms.txt"
+    ck = tmp_path / "ck"
+    rc = cli_main([
+        "fit", "shaped:3000:9000", "--k", "64", "--init", "seed",
+        "--max-sweeps", "30", "--out", str(out),
+        "--checkpoint-dir", str(ck), "--quiet",
+    ])
+    assert rc == 0
+    lines = out.read_text().strip().splitlines()
+    assert len(lines) > 10  # non-empty communities written
+    meta = _json.load(open(ck / "meta.json"))
+    assert meta["k"] == 64 and meta["n"] == 3000
