@@ -131,6 +131,9 @@ class ShardState:
         self._indices64: Optional[torch.Tensor] = None
         self._order64: Optional[torch.Tensor] = None
         self._last_nnz: Optional[torch.Tensor] = None  # from KAF scount
+        self._sp_scount: Optional[torch.Tensor] = None
+        self._dirty: Optional[torch.Tensor] = None  # uint8 [n_rows]
+        self._kaf_valid = False  # persistent support lists match F
         # per-edge source row (torch reference path); built lazily on CPU
         self._edge_src: Optional[torch.Tensor] = None
 
@@ -183,6 +186,7 @@ class ShardState:
     # ------------------------------------------------------------- model init
     def set_local_F(self, F_local: torch.Tensor):
         """Install owned rows (true-K width) and (re)compute the global sumF."""
+        self._kaf_valid = False
         self.F = torch.zeros(
             self.shard.n_rows,
             self.kp,
@@ -395,20 +399,32 @@ class ShardState:
         # Single host sync per sweep (the boolean order split below);
         # pools use a fixed per-row stride of `cap` so their sizes are
         # shape-derived, not data-dependent.
-        scount = torch.empty(n_rows, device=dev, dtype=torch.int32)
         if self._sp_soffset is None or self._sp_soffset.numel() != n_rows:
             self._sp_soffset = (
                 torch.arange(n_rows, device=dev, dtype=torch.int64) * cap
             )
             self._indices64 = self.indices.long()
             self._order64 = self.order.long()
+            self._sp_scount = torch.empty(
+                n_rows, device=dev, dtype=torch.int32
+            )
+            self._dirty = torch.ones(n_rows, device=dev, dtype=torch.uint8)
         soffset = self._sp_soffset
+        scount = self._sp_scount
         sidx = ops._pool(dev, "sidx", n_rows * cap, torch.int32)
         sval = ops._pool(dev, "sval", n_rows * cap, torch.float32)
-        # single KAF pass: writes counts AND fills lists for rows <= cap
+        # single KAF pass: counts AND lists for rows <= cap.  Incremental:
+        # rows unchanged since the last commit keep their persistent
+        # entries (the commit marks accepted rows dirty; halo rows are
+        # always dirty — refreshed by the exchange every sweep)
+        if self._kaf_valid:
+            dirty = self._dirty
+        else:
+            dirty = torch.empty(0, device=dev, dtype=torch.uint8)
         ops.ensure_loaded().sparse_support(
-            self.F, soffset, scount, sidx, sval, cap, True
+            self.F, soffset, scount, sidx, sval, cap, True, dirty
         )
+        self._kaf_valid = True
         self._last_nnz = scount[: self.n_local].sum()
         bound, cs = self.sparse_bounds(
             scount, self.indptr, self._indices64, self.n_local
@@ -464,11 +480,17 @@ class ShardState:
         """Commit a sweep's accepted steps: K3S for sparse-routed rows
         (first), then the dense commit + exact column-sum refresh (which
         reads the POST-commit F for every row)."""
+        accepted = steps  # full best: sparse + dense accepted rows
         if pack is not None:
             _hip_ops().sparse_commit(self.F_local, pack, pack["best"],
                                      self.cfg)
             steps = pack["steps_dense"]
         self.apply_step(grad, steps)
+        # incremental KAF bookkeeping: only accepted rows changed (halo
+        # rows stay always-dirty)
+        if self._dirty is not None:
+            self._dirty[: self.n_local] = (accepted > 0).to(torch.uint8)
+            self._kaf_valid = True
 
     def grad_llh_overlap(self, halo_work) -> Tuple[torch.Tensor, torch.Tensor]:
         """K1 overlapped with the in-flight halo exchange: interior nodes
@@ -533,6 +555,7 @@ class ShardState:
         recomputing the exact column sums + all-reduce (C12).  Recomputing
         (one read pass) keeps the sumF == colsum(F) invariant exact instead
         of accumulating incremental deltas."""
+        self._kaf_valid = False  # re-validated by apply_commit
         if self.use_hip and self.storage_dtype == torch.bfloat16:
             # fused commit + column sums: one pass over F (the separate
             # flow re-read F via an fp32 materialization, ~3.5 ms/sweep

@@ -2439,10 +2439,15 @@ template <bool BF16, bool FILL>
 __global__ void __launch_bounds__(BLOCK) kaf_support_t(
     const void* __restrict__ Fp, int n_rows, int K, int ldF, int cap,
     const long long* __restrict__ soffset, int* __restrict__ scount,
-    int* __restrict__ sidx, float* __restrict__ sval) {
+    int* __restrict__ sidx, float* __restrict__ sval,
+    const unsigned char* __restrict__ dirty) {
   __shared__ int scan[BLOCK];
   const int r = blockIdx.x;
   if (r >= n_rows) return;
+  // incremental mode: rows unchanged since the last commit keep their
+  // persistent counts/lists (the commit kernels only touch accepted
+  // rows; halo rows are always marked dirty at ws > 1)
+  if (dirty != nullptr && !dirty[r]) return;
   // chunk rounded to 8 so the row reads vectorize (uint4 of 8 bf16 /
   // two float4); K is padded (bf16: %8, fp32: %4 -> guard the last 4)
   const int chunk = (((K + BLOCK - 1) / BLOCK) + 7) & ~7;
@@ -3323,12 +3328,12 @@ extern "C" void launch_k1_chunked(const void* F, int bf16,
 extern "C" void launch_kaf(const void* F, int bf16, int n_rows, int K,
                            int cap, const long long* soffset, int* scount,
                            int* sidx, float* sval, int fill,
-                           hipStream_t stream) {
+                           const unsigned char* dirty, hipStream_t stream) {
   if (n_rows == 0) return;
 #define KAF_CASE(B, FI)                                                     \
   hipLaunchKernelGGL((kaf_support_t<B, FI>), dim3(n_rows), dim3(BLOCK), 0,  \
                      stream, F, n_rows, K, K, cap, soffset, scount, sidx,   \
-                     sval)
+                     sval, dirty)
   if (bf16) {
     if (fill)
       KAF_CASE(true, true);

@@ -345,7 +345,7 @@ void edge_grad_llh_chunked(torch::Tensor F, torch::Tensor indptr,
 // Sparse-adaptive sweep (KAF/K1S/K2S/K3S — docs/sparse_sweep_design.md).
 extern "C" void launch_kaf(const void*, int, int, int, int,
                            const long long*, int*, int*, float*, int,
-                           hipStream_t);
+                           const unsigned char*, hipStream_t);
 extern "C" void launch_kfs(const void*, int, const long long*, const int*,
                            const float*, const int*, int, const long long*,
                            const int*, const float*, const int*,
@@ -363,7 +363,8 @@ static const long long* i64p(const torch::Tensor& t) {
 
 void sparse_support(torch::Tensor F, torch::Tensor soffset,
                     torch::Tensor scount, torch::Tensor sidx,
-                    torch::Tensor sval, int64_t cap, bool fill) {
+                    torch::Tensor sval, int64_t cap, bool fill,
+                    torch::Tensor dirty) {
   CHECK_F(F);
   CHECK_IN(soffset, torch::kInt64);
   CHECK_IN(scount, torch::kInt32);
@@ -372,9 +373,15 @@ void sparse_support(torch::Tensor F, torch::Tensor soffset,
   const int n_rows = (int)F.size(0);
   const int K = (int)F.size(1);
   TORCH_CHECK(scount.size(0) == n_rows && soffset.size(0) == n_rows);
+  const unsigned char* dp = nullptr;
+  if (dirty.numel()) {
+    TORCH_CHECK(dirty.scalar_type() == torch::kUInt8 &&
+                dirty.is_contiguous() && dirty.size(0) == n_rows);
+    dp = dirty.data_ptr<uint8_t>();
+  }
   launch_kaf(F.data_ptr(), is_bf16(F) ? 1 : 0, n_rows, K, (int)cap,
              i64p(soffset), scount.data_ptr<int>(), sidx.data_ptr<int>(),
-             sval.data_ptr<float>(), fill ? 1 : 0, current_stream());
+             sval.data_ptr<float>(), fill ? 1 : 0, dp, current_stream());
 }
 
 void sparse_fused(torch::Tensor F, torch::Tensor indptr,
