@@ -365,12 +365,14 @@ class ShardState:
             self._last_nnz = None
             g, l, b = self.fused_grad_ls_overlap(halo_work)
             return g, l, b, None
-        if halo_work is not None:  # waited HERE so the fallback can't
-            halo_work.wait()       # ever run on stale halo rows
-            halo_work = None
         try:
             return self._grad_ls_sparse(halo_work)
         except RuntimeError as e:
+            if halo_work is not None:
+                try:  # Work.wait() is idempotent; make sure the halo
+                    halo_work.wait()  # landed before the dense fallback
+                except Exception:
+                    pass
             # The sparse path is an OPTIMIZATION of the dense sweep (same
             # math); if it ever fails at runtime (e.g. an untested shape
             # on a fresh topology), degrade loudly to the dense path for
@@ -390,9 +392,8 @@ class ShardState:
 
     def _grad_ls_sparse(self, halo_work):
         ops = _hip_ops()
-        # KAF needs fresh halo rows: wait before scanning F
-        if halo_work is not None:
-            halo_work.wait()
+        # NOTE: the halo wait happens mid-KAF below — the local-section
+        # scan overlaps the in-flight all_to_all (disjoint rows)
         dev = self.device
         n_rows = self.F.shape[0]
         cap = self.sparse_cap
@@ -416,14 +417,36 @@ class ShardState:
         # single KAF pass: counts AND lists for rows <= cap.  Incremental:
         # rows unchanged since the last commit keep their persistent
         # entries (the commit marks accepted rows dirty; halo rows are
-        # always dirty — refreshed by the exchange every sweep)
+        # always dirty — refreshed by the exchange every sweep).
+        # ws > 1: the LOCAL section scans while the halo all_to_all is in
+        # flight (the exchange writes ONLY the halo rows — disjoint), then
+        # the halo section scans after the wait.
         if self._kaf_valid:
             dirty = self._dirty
         else:
             dirty = torch.empty(0, device=dev, dtype=torch.uint8)
-        ops.ensure_loaded().sparse_support(
-            self.F, soffset, scount, sidx, sval, cap, True, dirty
-        )
+        ext = ops.ensure_loaded()
+        n_loc = self.n_local
+        if halo_work is not None and n_rows > n_loc:
+            ext.sparse_support(
+                self.F[:n_loc], soffset[:n_loc], scount[:n_loc], sidx,
+                sval, cap, True,
+                dirty[:n_loc] if dirty.numel() else dirty,
+            )
+            halo_work.wait()
+            halo_work = None
+            ext.sparse_support(
+                self.F[n_loc:], soffset[n_loc:], scount[n_loc:], sidx,
+                sval, cap, True,
+                torch.empty(0, device=dev, dtype=torch.uint8),
+            )
+        else:
+            if halo_work is not None:
+                halo_work.wait()
+                halo_work = None
+            ext.sparse_support(
+                self.F, soffset, scount, sidx, sval, cap, True, dirty
+            )
         self._kaf_valid = True
         self._last_nnz = scount[: self.n_local].sum()
         bound, cs = self.sparse_bounds(
