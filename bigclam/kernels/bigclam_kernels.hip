@@ -2696,6 +2696,13 @@ __global__ void __launch_bounds__(BLOCK) kfs_sparse_t(
   __syncthreads();
   unsigned short* pmap = reinterpret_cast<unsigned short*>(gacc);
   for (int i = tid; i < ns; i += BLOCK) pmap[kS[i]] = (unsigned short)i;
+  __syncthreads();
+  {  // transform the staged entries' column ids to COMPACT POSITIONS in
+     // place (phase 4 then reads one LDS word per element instead of a
+     // nidx -> pmap dependent chain)
+    const int total = (int)(epos[e1] - p0);
+    for (int i = tid; i < total; i += BLOCK) nidx[i] = pmap[nidx[i]];
+  }
   // one packed reduction round: (gg, fs, ff) + the llh doubles
 #pragma unroll
   for (int off = 32; off > 0; off >>= 1) {
@@ -2746,7 +2753,8 @@ __global__ void __launch_bounds__(BLOCK) kfs_sparse_t(
         const int sv = (int)(epos[e + 1] - epos[e]);
         float x = 0.f;
         for (int t = 0; t < sv; ++t) {
-          const int pos = (int)pmap[nidx[base + t]];
+          const int pos = (int)nidx[base + t];  // positions since the
+                                                // in-place transform
           const float fv =
               BF16 ? vget<true>(nval_h[base + t]) : nval_f[base + t];
           const float fu = BF16 ? vget<true>(fuS_h[pos]) : fuS_f[pos];
