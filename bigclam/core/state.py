@@ -132,6 +132,9 @@ class ShardState:
         self._order64: Optional[torch.Tensor] = None
         self._last_nnz: Optional[torch.Tensor] = None  # from KAF scount
         self._sp_scount: Optional[torch.Tensor] = None
+        self._sp_sidx: Optional[torch.Tensor] = None
+        self._sp_sval: Optional[torch.Tensor] = None
+        self._sp_cap: Optional[int] = None
         self._dirty: Optional[torch.Tensor] = None  # uint8 [n_rows]
         self._kaf_valid = False  # persistent support lists match F
         # per-edge source row (torch reference path); built lazily on CPU
@@ -400,7 +403,12 @@ class ShardState:
         # Single host sync per sweep (the boolean order split below);
         # pools use a fixed per-row stride of `cap` so their sizes are
         # shape-derived, not data-dependent.
-        if self._sp_soffset is None or self._sp_soffset.numel() != n_rows:
+        if self._sp_soffset is None or self._sp_cap != cap:
+            # per-STATE persistent buffers: the incremental lists must
+            # survive across sweeps of THIS state without another state
+            # (e.g. select-k's next-K trainer) clobbering them — a
+            # shared module pool would.  Rebuilt if the cap env changes.
+            self._sp_cap = cap
             self._sp_soffset = (
                 torch.arange(n_rows, device=dev, dtype=torch.int64) * cap
             )
@@ -409,11 +417,18 @@ class ShardState:
             self._sp_scount = torch.empty(
                 n_rows, device=dev, dtype=torch.int32
             )
+            self._sp_sidx = torch.empty(
+                n_rows * cap, device=dev, dtype=torch.int32
+            )
+            self._sp_sval = torch.empty(
+                n_rows * cap, device=dev, dtype=torch.float32
+            )
             self._dirty = torch.ones(n_rows, device=dev, dtype=torch.uint8)
+            self._kaf_valid = False
         soffset = self._sp_soffset
         scount = self._sp_scount
-        sidx = ops._pool(dev, "sidx", n_rows * cap, torch.int32)
-        sval = ops._pool(dev, "sval", n_rows * cap, torch.float32)
+        sidx = self._sp_sidx
+        sval = self._sp_sval
         # single KAF pass: counts AND lists for rows <= cap.  Incremental:
         # rows unchanged since the last commit keep their persistent
         # entries (the commit marks accepted rows dirty; halo rows are
