@@ -335,12 +335,20 @@ class ShardState:
         KFS holds a dense K-float LDS accumulator, so kp is bounded by
         the 160 KB workgroup LDS (the dense-node remainder runs the
         fused kernel where it covers, else the separate chunked path —
-        both subset-capable).  BIGCLAM_SPARSE=0 disables."""
-        return (
-            self.use_hip
-            and self.sparse_cap >= 64
-            and os.environ.get("BIGCLAM_SPARSE", "1") != "0"
-        )
+        both subset-capable).  The cap-strided support/grad pools cost
+        ~16·n_rows·cap bytes — at HBM-filling shapes (e.g. the 100M-edge
+        R-MAT ws=8 slice: 266 GB of pools next to a 207 GB model) that
+        budget doesn't exist, so the path stays off there.
+        BIGCLAM_SPARSE=0 disables."""
+        if (
+            not self.use_hip
+            or self.sparse_cap < 64
+            or os.environ.get("BIGCLAM_SPARSE", "1") == "0"
+        ):
+            return False
+        pool_bytes = 16 * self.F.shape[0] * self.sparse_cap
+        total = torch.cuda.get_device_properties(self.device).total_memory
+        return pool_bytes < 0.15 * total
 
     @staticmethod
     def sparse_bounds(scount: torch.Tensor, indptr: torch.Tensor,
