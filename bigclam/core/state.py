@@ -137,6 +137,7 @@ class ShardState:
         self._sp_cap: Optional[int] = None
         self._dirty: Optional[torch.Tensor] = None  # uint8 [n_rows]
         self._kaf_valid = False  # persistent support lists match F
+        self._side_stream = None  # hub-remainder overlap (sparse path)
         # per-edge source row (torch reference path); built lazily on CPU
         self._edge_src: Optional[torch.Tensor] = None
 
@@ -494,27 +495,41 @@ class ShardState:
         best = torch.empty(n, device=dev, dtype=torch.float32)
         n_d = int(order_d.numel())
         if n_d:
-            if self.fused_ok:
-                n_mfma_d = n_d if self.n_mfma == int(self.order.numel()) else 0
-                ops.fused_grad_ls(
-                    self.F, self.indptr, self.indices, self.sumF, order_d,
-                    self.cfg, out=(grad, llh, best), n_mfma=n_mfma_d,
-                )
-            else:
-                # large-K hub remainder: chunked K1 + K2, subset launches
-                ops.edge_grad_llh(
-                    self.F, self.indptr, self.indices, self.sumF, order_d,
-                    self.cfg, out=(grad, llh),
-                )
-                ops.linesearch(
-                    self.F, self.indptr, self.indices, self.sumF, grad,
-                    llh, order_d, self.cfg, out=best,
-                )
+            # the hub remainder is a FEW blocks (often one giant hub
+            # whose serial block latency is ~1 ms) — run it on a side
+            # stream so it overlaps the 300k-block KFS launch below.
+            # Independent: both only READ F/sumF; grad/llh/best writes
+            # are row-disjoint by the routing split.
+            if self._side_stream is None:
+                self._side_stream = torch.cuda.Stream(device=dev)
+            self._side_stream.wait_stream(torch.cuda.current_stream(dev))
+            with torch.cuda.stream(self._side_stream):
+                if self.fused_ok:
+                    n_mfma_d = (
+                        n_d if self.n_mfma == int(self.order.numel()) else 0
+                    )
+                    ops.fused_grad_ls(
+                        self.F, self.indptr, self.indices, self.sumF,
+                        order_d, self.cfg, out=(grad, llh, best),
+                        n_mfma=n_mfma_d,
+                    )
+                else:
+                    # large-K hub remainder: chunked K1 + K2 subsets
+                    ops.edge_grad_llh(
+                        self.F, self.indptr, self.indices, self.sumF,
+                        order_d, self.cfg, out=(grad, llh),
+                    )
+                    ops.linesearch(
+                        self.F, self.indptr, self.indices, self.sumF, grad,
+                        llh, order_d, self.cfg, out=best,
+                    )
         pack = ops.sparse_sweep_part(
             self.F, self.indptr, self.indices, self.sumF, order_s,
             soffset, sidx, sval, scount, cs, goffset, n_s * cap, cap, llh,
             best, self.cfg,
         )
+        if n_d:
+            torch.cuda.current_stream(dev).wait_stream(self._side_stream)
         pack["best"] = best
         # dense commit must skip sparse rows (their grad rows are unset)
         steps_dense = best.clone()
