@@ -204,14 +204,89 @@ class ShardState:
         comm.all_reduce_(self.sumF)
 
     # ---------------------------------------------------------- communication
+    def _halo_compress_on(self) -> bool:
+        """Compressed halo: send (col, val) nonzeros instead of dense
+        rows.  Engages when F is sparse enough that the packed stream
+        (8 B/nnz) beats the dense row bytes — the xGMI volume drops with
+        the density (~250x at the converged headline config).
+
+        The decision MUST be identical on every rank (the two paths run
+        different collectives), so it comes from an allreduce of
+        (have-counts, nnz) — one tiny extra collective per sweep, only
+        at ws > 1.  BIGCLAM_HALO_COMPRESS=1/0 forces (same env on every
+        rank under torchrun)."""
+        env = os.environ.get("BIGCLAM_HALO_COMPRESS")
+        if env is not None:
+            return env != "0"
+        flag = torch.zeros(2, dtype=torch.float64, device=self.device)
+        if self._last_nnz is not None:
+            flag[0] = 1.0
+            flag[1] = float(self._last_nnz.item())
+        comm.all_reduce_(flag)
+        if float(flag[0].item()) < self.shard.world_size:
+            return False  # some rank has no fresh counts: stay dense
+        density = float(flag[1].item()) / max(
+            1, self.shard.n_total * self.kp
+        )
+        # transport = ~8 B/nnz vs kp*esize dense; engage with ~2x margin
+        return density < 0.1 * self.F.element_size()
+
+    def _halo_exchange_compressed(self):
+        """C8 compressed: two all-to-alls (per-row nnz counts, then the
+        packed col/val streams) + a scatter into the zeroed halo section.
+        Synchronous (the volumes are tiny once this path engages)."""
+        st = self.shard
+        dev = self.device
+        Fs = self.F_local.index_select(0, self.send_idx).float()
+        mask = Fs != 0
+        c_send = mask.sum(1).to(torch.int64)
+        c_recv = torch.empty(st.n_halo, device=dev, dtype=torch.int64)
+        comm.all_to_all(
+            c_recv, c_send, self.recv_splits, self.send_splits
+        )
+        nz = mask.nonzero(as_tuple=False)  # row-major: rows grouped by peer
+        cols = nz[:, 1].to(torch.int32)
+        vals = Fs[nz[:, 0], nz[:, 1]]
+        # per-peer element splits (host): segment sums of the row counts
+        cs_s = torch.cat(
+            [torch.zeros(1, dtype=torch.int64, device=dev),
+             torch.cumsum(c_send, 0)]
+        )
+        cs_r = torch.cat(
+            [torch.zeros(1, dtype=torch.int64, device=dev),
+             torch.cumsum(c_recv, 0)]
+        )
+        sb = np.concatenate([[0], np.cumsum(self.send_splits)])
+        rb = np.concatenate([[0], np.cumsum(self.recv_splits)])
+        vs = (cs_s[sb[1:]] - cs_s[sb[:-1]]).cpu().tolist()
+        vr_t = cs_r[rb[1:]] - cs_r[rb[:-1]]
+        vr = vr_t.cpu().tolist()
+        total_r = int(sum(vr))
+        cols_r = torch.empty(total_r, device=dev, dtype=torch.int32)
+        vals_r = torch.empty(total_r, device=dev, dtype=torch.float32)
+        comm.all_to_all(cols_r, cols, vr, vs)
+        comm.all_to_all(vals_r, vals, vr, vs)
+        # unpack: zero the halo section, scatter (row, col) = val
+        halo = self.F[self.n_local :]
+        halo.zero_()
+        rows_r = torch.repeat_interleave(
+            torch.arange(st.n_halo, device=dev, dtype=torch.int64), c_recv
+        )
+        halo[rows_r, cols_r.to(torch.int64)] = vals_r.to(self.storage_dtype)
+        return None
+
     def halo_exchange(self, async_op: bool = False):
         """C8: refresh halo rows with peers' current F rows (p2p all-to-all).
 
         With ``async_op`` returns the Work handle (None when nothing to
         exchange); the send buffer stays referenced until the next call.
+        When F is sparse (tracked by the KAF counts), the compressed
+        variant sends only the nonzeros.
         """
         if self.shard.world_size == 1 or self.shard.n_halo == 0:
             return None
+        if self._halo_compress_on():
+            return self._halo_exchange_compressed()
         send = self.F_local.index_select(0, self.send_idx)
         recv = self.F[self.n_local :]
         if self.storage_dtype == torch.bfloat16 and comm.backend() == "gloo":
