@@ -72,18 +72,18 @@ class Trainer:
         conductance runs as the K5 HIP kernel; the ranking itself is a
         trivial host pass."""
         if self._seeds is None:
-            cond = None
-            if self.state.use_hip:
+            if self.state.use_hip and not self.cfg.seed_rank_compat:
+                from ..core.init import conductance_ranking_device
                 from ..ops import hip as hip_ops
 
-                cond = (
-                    hip_ops.conductance_full_graph(self.graph, self.state.device)
-                    .cpu()
-                    .numpy()
+                cond_t = hip_ops.conductance_full_graph(
+                    self.graph, self.state.device
                 )
-            self._seeds = conductance_ranking(
-                self.graph, compat=self.cfg.seed_rank_compat, cond=cond
-            )
+                self._seeds = conductance_ranking_device(self.graph, cond_t)
+            else:
+                self._seeds = conductance_ranking(
+                    self.graph, compat=self.cfg.seed_rank_compat
+                )
         return self._seeds
 
     def init_F(self, kind: str = "seed"):

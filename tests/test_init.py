@@ -113,3 +113,19 @@ def test_seed_init_pad_memory_bounded_at_10M_nodes():
     a = seed_init_local_F(g, k, n - 20000, n - 12000, seeds=seeds, rng_seed=3)
     b = seed_init_local_F(g, k, n - 12000, n - 4000, seeds=seeds, rng_seed=3)
     np.testing.assert_array_equal(np.concatenate([a, b]), sl)
+
+
+def test_conductance_ranking_device_matches_host():
+    """The device ranking (rank + segment-min) == the host lexsort path
+    (intended semantics), on CPU tensors."""
+    import torch
+
+    from bigclam.core.init import conductance_ranking_device, conductances
+    from bigclam.io import rmat_graph
+
+    for seed in (3, 9):
+        g = rmat_graph(9, 5.0, seed=seed)
+        cond = conductances(g)
+        host = conductance_ranking(g, compat=False, cond=cond)
+        dev = conductance_ranking_device(g, torch.from_numpy(cond))
+        np.testing.assert_array_equal(dev, host)
