@@ -618,14 +618,16 @@ class ShardState:
         reads the POST-commit F for every row)."""
         accepted = steps  # full best: sparse + dense accepted rows
         if pack is not None:
+            # K3S commits AND rewrites the sparse rows' support lists,
+            # so only DENSE-committed rows need a KAF rescan
             _hip_ops().sparse_commit(self.F_local, pack, pack["best"],
-                                     self.cfg)
+                                     self, self.cfg)
             steps = pack["steps_dense"]
         self.apply_step(grad, steps)
-        # incremental KAF bookkeeping: only accepted rows changed (halo
-        # rows stay always-dirty)
+        # incremental KAF bookkeeping (halo rows stay always-dirty)
         if self._dirty is not None:
-            self._dirty[: self.n_local] = (accepted > 0).to(torch.uint8)
+            dirty_rows = steps if pack is not None else accepted
+            self._dirty[: self.n_local] = (dirty_rows > 0).to(torch.uint8)
             self._kaf_valid = True
 
     def grad_llh_overlap(self, halo_work) -> Tuple[torch.Tensor, torch.Tensor]:

@@ -2799,27 +2799,74 @@ __global__ void __launch_bounds__(BLOCK) kfs_sparse_t(
   }
 }
 
+// K3S also REWRITES the committed row's persistent support list
+// (sidx/sval/scount) so the next sweep's KAF only rescans rows the
+// DENSE path committed: the new row's support is a subset of S_u, and
+// every new value is computed right here.  List values store the
+// POST-ROUND value (bf16 pack->unpack), matching what KAF would read.
 template <bool BF16>
 __global__ void __launch_bounds__(BLOCK) k3s_commit_t(
     void* __restrict__ Fp, int ldF, const int* __restrict__ order,
     const long long* __restrict__ goffset, const int* __restrict__ gidx,
     const float* __restrict__ gval, const int* __restrict__ gcount,
-    const float* __restrict__ best, float min_f, float max_f) {
+    const float* __restrict__ best, const long long* __restrict__ soffset,
+    int* __restrict__ sidx, float* __restrict__ sval,
+    int* __restrict__ scount, int cap, float min_f, float max_f) {
+  __shared__ int scan[NWAVE];
   const int u = order[blockIdx.x];
   const float s = best[u];
   if (s <= 0.f) return;
   const int ns = gcount[blockIdx.x];
   const long long go = goffset[blockIdx.x];
-  for (int i = threadIdx.x; i < ns; i += BLOCK) {
+  const int tid = threadIdx.x;
+  const int lane = tid & (WAVE - 1);
+  const int wid = tid >> 6;
+  // pass 1: commit + per-thread new-nnz count (strided i keeps the
+  // ascending-k order within each thread's subsequence? NO — strided
+  // threads interleave; use contiguous chunks so the scan emission
+  // stays ascending)
+  const int chunk = (ns + BLOCK - 1) / BLOCK;
+  const int i0 = min(tid * chunk, ns);
+  const int i1 = min(i0 + chunk, ns);
+  int cnt = 0;
+  for (int i = i0; i < i1; ++i) {
     const int k = gidx[go + i];
     const float f = f_elem<BF16>(Fp, ldF, u, k);
-    const float nf =
+    float nf =
         __builtin_amdgcn_fmed3f(fmaf(s, gval[go + i], f), min_f, max_f);
     if (BF16) {
-      reinterpret_cast<unsigned short*>(Fp)[(size_t)u * ldF + k] =
-          pack1_bf16_rne(nf);
+      const unsigned short h = pack1_bf16_rne(nf);
+      reinterpret_cast<unsigned short*>(Fp)[(size_t)u * ldF + k] = h;
+      nf = __uint_as_float((u32)h << 16);  // post-round value
     } else {
       reinterpret_cast<float*>(Fp)[(size_t)u * ldF + k] = nf;
+    }
+    cnt += nf != 0.f;
+  }
+  // 2-barrier block scan (wave shuffles + cross-wave offsets)
+  int incl = cnt;
+#pragma unroll
+  for (int off = 1; off < WAVE; off <<= 1) {
+    const int v = __shfl_up(incl, off, WAVE);
+    if (lane >= off) incl += v;
+  }
+  if (lane == WAVE - 1) scan[wid] = incl;
+  __syncthreads();
+  int wbase = 0;
+#pragma unroll
+  for (int wv = 0; wv < NWAVE; ++wv)
+    if (wv < wid) wbase += scan[wv];
+  const int total = scan[0] + scan[1] + scan[2] + scan[3];
+  if (tid == 0) scount[u] = total;
+  if (total > cap) return;  // over-cap rows keep count only (KAF rule)
+  long long w = soffset[u] + (wbase + incl - cnt);
+  for (int i = i0; i < i1; ++i) {
+    const int k = gidx[go + i];
+    const float f = f_elem<BF16>(Fp, ldF, u, k);  // post-commit value
+    if (f != 0.f) {
+      sidx[w] = k;
+      sval[w] = f;
+      ++w;
     }
   }
 }
@@ -3395,17 +3442,19 @@ extern "C" void launch_kfs(const void* F, int bf16,
 extern "C" void launch_k3s(void* F, int bf16, const int* order, int n_blocks,
                            const long long* goffset, const int* gidx,
                            const float* gval, const int* gcount,
-                           const float* best, int K, float min_f,
-                           float max_f, hipStream_t stream) {
+                           const float* best, const long long* soffset,
+                           int* sidx, float* sval, int* scount, int cap,
+                           int K, float min_f, float max_f,
+                           hipStream_t stream) {
   if (n_blocks == 0) return;
   if (bf16) {
     hipLaunchKernelGGL((k3s_commit_t<true>), dim3(n_blocks), dim3(BLOCK), 0,
                        stream, F, K, order, goffset, gidx, gval, gcount,
-                       best, min_f, max_f);
+                       best, soffset, sidx, sval, scount, cap, min_f, max_f);
   } else {
     hipLaunchKernelGGL((k3s_commit_t<false>), dim3(n_blocks), dim3(BLOCK), 0,
                        stream, F, K, order, goffset, gidx, gval, gcount,
-                       best, min_f, max_f);
+                       best, soffset, sidx, sval, scount, cap, min_f, max_f);
   }
   HIP_CHECK(hipGetLastError());
 }

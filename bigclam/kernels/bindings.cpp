@@ -355,7 +355,8 @@ extern "C" void launch_kfs(const void*, int, const long long*, const int*,
                            float, float, hipStream_t);
 extern "C" void launch_k3s(void*, int, const int*, int, const long long*,
                            const int*, const float*, const int*,
-                           const float*, int, float, float, hipStream_t);
+                           const float*, const long long*, int*, float*,
+                           int*, int, int, float, float, hipStream_t);
 
 static const long long* i64p(const torch::Tensor& t) {
   return reinterpret_cast<const long long*>(t.data_ptr<int64_t>());
@@ -435,14 +436,22 @@ void sparse_fused(torch::Tensor F, torch::Tensor indptr,
 void sparse_commit(torch::Tensor F, torch::Tensor order,
                    torch::Tensor goffset, torch::Tensor gidx,
                    torch::Tensor gval, torch::Tensor gcount,
-                   torch::Tensor best, double min_f, double max_f) {
+                   torch::Tensor best, torch::Tensor soffset,
+                   torch::Tensor sidx, torch::Tensor sval,
+                   torch::Tensor scount, int64_t cap, double min_f,
+                   double max_f) {
   CHECK_F(F);
   CHECK_IN(best, torch::kFloat32);
+  CHECK_IN(soffset, torch::kInt64);
+  CHECK_IN(sidx, torch::kInt32);
+  CHECK_IN(sval, torch::kFloat32);
+  CHECK_IN(scount, torch::kInt32);
   launch_k3s(F.data_ptr(), is_bf16(F) ? 1 : 0, order.data_ptr<int>(),
              (int)order.size(0), i64p(goffset), gidx.data_ptr<int>(),
              gval.data_ptr<float>(), gcount.data_ptr<int>(),
-             best.data_ptr<float>(), (int)F.size(1), (float)min_f,
-             (float)max_f, current_stream());
+             best.data_ptr<float>(), i64p(soffset), sidx.data_ptr<int>(),
+             sval.data_ptr<float>(), scount.data_ptr<int>(), (int)cap,
+             (int)F.size(1), (float)min_f, (float)max_f, current_stream());
 }
 
 // K6: seed-init F scatter from the seeds' compact adjacency (see

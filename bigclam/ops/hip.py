@@ -313,11 +313,15 @@ def _pool(dev, name, size, dtype) -> torch.Tensor:
 
 
 def sparse_commit(F_local: torch.Tensor, pack: dict, best: torch.Tensor,
-                  cfg: BigClamConfig):
-    """K3S: projected commit confined to each routed node's active set."""
+                  state, cfg: BigClamConfig):
+    """K3S: projected commit confined to each routed node's active set,
+    rewriting the committed rows' persistent support lists in place (the
+    next sweep's KAF then skips them)."""
     ensure_loaded().sparse_commit(
         F_local, pack["order"], pack["goffset"], pack["gidx"], pack["gval"],
-        pack["gcount"], best, cfg.min_f, cfg.max_f,
+        pack["gcount"], best, state._sp_soffset, state._sp_sidx,
+        state._sp_sval, state._sp_scount, state._sp_cap,
+        cfg.min_f, cfg.max_f,
     )
 
 
