@@ -2580,9 +2580,8 @@ __global__ void __launch_bounds__(BLOCK) kfs_sparse_t(
   __shared__ float red[NWAVE];
   __shared__ int scan[NWAVE];
   __shared__ float s_lad[MAX_LS];
-  __shared__ double cllh[8][MAX_LS];
+  __shared__ double cllh[16][MAX_LS];
   __shared__ float cnt16[16][MAX_LS];
-  __shared__ float xpart[2][8][MAX_LS];
   __shared__ float red3[NWAVE][3];
   __shared__ double dred[NWAVE];
   __shared__ double sh_llh_base;
@@ -2742,21 +2741,18 @@ __global__ void __launch_bounds__(BLOCK) kfs_sparse_t(
   // reductions at all (the 17-shuffle butterflies were the phase's cost
   // at these tiny per-node sizes; measured in the r04 phase bisect).
   if (phases & 4) {
-    // (stripe, edge-slot, candidate) mapping: 2 stripes halve the
-    // serial per-thread entry chain; stripe partials combine via LDS
-    const int sj_ = tid >> 7;        // stripe 0/1
-    const int ej = (tid >> 4) & 7;   // edge slot 0..7
-    const int jc = tid & 15;         // candidate 0..15
+    const int ej = tid >> 4;  // edge slot 0..15
+    const int jc = tid & 15;  // candidate 0..15
     const float sj = s_lad[jc];
     double myllh = 0.0;
-    for (long long et = e0; et < e1; et += 8) {
-      const int ne = (int)min((long long)8, e1 - et);
-      float x = 0.f;
+    for (long long et = e0; et < e1; et += 16) {
+      const int ne = (int)min((long long)16, e1 - et);
       if (ej < ne && jc < n_ladder) {
         const long long e = et + ej;
         const int base = (int)(epos[e] - p0);
         const int sv = (int)(epos[e + 1] - epos[e]);
-        for (int t = sj_; t < sv; t += 2) {
+        float x = 0.f;
+        for (int t = 0; t < sv; ++t) {
           const int pos = (int)nidx[base + t];  // positions since the
                                                 // in-place transform
           const float fv =
@@ -2766,37 +2762,29 @@ __global__ void __launch_bounds__(BLOCK) kfs_sparse_t(
                                                   min_f, max_f);
           x = fmaf(c, fv, x);
         }
+        const float p = clamp_p(__expf(-x), min_p, max_p);
+        myllh += (double)log1pf(-p) + (double)x;
       }
-      xpart[sj_][ej][jc] = x;
-      __syncthreads();
-      if (sj_ == 0 && ej < ne && jc < n_ladder) {
-        const float xt = x + xpart[1][ej][jc];
-        const float p = clamp_p(__expf(-xt), min_p, max_p);
-        myllh += (double)log1pf(-p) + (double)xt;
-      }
-      __syncthreads();
     }
-    // node terms: thread (stripe = tid>>4, j=jc) over strided elements
-    const int nts = tid >> 4;  // 0..15
+    // node terms: thread (stripe=ej, j=jc) over strided elements
     float mynt = 0.f;
     if (jc < n_ladder) {
-      for (int i = nts; i < ns; i += 16) {
+      for (int i = ej; i < ns; i += 16) {
         const float fu = BF16 ? vget<true>(fuS_h[i]) : fuS_f[i];
         const float c = __builtin_amdgcn_fmed3f(fmaf(sj, gS[i], fu), min_f,
                                                 max_f);
         mynt = fmaf(c, fu - sumF[kS[i]], mynt);
       }
     }
-    if (sj_ == 0) cllh[ej][jc] = myllh;
-    cnt16[nts][jc] = mynt;
+    cllh[ej][jc] = myllh;
+    cnt16[ej][jc] = mynt;
     __syncthreads();
     if (tid < MAX_LS) {
       double trial = 0.0;
       float nt = 0.f;
 #pragma unroll
-      for (int e = 0; e < 8; ++e) trial += cllh[e][tid];
-#pragma unroll
       for (int e = 0; e < 16; ++e) {
+        trial += cllh[e][tid];
         nt += cnt16[e][tid];
       }
       const bool ok = (tid < n_ladder) &&
