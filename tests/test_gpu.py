@@ -730,3 +730,40 @@ def test_cli_fit_end_to_end_gpu(tmp_path):
     assert len(lines) > 10  # non-empty communities written
     meta = _json.load(open(ck / "meta.json"))
     assert meta["k"] == 64 and meta["n"] == 3000
+
+
+@pytest.mark.parametrize("sparse", ["0", "1"])
+def test_gpu_resume_trajectory(sparse, monkeypatch, tmp_path):
+    """Checkpoint at sweep 3, resume, continue on GPU: trajectory equals
+    the uninterrupted run — bitwise on the dense path (deterministic
+    kernels), to fp32-noise tolerance on the sparse path (its LDS-atomic
+    gradient order is run-dependent by design)."""
+    from bigclam.ckpt.checkpoint import resume, save_shard_checkpoint
+
+    monkeypatch.setenv("BIGCLAM_SPARSE", sparse)
+    monkeypatch.setenv("BIGCLAM_SPARSE_CAP", "1024")  # small-K routing
+    g = rmat_graph(10, 6.0, seed=61)
+
+    def mk(max_sweeps):
+        cfg = BigClamConfig(k=192, device="cuda", seed=8, tol=0.0,
+                            max_sweeps=max_sweeps)
+        return Trainer(g, cfg, rank=0, world_size=1,
+                       device=torch.device("cuda"))
+
+    tr_full = mk(6)
+    res_full = tr_full.fit(init="seed")
+    tr1 = mk(3)
+    res1 = tr1.fit(init="seed")
+    save_shard_checkpoint(str(tmp_path), tr1, sweep=res1.sweeps,
+                          llh=res1.llh)
+    tr2 = mk(3)
+    sweep0, llh0 = resume(str(tmp_path), tr2)
+    res2 = tr2.fit(skip_init=True, llh_old=llh0, sweep0=sweep0)
+    traj = res1.llh_history + res2.llh_history
+    if sparse == "0":
+        assert traj == res_full.llh_history
+    else:
+        for a, b in zip(traj, res_full.llh_history):
+            assert abs(a - b) < 5e-5 * max(1.0, abs(b)), (
+                traj, res_full.llh_history
+            )
