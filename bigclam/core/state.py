@@ -163,9 +163,9 @@ class ShardState:
         if self.use_hip:
             _hip_ops().ensure_loaded()  # fail loudly if the .so is missing
             # no K cap: shapes beyond the fused kernels' LDS coverage
-            # (fp32 > 8192, bf16 > 26000, or bf16 16384..26000 with MFMA
-            # disabled) take the separate chunked-K1 + K2 path
-            # (ops/hip.py edge_grad_llh_chunked)
+            # (fp32 > 8192, bf16 > 16384) take the chunked-K1 + K2 path
+            # (ops/hip.py edge_grad_llh_chunked; measured dispatch table
+            # in profiles/r02_largek_dispatch.md)
 
     # ------------------------------------------------------------------ util
     @property
@@ -445,8 +445,9 @@ class ShardState:
     def grad_ls_auto(self, halo_work):
         """Per-sweep adaptive dispatch: route nodes whose active-set bound
         (own support + sum of neighbor supports, exact upper bound on
-        |S_u|) fits under ``sparse_cap`` to the compact K1S/K2S kernels,
-        the rest to the dense fused path.  Returns
+        |S_u|) fits under ``sparse_cap`` to the fused sparse kernel
+        (KFS), the rest to the dense path (fused kernel, or chunked
+        K1 + K2 subsets above its caps).  Returns
         (grad, llh_nodes, best, sparse_pack|None)."""
         if not self.sparse_allowed:
             self._last_nnz = None

@@ -170,10 +170,11 @@ class Trainer:
 
     def pipelined_sweep(self, carry):
         """One iteration: commit of the carried steps (K3S for
-        sparse-routed rows + dense K3) -> async halo -> adaptive grad +
-        line search (compact K1S/K2S for nodes whose active-set bound
-        fits, the fused dense kernels for the rest; dense-only sweeps
-        keep the interior/boundary halo overlap).  Returns
+        sparse-routed rows, with in-place support-list rewrite, + dense
+        K3) -> async halo -> adaptive grad + line search (the fused
+        sparse kernel KFS for nodes whose active-set bound fits, the
+        dense kernels for the rest; dense-only sweeps keep the
+        interior/boundary halo overlap).  Returns
         (carry', llh_total_after_commit, committed_steps)."""
         st = self.state
         grad, steps, pack = carry
