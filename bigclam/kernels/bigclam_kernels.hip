@@ -17,6 +17,12 @@
 //                       GEMM tiles on v_mfma_f32_16x16x32_bf16 /
 //                       v_mfma_f32_16x16x4_f32 (dispatch is K-aware,
 //                       see launch_kf_mfma* and core/state.py)
+//   KD/KW kd_dot/kw_grad — chunked large-K K1 (no K cap; x-buffer split)
+//   KAF/KFS/K3S         — sparse-adaptive sweep on exact active column
+//                       sets (support compaction, fused compact
+//                       grad+Armijo, sparse commit + list rewrite)
+//   K6 k6_seed_init     — device seed-init scatter (scala:60-87)
+//   K7 k7_membership    — device community extraction (Bigclamv2:223-230)
 //
 // Design notes (measured rationale in profiles/r01_kernel_opt_log.md):
 //  * K1 is BLOCK-per-node: per edge the workgroup does one cooperative dot
