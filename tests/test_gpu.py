@@ -767,3 +767,39 @@ def test_gpu_resume_trajectory(sparse, monkeypatch, tmp_path):
             assert abs(a - b) < 5e-5 * max(1.0, abs(b)), (
                 traj, res_full.llh_history
             )
+
+
+def test_incremental_kaf_lists_match_full_rescan(monkeypatch):
+    """After several incremental sweeps (K3S list rewrites + dirty-row
+    skips), the persistent support counts/lists equal a forced FULL
+    rescan of the same F."""
+    monkeypatch.setenv("BIGCLAM_SPARSE", "1")
+    monkeypatch.setenv("BIGCLAM_SPARSE_CAP", "1024")
+    g = rmat_graph(10, 6.0, seed=63)
+    cfg = BigClamConfig(k=256, device="cuda", seed=5, tol=0.0, max_sweeps=12)
+    tr = Trainer(g, cfg, rank=0, world_size=1, device=torch.device("cuda"))
+    tr.fit(init="seed")  # several sweeps of incremental maintenance
+    st = tr.state
+    assert st._kaf_valid
+    sc_inc = st._sp_scount.clone()
+    sidx_inc = st._sp_sidx.clone()
+    sval_inc = st._sp_sval.clone()
+    st._kaf_valid = False  # force a full rescan of the SAME F
+    grad, llh, best, pack = st.grad_ls_auto(None)
+    assert pack is not None
+    torch.testing.assert_close(st._sp_scount, sc_inc, rtol=0, atol=0)
+    # compare list contents within each row's valid prefix
+    cap = st._sp_cap
+    sc = sc_inc.cpu().numpy()
+    for r in range(0, st.n_local, 37):
+        c = int(sc[r])
+        if c == 0 or c > cap:
+            continue
+        torch.testing.assert_close(
+            st._sp_sidx[r * cap : r * cap + c],
+            sidx_inc[r * cap : r * cap + c], rtol=0, atol=0,
+        )
+        torch.testing.assert_close(
+            st._sp_sval[r * cap : r * cap + c],
+            sval_inc[r * cap : r * cap + c], rtol=0, atol=0,
+        )
