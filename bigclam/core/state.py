@@ -398,9 +398,12 @@ class ShardState:
         env = os.environ.get("BIGCLAM_SPARSE_CAP")
         if env is not None:
             return int(env)
-        per = 12 if self.storage_dtype == torch.bfloat16 else 16
+        per = 16 if self.storage_dtype == torch.bfloat16 else 20
         nw4 = ((self.kp + 31) // 32) * 4
-        lds_cap = (160 * 1024 - 4096 - self.kp * 4 - nw4 - 64) // per
+        # KFS v10's compact accumulator: LDS = bitmap + cap*per.  Target
+        # <= 78 KB so at least 2 blocks/CU always fit (the K=5000
+        # headline sits at ~21 KB -> 7 blocks).
+        lds_cap = (78 * 1024 - 4096 - nw4) // per
         return max(0, min(self.kp // 4, lds_cap) & ~7)
 
     @property
@@ -408,10 +411,11 @@ class ShardState:
         """Active-column sweep (docs/sparse_sweep_design.md): exact
         per-node routing by active-set bound; pays off once F sparsifies
         (measured 0.4-4% converged density at the headline configs).
-        KFS holds a dense K-float LDS accumulator, so kp is bounded by
-        the 160 KB workgroup LDS (the dense-node remainder runs the
-        fused kernel where it covers, else the separate chunked path —
-        both subset-capable).  The cap-strided support/grad pools cost
+        KFS's LDS cost is bitmap + compact-set arrays (occupancy-aware
+        cap), so there is no hard K bound from the kernel itself; the
+        dense-node remainder runs the fused kernel where it covers, else
+        the separate chunked path — both subset-capable.
+        The cap-strided support/grad pools cost
         ~16·n_rows·cap bytes — at HBM-filling shapes (e.g. the 100M-edge
         R-MAT ws=8 slice: 266 GB of pools next to a 207 GB model) that
         budget doesn't exist, so the path stays off there.

@@ -2519,24 +2519,28 @@ __global__ void __launch_bounds__(BLOCK) kaf_support_t(
   }
 }
 
-// KFS: fused K1S+K2S — ONE launch per sweep for the routed nodes.
-// The first cut ran K1S and K2S separately with a dense K-column scan
-// and global compact-pool round-trips (8.0 + 5.6 ms/sweep measured at
-// the converged headline config); the fusion + the optimizations below
-// take the whole routed sweep under the dense kernel's time:
+// KFS: the fused sparse sweep — ONE launch per sweep for the routed
+// nodes.  Evolution is measured in profiles/r02_sparse_sweep.md; the
+// current (v10) structure:
 //   * ALL neighbor-list entries staged in LDS once (<= cap by routing;
 //     per-edge bases come FREE from the host's epos prefix — the same
 //     cumsum grad_ls_auto builds for the routing bounds);
+//   * the active set S_u = supp(gacc) ∪ supp(fu) is built FIRST from an
+//     LDS bitmap (phase 1 ORs bits while staging), so the accumulator
+//     gacc is COMPACT (cap slots, not K): no O(K) LDS residency, no
+//     O(K) zeroing, and staged entries transform to slot positions once
+//     (binary search) — after that every phase runs on LDS positions
+//     with zero global gathers;
 //   * WAVE-per-edge dot + scatter via LDS atomicAdd — barrier-free edge
 //     parallelism.  The atomics make gacc's fp32 summation order
 //     run-dependent (unlike every dense kernel): the sparse path's
 //     documented determinism trade (BIGCLAM_SPARSE=0 restores bitwise
 //     reproducibility);
-//   * the active set tracked in an LDS BITMAP (K/32 words): compact
-//     emission scans words, not K columns;
-//   * LDS diet: u16 column ids, storage-dtype values -> 31 KB (bf16) /
-//     38 KB (fp32) at K=5000, cap=K/4 -> 4-5 blocks/CU;
-//   * one packed 3-value block reduction (gg, fs, ff) instead of three.
+//   * the 16-candidate trial phase runs on an (edge-slot × candidate)
+//     THREAD mapping — no wave-wide reductions;
+//   * one packed 3-value block reduction (gg, fs, ff).
+// LDS = bitmap K/32 + cap*(16|20) bytes: 21 KB at the headline K=5000
+// bf16 (7 blocks/CU), ~78 KB at K=25000 with the occupancy-aware cap.
 
 template <bool BF16>
 __device__ __forceinline__ float vget(unsigned short v);
@@ -2566,24 +2570,20 @@ __global__ void __launch_bounds__(BLOCK) kfs_sparse_t(
   const int wid = tid >> 6;
   const int nw = (K + 31) >> 5;
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  // layout: gacc[K] f32 | gS[cap] f32 | fuS[cap] f32|u16 | nval[cap]
-  // f32|u16 | nidx[cap] u16 | kS[cap] u16 | bmap[nw] u32
-  float* gacc = reinterpret_cast<float*>(smem);
-  float* gS = gacc + K;
-  char* pv = reinterpret_cast<char*>(gS + cap);
-  float* fuS_f = reinterpret_cast<float*>(pv);
-  unsigned short* fuS_h = reinterpret_cast<unsigned short*>(pv);
-  pv += (size_t)cap * (BF16 ? 2 : 4);
+  // layout: bmap[nw] u32 | gacc[cap] f32 | gS[cap] f32 | nidx[cap] u16 |
+  //         kS[cap] u16 | nval[cap] u16|f32 | fuS[cap] u16|f32
+  u32* bmap = reinterpret_cast<u32*>(smem);
+  float* gacc = reinterpret_cast<float*>(bmap + nw);
+  float* gS = gacc + cap;
+  unsigned short* nidx = reinterpret_cast<unsigned short*>(gS + cap);
+  unsigned short* kS = nidx + cap;
+  char* pv = reinterpret_cast<char*>(kS + cap);
   float* nval_f = reinterpret_cast<float*>(pv);
   unsigned short* nval_h = reinterpret_cast<unsigned short*>(pv);
   pv += (size_t)cap * (BF16 ? 2 : 4);
-  unsigned short* nidx = reinterpret_cast<unsigned short*>(pv);
-  pv += (size_t)cap * 2;
-  unsigned short* kS = reinterpret_cast<unsigned short*>(pv);
-  pv += (size_t)cap * 2;
   pv = (char*)(((size_t)pv + 3) & ~(size_t)3);
-  u32* bmap = reinterpret_cast<u32*>(pv);
-  __shared__ float red[NWAVE];
+  float* fuS_f = reinterpret_cast<float*>(pv);
+  unsigned short* fuS_h = reinterpret_cast<unsigned short*>(pv);
   __shared__ int scan[NWAVE];
   __shared__ float s_lad[MAX_LS];
   __shared__ double cllh[16][MAX_LS];
@@ -2594,8 +2594,6 @@ __global__ void __launch_bounds__(BLOCK) kfs_sparse_t(
   __shared__ float sh_gg;
 
   if (tid < MAX_LS) s_lad[tid] = (tid < n_ladder) ? ladder[tid] : 0.f;
-  for (int k = tid * 4; k < K; k += BLOCK * 4)
-    *reinterpret_cast<float4*>(gacc + k) = float4{0.f, 0.f, 0.f, 0.f};
   for (int i = tid; i < nw; i += BLOCK) bmap[i] = 0u;
   __syncthreads();
 
@@ -2625,31 +2623,9 @@ __global__ void __launch_bounds__(BLOCK) kfs_sparse_t(
   }
   __syncthreads();
 
-  // phase 2: WAVE-per-edge dot (fu gathered L1-hot) -> w-weighted LDS
-  // atomicAdd scatter; per-wave edge llh partials
-  double llh_acc = 0.0;
-  if (phases & 1)
-  for (long long e = e0 + wid; e < e1; e += NWAVE) {
-    const int base = (int)(epos[e] - p0);
-    const int sv = (int)(epos[e + 1] - epos[e]);
-    float part = 0.f;
-    for (int j = lane; j < sv; j += WAVE) {
-      const float fv = BF16 ? vget<true>(nval_h[base + j]) : nval_f[base + j];
-      part += f_elem<BF16>(Fp, K, u, nidx[base + j]) * fv;
-    }
-    const float x = wave_allreduce_sum(part);
-    const float p = clamp_p(__expf(-x), min_p, max_p);
-    const float w = 1.f / (1.f - p);
-    if (lane == 0) llh_acc += (double)log1pf(-p) + (double)x;
-    for (int j = lane; j < sv; j += WAVE) {
-      const float fv = BF16 ? vget<true>(nval_h[base + j]) : nval_f[base + j];
-      atomicAdd(&gacc[nidx[base + j]], w * fv);
-    }
-  }
-  __syncthreads();
-
-  // phase 3: bitmap scan -> compact (k, g, fu) into LDS + global pools;
-  // 2-barrier block scan (wave shuffles + cross-wave offsets)
+  // phase 1.5: bitmap scan -> sorted compact kS + fuS gather + gacc
+  // slot zeroing; then transform the staged entries to SLOT POSITIONS
+  // (one binary search per entry) — no global gathers after this point
   const int wchunk = (nw + BLOCK - 1) / BLOCK;
   const int w0 = min(tid * wchunk, nw);
   const int w1 = min(w0 + wchunk, nw);
@@ -2670,44 +2646,80 @@ __global__ void __launch_bounds__(BLOCK) kfs_sparse_t(
     if (wv < wid) wbase += scan[wv];
   const int ns = scan[0] + scan[1] + scan[2] + scan[3];
   int pos = wbase + incl - cnt;
-  const long long go = goffset[blockIdx.x];
-  float gg_p = 0.f, fs_p = 0.f, ff_p = 0.f;
   for (int wv = w0; wv < w1; ++wv) {
     u32 bits = bmap[wv];
     while (bits) {
       const int b = __ffs(bits) - 1;
       bits &= bits - 1;
       const int k = (wv << 5) + b;
-      const float a = gacc[k];
-      const float f = f_elem<BF16>(Fp, K, u, k);
-      const float sfk = sumF[k];
-      const float g = a - sfk + f;
       kS[pos] = (unsigned short)k;
-      gS[pos] = g;
+      const float f = f_elem<BF16>(Fp, K, u, k);
       if (BF16)
         fuS_h[pos] = pack1_bf16_rne(f);
       else
         fuS_f[pos] = f;
-      gidx[go + pos] = k;
-      gval[go + pos] = g;
-      gg_p += g * g - sfk * sfk;
-      fs_p = fmaf(f, sfk, fs_p);
-      ff_p = fmaf(f, f, ff_p);
+      gacc[pos] = 0.f;
       ++pos;
     }
   }
-  // gacc is dead after emission: REUSE its LDS as an O(1) position map
-  // for phase 4 (replaces a per-element binary search — the search was
-  // ~25% of the kernel's VALU instructions)
   __syncthreads();
-  unsigned short* pmap = reinterpret_cast<unsigned short*>(gacc);
-  for (int i = tid; i < ns; i += BLOCK) pmap[kS[i]] = (unsigned short)i;
-  __syncthreads();
-  {  // transform the staged entries' column ids to COMPACT POSITIONS in
-     // place (phase 4 then reads one LDS word per element instead of a
-     // nidx -> pmap dependent chain)
+  {  // staged column ids -> compact positions, in place
     const int total = (int)(epos[e1] - p0);
-    for (int i = tid; i < total; i += BLOCK) nidx[i] = pmap[nidx[i]];
+    for (int i = tid; i < total; i += BLOCK) {
+      const unsigned short k = nidx[i];
+      int lo = 0, hi = ns;
+      while (lo < hi) {
+        const int mid = (lo + hi) >> 1;
+        if (kS[mid] < k)
+          lo = mid + 1;
+        else
+          hi = mid;
+      }
+      nidx[i] = (unsigned short)lo;  // always found: k came from bmap
+    }
+  }
+  __syncthreads();
+
+  // phase 2: WAVE-per-edge dot (all operands in LDS) -> w-weighted LDS
+  // atomicAdd scatter into the compact gacc; per-wave edge llh partials
+  double llh_acc = 0.0;
+  if (phases & 1)
+  for (long long e = e0 + wid; e < e1; e += NWAVE) {
+    const int base = (int)(epos[e] - p0);
+    const int sv = (int)(epos[e + 1] - epos[e]);
+    float part = 0.f;
+    for (int j = lane; j < sv; j += WAVE) {
+      const int p_ = (int)nidx[base + j];
+      const float fv = BF16 ? vget<true>(nval_h[base + j]) : nval_f[base + j];
+      const float fu = BF16 ? vget<true>(fuS_h[p_]) : fuS_f[p_];
+      part += fu * fv;
+    }
+    const float x = wave_allreduce_sum(part);
+    const float p = clamp_p(__expf(-x), min_p, max_p);
+    const float w = 1.f / (1.f - p);
+    if (lane == 0) llh_acc += (double)log1pf(-p) + (double)x;
+    for (int j = lane; j < sv; j += WAVE) {
+      const float fv = BF16 ? vget<true>(nval_h[base + j]) : nval_f[base + j];
+      atomicAdd(&gacc[nidx[base + j]], w * fv);
+    }
+  }
+  __syncthreads();
+
+  // phase 3: finalize g over the compact set + write the global pools;
+  // node terms fs/ff and gg accumulate alongside
+  const long long go = goffset[blockIdx.x];
+  float gg_p = 0.f, fs_p = 0.f, ff_p = 0.f;
+  for (int i = tid; i < ns; i += BLOCK) {
+    const int k = (int)kS[i];
+    const float f = BF16 ? vget<true>(fuS_h[i]) : fuS_f[i];
+    const float sfk = sumF[k];
+    const float g = gacc[i] - sfk + f;
+    gS[i] = g;
+    gidx[go + i] = k;
+    gval[go + i] = g;
+    gg_p += g * g - sfk * sfk;
+    fs_p = fmaf(f, sfk, fs_p);
+    ff_p = fmaf(f, f, ff_p);
   }
   // one packed reduction round: (gg, fs, ff) + the llh doubles
 #pragma unroll
@@ -2743,9 +2755,7 @@ __global__ void __launch_bounds__(BLOCK) kfs_sparse_t(
 
   // phase 4: 16-candidate trial scoring with an (edge-slot x candidate)
   // THREAD mapping: thread (e, j) serially accumulates one edge's
-  // candidate-j trial dot over the staged entries — no wave-wide
-  // reductions at all (the 17-shuffle butterflies were the phase's cost
-  // at these tiny per-node sizes; measured in the r04 phase bisect).
+  // candidate-j trial dot over the staged entries
   if (phases & 4) {
     const int ej = tid >> 4;  // edge slot 0..15
     const int jc = tid & 15;  // candidate 0..15
@@ -2759,12 +2769,11 @@ __global__ void __launch_bounds__(BLOCK) kfs_sparse_t(
         const int sv = (int)(epos[e + 1] - epos[e]);
         float x = 0.f;
         for (int t = 0; t < sv; ++t) {
-          const int pos = (int)nidx[base + t];  // positions since the
-                                                // in-place transform
+          const int p_ = (int)nidx[base + t];
           const float fv =
               BF16 ? vget<true>(nval_h[base + t]) : nval_f[base + t];
-          const float fu = BF16 ? vget<true>(fuS_h[pos]) : fuS_f[pos];
-          const float c = __builtin_amdgcn_fmed3f(fmaf(sj, gS[pos], fu),
+          const float fu = BF16 ? vget<true>(fuS_h[p_]) : fuS_f[p_];
+          const float c = __builtin_amdgcn_fmed3f(fmaf(sj, gS[p_], fu),
                                                   min_f, max_f);
           x = fmaf(c, fv, x);
         }
@@ -3423,8 +3432,8 @@ extern "C" void launch_kfs(const void* F, int bf16,
                            float min_f, float max_f, hipStream_t stream) {
   if (n_blocks == 0) return;
   if (n_ladder > 16) throw std::runtime_error("ladder length > 16 unsupported");
-  const size_t lds = (size_t)K * 4 + (size_t)cap * (bf16 ? 12 : 16) +
-                     ((K + 31) / 32) * 4 + 16;
+  const size_t lds = ((K + 31) / 32) * 4 + (size_t)cap * (bf16 ? 16 : 20)
+                     + 16;
   const char* ph = getenv("BIGCLAM_KFS_PHASES");  // measurement bisect only
   const int phases = ph ? atoi(ph) : 0xF;
   if (bf16) {
