@@ -803,3 +803,18 @@ def test_incremental_kaf_lists_match_full_rescan(monkeypatch):
             st._sp_sval[r * cap : r * cap + c],
             sval_inc[r * cap : r * cap + c], rtol=0, atol=0,
         )
+
+
+def test_select_k_gpu_small():
+    """v4 model selection end-to-end on GPU (each grid point a full fit
+    with the adaptive sweep)."""
+    from bigclam.engine.model_select import select_k
+
+    g = rmat_graph(10, 6.0, seed=71)
+    cfg = BigClamConfig(k=8, device="cuda", seed=2, max_sweeps=25,
+                        k_min=8, k_max=64, k_div=4, k_tol=1e-3)
+    out = select_k(g, cfg, init="seed")
+    assert out["k"] in out["grid"]
+    assert len(out["history"]) >= 2
+    lls = [h["llh"] for h in out["history"]]
+    assert all(np.isfinite(lls))
