@@ -53,6 +53,11 @@ class Trainer:
         device: Optional[torch.device] = None,
         metrics: Optional[MetricsLogger] = None,
     ):
+        if graph.num_nodes == 0:
+            raise ValueError(
+                "empty graph: no edges survived parsing (self-loops are "
+                "dropped; check the edge-list file)"
+            )
         self.graph = graph
         self.cfg = cfg
         self.rank = comm.get_rank() if rank is None else rank

@@ -264,3 +264,22 @@ def test_native_rmat_deterministic():
 
     g = build_graph(a)
     assert g.degrees().max() > 5 * g.degrees().mean()
+
+
+def test_empty_graph_friendly_error(tmp_path):
+    """A file with only comments/self-loops produces a clear error at
+    Trainer construction instead of an obscure downstream crash."""
+    import pytest as _pytest
+    import torch
+
+    from bigclam.config import BigClamConfig
+    from bigclam.engine.trainer import Trainer
+    from bigclam.io.edgelist import load_graph
+
+    p = tmp_path / "empty.txt"
+    p.write_text("# only a comment\n7 7\n")
+    g = load_graph(str(p))
+    assert g.num_nodes == 0
+    with _pytest.raises(ValueError, match="empty graph"):
+        Trainer(g, BigClamConfig(k=4, device="cpu"), rank=0, world_size=1,
+                device=torch.device("cpu"))
