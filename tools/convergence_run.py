@@ -37,6 +37,9 @@ RUNS = {
     "enron-bf16": ("enron", 500, "bf16"),
     "amazon-bf16": ("amazon", 5000, "bf16"),
     "amazon-fp32": ("amazon", 5000, "fp32"),
+    # BASELINE config #4's algorithmic core (com-Amazon K=25000), run at
+    # 1 GPU; the 8-GPU sharded version is the driver's SCALE territory
+    "amazon25k-bf16": ("amazon", 25000, "bf16"),
 }
 
 
