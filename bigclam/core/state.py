@@ -401,9 +401,12 @@ class ShardState:
         per = 16 if self.storage_dtype == torch.bfloat16 else 20
         nw4 = ((self.kp + 31) // 32) * 4
         # KFS v10's compact accumulator: LDS = bitmap + cap*per.  Target
-        # <= 78 KB so at least 2 blocks/CU always fit (the K=5000
-        # headline sits at ~21 KB -> 7 blocks).
-        lds_cap = (78 * 1024 - 4096 - nw4) // per
+        # ~32 KB -> 4-5 blocks/CU: a measured cap sweep at K=25000 put
+        # the 28.7 KB / 5-block point at 14.2 ms/sweep vs 21.5 at the
+        # 78 KB / 2-block sizing — occupancy beats routing the last few
+        # hub-adjacent nodes.  (The K=5000 headline stays at cap = K/4,
+        # 20.6 KB, 7 blocks/CU.)
+        lds_cap = (32 * 1024 - 1024 - nw4) // per
         return max(0, min(self.kp // 4, lds_cap) & ~7)
 
     @property
