@@ -649,6 +649,10 @@ def test_sparse_large_k_matches_dense(monkeypatch):
     g = rmat_graph(9, 5.0, seed=35)
     cfg = BigClamConfig(k=17000, device="cuda", dtype="bf16", seed=3,
                         max_sweeps=12, tol=0.0)
+    # the fixture's thresholded density (~1.5%) sits above the default
+    # occupancy-targeted cap's routing reach at this K — pin a wider cap
+    # (the test exercises the large-K mechanism, not the default policy)
+    monkeypatch.setenv("BIGCLAM_SPARSE_CAP", "4096")
     monkeypatch.setenv("BIGCLAM_SPARSE", "0")
     tr = Trainer(g, cfg, rank=0, world_size=1, device=torch.device("cuda"))
     # random init, then threshold to ~1.5% density (any valid F state is
