@@ -628,9 +628,36 @@ class ShardState:
         if pack is not None:
             # K3S commits AND rewrites the sparse rows' support lists,
             # so only DENSE-committed rows need a KAF rescan
-            _hip_ops().sparse_commit(self.F_local, pack, pack["best"],
-                                     self, self.cfg)
+            ops = _hip_ops()
+            ops.sparse_commit(self.F_local, pack, pack["best"], self,
+                              self.cfg)
             steps = pack["steps_dense"]
+            if self._kaf_valid and os.environ.get(
+                "BIGCLAM_SPARSE_COLSUM", "1"
+            ) != "0":
+                # list-based sumF refresh: the persistent lists are
+                # current for everything except the dense-committed
+                # (dirty) and over-cap rows, which the kernel reads
+                # dense — exact for the current F, ~60x fewer bytes
+                ops.apply_step(self.F_local, grad, steps, self.cfg)
+                self._dirty[: self.n_local] = (steps > 0).to(torch.uint8)
+                ns = (self.n_local + 511) // 512
+                if (
+                    self._colsum_partials is None
+                    or self._colsum_partials.shape[0] != ns
+                ):
+                    self._colsum_partials = torch.empty(
+                        ns, self.kp, device=self.device,
+                        dtype=torch.float32,
+                    )
+                ops.ensure_loaded().sparse_colsum(
+                    self.F_local, self._sp_soffset, self._sp_sidx,
+                    self._sp_sval, self._sp_scount, self._dirty,
+                    self._sp_cap, self._colsum_partials,
+                )
+                self.sumF = self._colsum_partials.sum(dim=0)
+                comm.all_reduce_(self.sumF)
+                return
         self.apply_step(grad, steps)
         # incremental KAF bookkeeping (halo rows stay always-dirty)
         if self._dirty is not None:

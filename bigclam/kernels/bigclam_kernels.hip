@@ -2886,6 +2886,65 @@ __global__ void __launch_bounds__(BLOCK) k3s_commit_t(
   }
 }
 
+// ------------------------------------------------- list-based column sum
+//
+// Sparse-path sumF refresh: column sums computed from the PERSISTENT
+// support lists instead of a dense read of F (~60x fewer bytes at
+// converged density).  Rows whose lists are not current — over-cap hubs
+// and rows the DENSE path just committed (dirty) — read their F row
+// directly, so the result is exact for the current F.  Per-stripe fp32
+// partials land in the same [n_stripes, K] buffer as k3_colsum_bf16;
+// stage 2 (partials.sum(0)) stays deterministic.  Within a block the
+// list entries accumulate via LDS atomics — the same run-to-run fp-order
+// trade the sparse gradient already makes (the dense path keeps the
+// bitwise-deterministic k3_colsum).
+
+#define KCS_ROWS 512
+#define KCS_KCH 8192  // 32 KB LDS chunk
+
+template <bool BF16>
+__global__ void __launch_bounds__(BLOCK) kcs_lists_t(
+    const void* __restrict__ Fp, int n_local, int K,
+    const long long* __restrict__ soffset, const int* __restrict__ sidx,
+    const float* __restrict__ sval, const int* __restrict__ scount,
+    const unsigned char* __restrict__ dirty, int cap,
+    float* __restrict__ partials) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  float* acc = reinterpret_cast<float*>(smem);  // KCS_KCH floats
+  const int r0 = blockIdx.x * KCS_ROWS;
+  const int r1 = min(n_local, r0 + KCS_ROWS);
+  const int kc0 = blockIdx.y * KCS_KCH;
+  const int kc1 = min(K, kc0 + KCS_KCH);
+  const int cw = kc1 - kc0;
+  const int tid = threadIdx.x;
+  for (int k = tid * 4; k < cw; k += BLOCK * 4)
+    *reinterpret_cast<float4*>(acc + k) = float4{0.f, 0.f, 0.f, 0.f};
+  __syncthreads();
+  for (int r = r0 + (tid >> 6); r < r1; r += NWAVE) {  // wave-per-row
+    const int lane = tid & (WAVE - 1);
+    const int c = scount[r];
+    if (c <= cap && !dirty[r]) {
+      // current list: add its entries that fall in this k-chunk
+      const long long off = soffset[r];
+      for (int j = lane; j < c; j += WAVE) {
+        const int k = sidx[off + j];
+        if (k >= kc0 && k < kc1) atomicAdd(&acc[k - kc0], sval[off + j]);
+      }
+    } else {
+      // stale/over-cap row: dense read of the F row chunk
+      for (int k = lane; k < cw; k += WAVE) {
+        const float f = f_elem<BF16>(Fp, K, r, kc0 + k);
+        if (f != 0.f) atomicAdd(&acc[k], f);
+      }
+    }
+  }
+  __syncthreads();
+  float* __restrict__ out = partials + (size_t)blockIdx.x * K + kc0;
+  for (int k = tid * 4; k < cw; k += BLOCK * 4)
+    *reinterpret_cast<float4*>(out + k) =
+        *reinterpret_cast<const float4*>(acc + k);
+}
+
 // ------------------------------------------------------------------- K6
 //
 // Seed-init F scatter (codes/bigclamv3-7.scala:60-87): community c's
@@ -3487,6 +3546,28 @@ extern "C" void launch_k6(void* F, int bf16, int K, const long long* sindptr,
     hipLaunchKernelGGL((k6_seed_init_t<false>), dim3(n_seeds), dim3(BLOCK),
                        0, stream, F, K, sindptr, snbrs, seeds, n_seeds,
                        start, stop, include_seed);
+  }
+  HIP_CHECK(hipGetLastError());
+}
+
+extern "C" void launch_kcs_lists(const void* F, int bf16, int n_local,
+                                 int K, const long long* soffset,
+                                 const int* sidx, const float* sval,
+                                 const int* scount,
+                                 const unsigned char* dirty, int cap,
+                                 float* partials, hipStream_t stream) {
+  if (n_local == 0) return;
+  const int ns = (n_local + KCS_ROWS - 1) / KCS_ROWS;
+  const int nk = (K + KCS_KCH - 1) / KCS_KCH;
+  const size_t lds = (size_t)(K < KCS_KCH ? ((K + 3) & ~3) : KCS_KCH) * 4;
+  if (bf16) {
+    hipLaunchKernelGGL((kcs_lists_t<true>), dim3(ns, nk), dim3(BLOCK), lds,
+                       stream, F, n_local, K, soffset, sidx, sval, scount,
+                       dirty, cap, partials);
+  } else {
+    hipLaunchKernelGGL((kcs_lists_t<false>), dim3(ns, nk), dim3(BLOCK), lds,
+                       stream, F, n_local, K, soffset, sidx, sval, scount,
+                       dirty, cap, partials);
   }
   HIP_CHECK(hipGetLastError());
 }

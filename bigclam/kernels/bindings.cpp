@@ -58,6 +58,10 @@ extern "C" void launch_k1_chunked(const void*, int, const long long*,
 extern "C" void launch_k6(void*, int, int, const long long*,
                           const long long*, const long long*, int, long long,
                           long long, int, hipStream_t);
+extern "C" void launch_kcs_lists(const void*, int, int, int,
+                                 const long long*, const int*, const float*,
+                                 const int*, const unsigned char*, int,
+                                 float*, hipStream_t);
 extern "C" void launch_k7_count(const void*, int, int, int, int, float, int*,
                                 hipStream_t);
 extern "C" void launch_k7_fill(const void*, int, int, int, int, float,
@@ -475,6 +479,29 @@ void seed_init(torch::Tensor F_local, torch::Tensor sindptr,
             include_seed ? 1 : 0, current_stream());
 }
 
+// List-based column sums for the sparse path (see kcs_lists_t).
+void sparse_colsum(torch::Tensor F_local, torch::Tensor soffset,
+                   torch::Tensor sidx, torch::Tensor sval,
+                   torch::Tensor scount, torch::Tensor dirty, int64_t cap,
+                   torch::Tensor partials) {
+  CHECK_F(F_local);
+  CHECK_IN(soffset, torch::kInt64);
+  CHECK_IN(sidx, torch::kInt32);
+  CHECK_IN(sval, torch::kFloat32);
+  CHECK_IN(scount, torch::kInt32);
+  CHECK_IN(partials, torch::kFloat32);
+  TORCH_CHECK(dirty.scalar_type() == torch::kUInt8 && dirty.is_contiguous());
+  const int n_local = (int)F_local.size(0);
+  const int K = (int)F_local.size(1);
+  TORCH_CHECK(partials.size(0) == (n_local + 511) / 512 &&
+              partials.size(1) == K);
+  launch_kcs_lists(F_local.data_ptr(), is_bf16(F_local) ? 1 : 0, n_local, K,
+                   i64p(soffset), sidx.data_ptr<int>(),
+                   sval.data_ptr<float>(), scount.data_ptr<int>(),
+                   dirty.data_ptr<uint8_t>(), (int)cap,
+                   partials.data_ptr<float>(), current_stream());
+}
+
 // K7 community extraction: two deterministic passes (count, then fill
 // after a host/torch prefix-sum) — see k7_membership in the .hip file.
 void extract_count(torch::Tensor F_local, int64_t k_true, double delta,
@@ -540,6 +567,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("sparse_fused", &sparse_fused,
         "KFS: fused compact gradient + 16-candidate Armijo for routed "
         "nodes (LDS bitmap active sets)");
+  m.def("sparse_colsum", &sparse_colsum,
+        "list-based column sums (sparse path; stale/over-cap rows dense)");
   m.def("sparse_commit", &sparse_commit,
         "K3S: sparse projected commit confined to the active set");
   m.def("edge_grad_llh_chunked", &edge_grad_llh_chunked,

@@ -822,3 +822,20 @@ def test_select_k_gpu_small():
     assert len(out["history"]) >= 2
     lls = [h["llh"] for h in out["history"]]
     assert all(np.isfinite(lls))
+
+
+@pytest.mark.parametrize("dtype", ["bf16", "fp32"])
+def test_sparse_colsum_invariant(dtype, monkeypatch):
+    """With the list-based sumF refresh active, the sumF == colsum(F)
+    invariant holds (exact modulo fp32 summation order) after several
+    sparse sweeps with mixed sparse/dense commits."""
+    monkeypatch.setenv("BIGCLAM_SPARSE", "1")
+    monkeypatch.setenv("BIGCLAM_SPARSE_CAP", "1024")
+    g = rmat_graph(10, 6.0, seed=65)
+    cfg = BigClamConfig(k=192, device="cuda", dtype=dtype, seed=6, tol=0.0,
+                        max_sweeps=12)
+    tr = Trainer(g, cfg, rank=0, world_size=1, device=torch.device("cuda"))
+    tr.fit(init="seed")
+    st = tr.state
+    ref = st.F_local.float().sum(dim=0)
+    torch.testing.assert_close(st.sumF, ref, rtol=1e-5, atol=1e-3)
