@@ -606,10 +606,13 @@ class ShardState:
                         self.F, self.indptr, self.indices, self.sumF, grad,
                         llh, order_d, self.cfg, out=best,
                     )
+        # grad pools sized at the n_local bound so the backing buffers
+        # allocate ONCE (growing them as more nodes route caused 60-90 ms
+        # hipMalloc spikes mid-fit)
         pack = ops.sparse_sweep_part(
             self.F, self.indptr, self.indices, self.sumF, order_s,
-            soffset, sidx, sval, scount, cs, goffset, n_s * cap, cap, llh,
-            best, self.cfg,
+            soffset, sidx, sval, scount, cs, goffset,
+            self.n_local * cap, cap, llh, best, self.cfg,
         )
         if n_d:
             torch.cuda.current_stream(dev).wait_stream(self._side_stream)
