@@ -134,6 +134,8 @@ class ShardState:
         self._sp_scount: Optional[torch.Tensor] = None
         self._sp_sidx: Optional[torch.Tensor] = None
         self._sp_sval: Optional[torch.Tensor] = None
+        self._sp_gidx: Optional[torch.Tensor] = None
+        self._sp_gval: Optional[torch.Tensor] = None
         self._sp_cap: Optional[int] = None
         self._dirty: Optional[torch.Tensor] = None  # uint8 [n_rows]
         self._kaf_valid = False  # persistent support lists match F
@@ -515,6 +517,12 @@ class ShardState:
             self._sp_sval = torch.empty(
                 n_rows * cap, device=dev, dtype=torch.float32
             )
+            self._sp_gidx = torch.empty(
+                self.n_local * cap, device=dev, dtype=torch.int32
+            )
+            self._sp_gval = torch.empty(
+                self.n_local * cap, device=dev, dtype=torch.float32
+            )
             self._dirty = torch.ones(n_rows, device=dev, dtype=torch.uint8)
             self._kaf_valid = False
         soffset = self._sp_soffset
@@ -613,6 +621,7 @@ class ShardState:
             self.F, self.indptr, self.indices, self.sumF, order_s,
             soffset, sidx, sval, scount, cs, goffset,
             self.n_local * cap, cap, llh, best, self.cfg,
+            state_pools=(self._sp_gidx, self._sp_gval),
         )
         if n_d:
             torch.cuda.current_stream(dev).wait_stream(self._side_stream)

@@ -272,6 +272,7 @@ def sparse_sweep_part(
     llh_out: torch.Tensor,
     best_out: torch.Tensor,
     cfg: BigClamConfig,
+    state_pools=None,
 ):
     """KFS for the routed (sparse) nodes: one fused launch computing the
     compact gradient pools, llh per node, and the Armijo best step.
@@ -280,8 +281,10 @@ def sparse_sweep_part(
     ext = ensure_loaded()
     dev = F.device
     n_s = int(order_sparse.numel())
-    gidx = _pool(dev, "gidx", gpool_size, torch.int32)
-    gval = _pool(dev, "gval", gpool_size, torch.float32)
+    # per-STATE grad pools (via state_pools): the commit consumes them at
+    # the START of the next sweep (the pipelined carry), so a shared
+    # module pool could be clobbered by another coexisting state
+    gidx, gval = state_pools
     gcount = torch.empty(n_s, device=dev, dtype=torch.int32)
     GG = (sumF * sumF).sum().reshape(1)  # device scalar: no host sync
     ext.sparse_fused(
