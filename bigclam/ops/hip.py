@@ -267,7 +267,7 @@ def sparse_sweep_part(
     scount: torch.Tensor,
     epos: torch.Tensor,
     goffset: torch.Tensor,
-    gpool_size: int,
+    gpool_size: int,  # kept for API stability; pools come from the state
     cap: int,
     llh_out: torch.Tensor,
     best_out: torch.Tensor,
@@ -299,20 +299,6 @@ def sparse_sweep_part(
         "gval": gval,
         "gcount": gcount,
     }
-
-
-_pools = {}
-
-
-def _pool(dev, name, size, dtype) -> torch.Tensor:
-    """Grow-only cached device buffer (avoids per-sweep allocator churn
-    and the size-sync it would force)."""
-    key = (str(dev), name)
-    t = _pools.get(key)
-    if t is None or t.numel() < size:
-        t = torch.empty(size, device=dev, dtype=dtype)
-        _pools[key] = t
-    return t
 
 
 def sparse_commit(F_local: torch.Tensor, pack: dict, best: torch.Tensor,
