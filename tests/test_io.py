@@ -283,3 +283,39 @@ def test_empty_graph_friendly_error(tmp_path):
     with _pytest.raises(ValueError, match="empty graph"):
         Trainer(g, BigClamConfig(k=4, device="cpu"), rank=0, world_size=1,
                 device=torch.device("cpu"))
+
+
+@pytest.mark.parametrize(
+    "name,edges",
+    [
+        ("star_hub", [[0, i] for i in range(1, 1001)]),
+        ("dup_heavy", [[1, 2], [2, 1], [1, 2], [3, 3], [2, 3]] * 10),
+        ("single_edge", [[7, 9]]),
+        # sparse 64-bit-ish id space just under the native builder's
+        # 2^31 presence-bitmap bound (io_native.cpp build_csr)
+        ("boundary_ids", [[0, 1 << 27], [5, 1 << 27], [0, 5], [123456789, 5]]),
+    ],
+)
+def test_native_csr_adversarial_shapes(monkeypatch, name, edges):
+    """Native build_csr == numpy reference on hub rows, heavy duplication,
+    minimal inputs and sparse high-id spaces (same dedupe/self-loop/remap
+    semantics, bit-identical CSR)."""
+    pytest.importorskip("bigclam._io_native")
+    e = np.asarray(edges, dtype=np.int64)
+    monkeypatch.setenv("BIGCLAM_NATIVE_CSR", "1")
+    g1 = build_graph(e)
+    monkeypatch.setenv("BIGCLAM_NATIVE_CSR", "0")
+    g2 = build_graph(e)
+    np.testing.assert_array_equal(g1.indptr, g2.indptr)
+    np.testing.assert_array_equal(g1.indices, g2.indices)
+    np.testing.assert_array_equal(g1.raw_ids, g2.raw_ids)
+
+
+def test_native_csr_large_id_fallback(monkeypatch):
+    """Raw ids >= 2^31 must take the numpy path (the native builder's
+    presence bitmap cannot cover them) and still build correctly."""
+    e = np.array([[1 << 33, 4], [4, 1 << 40], [1 << 33, 1 << 40]], dtype=np.int64)
+    monkeypatch.setenv("BIGCLAM_NATIVE_CSR", "1")
+    g = build_graph(e)
+    assert g.num_nodes == 3 and g.num_edges == 3
+    assert list(g.raw_ids) == [4, 1 << 33, 1 << 40]
