@@ -86,8 +86,10 @@ def load_F_slice(dirpath: str, start: int, stop: int) -> np.ndarray:
             os.path.join(dirpath, f"F_rank{r}.npy"), mmap_mode="r"
         )[lo - rs : hi - rs]
         if meta["dtype"] == "bf16":
+            # copy: the mmap slice is read-only and non-contiguous rows
+            # can't be viewed as bf16 anyway
             arr = (
-                torch.from_numpy(np.ascontiguousarray(arr))
+                torch.from_numpy(np.array(arr, copy=True))
                 .view(torch.bfloat16)
                 .float()
                 .numpy()
