@@ -36,7 +36,11 @@ def _add_common(p: argparse.ArgumentParser):
     p.add_argument("--beta", type=float, default=d.beta)
     p.add_argument("--ls-steps", type=int, default=d.ls_steps)
     p.add_argument("--tol", type=float, default=d.tol)
-    p.add_argument("--max-sweeps", type=int, default=d.max_sweeps)
+    p.add_argument(
+        "--max-sweeps", type=int, default=d.max_sweeps,
+        help="sweep budget for THIS invocation (with --resume: additional "
+        "sweeps on top of the checkpoint's sweep count)",
+    )
     p.add_argument("--dtype", choices=["fp32", "bf16"], default=d.dtype)
     p.add_argument("--device", default=None, help="cuda|cpu (auto)")
     p.add_argument("--seed", type=int, default=d.seed)

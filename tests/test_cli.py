@@ -128,3 +128,28 @@ def test_cli_bench_delegates(capfd):
     assert rec["data"] == "synthetic"
     assert rec["value"] > 0 and rec["ms_per_step"] > 0
     assert "parallelism" in rec["config"] and "k" in rec["config"]
+
+
+def test_cli_fit_resume(tmp_path):
+    """CLI --resume continues a checkpointed fit: the resumed run picks up
+    the saved sweep/llh and matches an uninterrupted run's final LLH."""
+    g, _ = planted_partition(3, 12, p_in=0.5, p_out=0.02, seed=23)
+    edges = tmp_path / "g.txt"
+    _write_edgelist(edges, g)
+    base = ["fit", str(edges), "--k", "3", "--device", "cpu",
+            "--init", "random", "--quiet"]
+    # uninterrupted 8-sweep run
+    r_full = _run(base + ["--max-sweeps", "8", "--tol", "0"])
+    assert r_full.returncode == 0, r_full.stderr
+    llh_full = json.loads(r_full.stdout.strip().splitlines()[-1])["llh"]
+    # 4 sweeps -> checkpoint -> resume 4 more
+    ck = tmp_path / "ck"
+    r1 = _run(base + ["--max-sweeps", "4", "--tol", "0",
+                      "--checkpoint-dir", str(ck)])
+    assert r1.returncode == 0, r1.stderr
+    # max-sweeps counts sweeps RUN THIS INVOCATION (4 more -> 8 total)
+    r2 = _run(base + ["--max-sweeps", "4", "--tol", "0",
+                      "--resume", str(ck)])
+    assert r2.returncode == 0, r2.stderr
+    res2 = json.loads(r2.stdout.strip().splitlines()[-1])
+    assert res2["llh"] == pytest.approx(llh_full, rel=1e-12)
