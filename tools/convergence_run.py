@@ -32,6 +32,8 @@ SHAPES = {
     "enron": (36692, 183831),
     # com-Amazon: 334,863 nodes / 925,872 undirected edges (SNAP; missing blob upstream)
     "amazon": (334863, 925872),
+    # com-Youtube: 1,134,890 nodes / 2,987,624 undirected edges (data/com-youtube.ungraph.txt)
+    "youtube": (1134890, 2987624),
 }
 RUNS = {
     "enron-bf16": ("enron", 500, "bf16"),
@@ -40,6 +42,9 @@ RUNS = {
     # BASELINE config #4's algorithmic core (com-Amazon K=25000), run at
     # 1 GPU; the 8-GPU sharded version is the driver's SCALE territory
     "amazon25k-bf16": ("amazon", 25000, "bf16"),
+    # the reference v3's actual experiment: com-Youtube at its fixed
+    # K=8385 (codes/bigclamv3-7.scala:15,26)
+    "youtube-bf16": ("youtube", 8385, "bf16"),
 }
 
 
