@@ -104,10 +104,16 @@ def _load(args):
 
 
 def main(argv=None):
-    ap = argparse.ArgumentParser(prog="bigclam")
+    ap = argparse.ArgumentParser(
+        prog="bigclam",
+        description="BigCLAM overlapping community detection, "
+        "MI355X-native (see README.md)",
+    )
     sub = ap.add_subparsers(dest="cmd", required=True)
 
-    p_fit = sub.add_parser("fit")
+    p_fit = sub.add_parser(
+        "fit", help="fit a K-community model on an edge list, "
+        "write communities + checkpoint")
     p_fit.add_argument("edgelist")
     _add_common(p_fit)
     p_fit.add_argument(
@@ -117,7 +123,9 @@ def main(argv=None):
         help="resume the fit from a checkpoint directory (any world size)",
     )
 
-    p_sel = sub.add_parser("select-k")
+    p_sel = sub.add_parser(
+        "select-k", help="model selection over the geometric K grid "
+        "(the reference v4 semantics)")
     p_sel.add_argument("edgelist")
     _add_common(p_sel)
     p_sel.add_argument("--k-min", dest="k_min", type=int, default=1000)
@@ -125,13 +133,15 @@ def main(argv=None):
     p_sel.add_argument("--k-div", dest="k_div", type=int, default=100)
     p_sel.add_argument("--k-tol", dest="k_tol", type=float, default=1e-3)
 
-    p_ext = sub.add_parser("extract")
+    p_ext = sub.add_parser(
+        "extract", help="re-extract communities from a checkpoint")
     p_ext.add_argument("checkpoint_dir")
     p_ext.add_argument("edgelist")
     p_ext.add_argument("--out", required=True)
 
     sub.add_parser(
         "bench",
+        help="flagship benchmark (forwards to bench.py)",
         add_help=False,
         description="flagship benchmark; all flags forwarded to bench.py",
     )
